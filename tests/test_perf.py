@@ -1,0 +1,141 @@
+"""End-to-end PerfLLM tests: schedules, memory model, search, MoE."""
+
+import pytest
+
+from simumax_amd import (ModelConfig, PerfLLM, StrategyConfig, SystemConfig,
+                         get_simu_model_config, get_simu_strategy_config,
+                         get_simu_system_config)
+from simumax_amd.perf.perf_llm import (estimate_straggler_increase_ratio,
+                                       schedule_1f1b)
+
+
+def build(model="llama3-8b", strategy="tp1_pp2_dp4_mbs1", system="mi355x", **over):
+    p = PerfLLM()
+    st = StrategyConfig.init_from_config_file(get_simu_strategy_config(strategy))
+    for k, v in over.items():
+        setattr(st, k, v)
+    p.configure(
+        st,
+        ModelConfig.init_from_config_file(get_simu_model_config(model)),
+        SystemConfig.init_from_config_file(get_simu_system_config(system)),
+    )
+    p.run_estimate()
+    return p
+
+
+def test_1f1b_uniform_closed_form():
+    """Uniform stages, no p2p: total = (mbc + pp - 1) * (F + B)."""
+    pp, mbc, F, B = 4, 8, 1.0, 2.0
+    total, recs = schedule_1f1b(pp, mbc, [F] * pp, [B] * pp, 0.0)
+    assert total == pytest.approx((mbc + pp - 1) * (F + B))
+    assert len(recs) == pp * mbc * 2
+
+
+def test_1f1b_pp1_equivalent():
+    total, _ = schedule_1f1b(1, 4, [1.0], [2.0], 0.0)
+    assert total == pytest.approx(12.0)
+
+
+def test_perf_llama3_8b_tp1_pp2():
+    p = build()
+    cost = p.analysis_cost()
+    mem = p.analysis_mem()
+    assert 0.05 < cost["mfu"] < 0.95
+    assert cost["iter_time"] > 0
+    assert mem["max_peak_mem"] > 10 * 2**30
+    assert not mem["oom"]
+    # pipeline ideal time <= pipeline_time (bubble >= 0)
+    assert cost["bubble_time"] >= -1e-6
+
+
+def test_mem_scaling_with_tp():
+    p1 = build(strategy="tp1_pp1_dp8_mbs1")
+    p8 = build(strategy="tp8_pp1_dp1_mbs1")
+    m1 = p1.analysis_mem()["max_peak_mem"]
+    m8 = p8.analysis_mem()["max_peak_mem"]
+    assert m8 < m1 / 2  # TP8 shards weights+acts
+
+
+def test_recompute_reduces_cache():
+    base = build(strategy="tp1_pp1_dp8_mbs1")
+    rc = build(strategy="tp1_pp1_dp8_mbs1", enable_recompute=True,
+               recompute_granularity="full_block")
+    c_base = base.chunks[0].peak_point.cache_mem
+    c_rc = rc.chunks[0].peak_point.cache_mem
+    assert c_rc < 0.2 * c_base
+    # recompute adds time
+    assert rc.analysis_cost()["iter_time"] > base.analysis_cost()["iter_time"]
+
+
+def test_selective_recompute_between():
+    base = build(strategy="tp1_pp1_dp8_mbs1")
+    sel = build(strategy="tp1_pp1_dp8_mbs1", enable_recompute=True,
+                recompute_granularity="selective_recompute")
+    full = build(strategy="tp1_pp1_dp8_mbs1", enable_recompute=True,
+                 recompute_granularity="full_block")
+    cb = base.chunks[0].peak_point.cache_mem
+    cs = sel.chunks[0].peak_point.cache_mem
+    cf = full.chunks[0].peak_point.cache_mem
+    assert cf < cs < cb
+
+
+def test_activation_replay_consistency():
+    p = build()
+    for chunk in p.chunks:
+        pp = chunk.peak_point
+        assert pp.fwd_peak_mem >= pp.cache_mem > 0
+        assert pp.bwd_peak_mem >= 0
+
+
+def test_straggler_model():
+    assert estimate_straggler_increase_ratio(1) == 1.0
+    r4 = estimate_straggler_increase_ratio(4)
+    assert r4 == pytest.approx(1 + 4 / 5 * 0.09 * (2 ** 0.5))
+
+
+def test_moe_deepseek():
+    p = build(model="deepseekv2-l4", strategy="ep8_pp1_dp8_mbs1")
+    cost = p.analysis_cost()
+    mem = p.analysis_mem()
+    assert cost["mfu"] > 0.02
+    assert not mem["oom"]
+    # EP all2all events exist
+    evs = [ev for ev in p.chunks[0].all_comm_ops() if ev.comm_stage == "ep"]
+    assert any(ev.op_name == "all2all" for ev in evs)
+    # expert state is sharded over edp=1 here (ep8 world8 -> edp1)
+    mi = p.chunks[0].get_model_info()
+    assert mi.moe_weight_bytes > 0
+
+
+def test_analysis_op_info_keys():
+    p = build()
+    ops = p.analysis_op_info()
+    assert "matmul" in ops and "sdp_fwd" in ops
+    for desc in ops["matmul"]:
+        assert desc.startswith("b=") and "layout=" in desc
+    for desc in ops["sdp_fwd"]:
+        assert desc.startswith("batch=")
+
+
+def test_search_max_mbs():
+    p = build(strategy="tp8_pp1_dp1_mbs1")
+    best = p.search_max_micro_batch_size(max_mbs=4)
+    assert best is not None and best >= 1
+
+
+def test_vpp_estimate_smaller_bubble():
+    p1 = build(model="llama3-70b-l12", strategy="tp1_pp2_dp4_mbs1",
+               micro_batch_num=8)
+    pv = build(model="llama3-70b-l12", strategy="tp1_pp2_dp4_mbs1",
+               micro_batch_num=8, interleaving_size=2)
+    b1 = p1.analysis_cost()["bubble_time"]
+    bv = pv.analysis_cost()["bubble_time"]
+    assert bv < b1
+
+
+def test_analysis_writes_artifacts(tmp_path):
+    p = build(model="llama2-tiny", strategy="tp1_pp1_dp8_mbs1")
+    p.analysis(str(tmp_path / "out"))
+    for f in ("compute_result.json", "mem_result.json", "base_info.json",
+              "net_info.json", "model_arch"):
+        assert (tmp_path / "out" / f).exists()
