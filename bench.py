@@ -1,0 +1,187 @@
+#!/usr/bin/env python3
+"""Headline benchmark: predicted-vs-measured MFU & peak-memory error for
+Llama-3-8B training, world_size 1/2/4/8 (BASELINE.json metric).
+
+Runs the in-repo Megatron-ROCm-style trainer (gfx950 HIP kernels +
+hipBLASLt GEMMs + RCCL DP) on synthetic data / random-init weights, times
+K steps after W warmup, then runs PerfLLM's prediction for the identical
+config and reports the error.
+
+Launch: python bench.py --gpus N --steps K --warmup W
+(N>1 is launched as one torchrun rank per GPU; RANK/WORLD_SIZE from env.)
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+MODEL_NAME = "llama3-8b"
+SEQ_LEN = 4096
+MICRO_BATCH_SIZE = 1
+MICRO_BATCH_NUM = 4
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--model", type=str, default=MODEL_NAME)
+    ap.add_argument("--seq-len", type=int, default=SEQ_LEN)
+    ap.add_argument("--mbs", type=int, default=MICRO_BATCH_SIZE)
+    ap.add_argument("--mbc", type=int, default=MICRO_BATCH_NUM)
+    ap.add_argument("--layers", type=int, default=0,
+                    help="override layer count (0 = full model)")
+    return ap.parse_args()
+
+
+def predict(model_cfg, world, args):
+    """PerfLLM prediction for the trainer's exact config."""
+    from simumax_amd import (PerfLLM, StrategyConfig, SystemConfig,
+                             get_simu_system_config)
+
+    st = StrategyConfig(
+        seq_len=args.seq_len,
+        micro_batch_size=args.mbs,
+        micro_batch_num=args.mbc,
+        world_size=world,
+        tp_size=1, pp_size=1, ep_size=1,
+        enable_sequence_parallel=False,
+        zero_state=0,                # trainer replicates optimizer state
+        use_fp32_accum_grad=True,
+        enable_recompute=False,
+        overlap_grad_reduce=True,
+        cross_entropy_loss_fusion=True,
+        attention_sparse_ratio=0.5,  # causal flash attention
+        enable_dropout=False,
+        mem_factor=1.0,
+    )
+    p = PerfLLM()
+    sysc = SystemConfig.init_from_config_file(get_simu_system_config("mi355x"))
+    import copy
+
+    mc = copy.deepcopy(model_cfg)
+    mc.padded_vocab_size = False  # trainer uses raw vocab
+    p.configure(st, mc, sysc)
+    p.run_estimate()
+    cost = p.analysis_cost()
+    mem = p.analysis_mem()
+    return cost, mem
+
+
+def main():
+    args = parse_args()
+    import torch
+    import torch.distributed as dist
+
+    from simumax_amd import ModelConfig, get_simu_model_config
+
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", args.gpus))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    distributed = world > 1
+    if distributed:
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+
+    model_cfg = ModelConfig.init_from_config_file(
+        get_simu_model_config(args.model))
+    if args.layers:
+        model_cfg.layer_num = args.layers
+
+    from simumax_amd.train.trainer import (TrainConfig, build_trainer,
+                                           make_synthetic_batch, train_step)
+
+    tc = TrainConfig(seq_len=args.seq_len, micro_batch_size=args.mbs,
+                     micro_batch_num=args.mbc)
+    device = f"cuda:{local_rank}"
+    t0 = time.time()
+    model, opt, reducer = build_trainer(model_cfg, tc, device)
+    if rank == 0:
+        print(f"[bench] built {args.model} ({model.num_params()/1e9:.2f}B params) "
+              f"in {time.time()-t0:.1f}s", file=sys.stderr)
+    toks, labels = make_synthetic_batch(model_cfg.vocab_size, args.mbc,
+                                        args.mbs, args.seq_len, device,
+                                        seed=1000 + rank)
+
+    for _ in range(args.warmup):
+        train_step(model, opt, reducer, toks, labels, args.mbc)
+    torch.cuda.reset_peak_memory_stats()
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t_start = time.time()
+    for _ in range(args.steps):
+        train_step(model, opt, reducer, toks, labels, args.mbc)
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.time() - t_start
+
+    # max over ranks
+    el = torch.tensor([elapsed], device=device)
+    pk = torch.tensor([float(torch.cuda.max_memory_allocated())], device=device)
+    if distributed:
+        dist.all_reduce(el, op=dist.ReduceOp.MAX)
+        dist.all_reduce(pk, op=dist.ReduceOp.MAX)
+    ms_per_step = el.item() / args.steps * 1e3
+    peak_bytes = pk.item()
+
+    if rank == 0:
+        tokens_per_iter = args.mbs * args.mbc * world * args.seq_len
+        flops_token = model_cfg.flops_per_token(args.seq_len)
+        peak_tflops = 2500.0
+        measured_mfu = (flops_token * tokens_per_iter / (ms_per_step / 1e3)
+                        / (world * peak_tflops * 1e12))
+        tokens_per_s = tokens_per_iter / (ms_per_step / 1e3)
+
+        cost, mem = predict(model_cfg, world, args)
+        pred_ms = cost["iter_time"]
+        pred_mfu = cost["mfu"]
+        pred_peak = mem["max_peak_mem"]
+        time_err = (pred_ms - ms_per_step) / ms_per_step * 100.0
+        mem_err = (pred_peak - peak_bytes) / peak_bytes * 100.0
+        value = abs(time_err)
+
+        out = {
+            "metric": "predicted-vs-measured MFU & peak-mem error (%), "
+                      "Llama-3-8B at world_size 1/2/4/8",
+            "value": round(value, 3),
+            "unit": "%",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.mbs * args.mbc * world,
+                "seq_len": args.seq_len,
+                "parallelism": f"dp{world}",
+                "layers": model_cfg.layer_num,
+                "timing_error_pct": round(time_err, 3),
+                "mem_error_pct": round(mem_err, 3),
+                "measured_ms_per_step": round(ms_per_step, 2),
+                "predicted_ms_per_step": round(pred_ms, 2),
+                "measured_mfu": round(measured_mfu, 4),
+                "predicted_mfu": round(pred_mfu, 4),
+                "measured_peak_gib": round(peak_bytes / 2**30, 2),
+                "predicted_peak_gib": round(pred_peak / 2**30, 2),
+                "tokens_per_s": round(tokens_per_s, 1),
+            },
+        }
+        print(json.dumps(out))
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,94 @@
+"""Megatron-style Llama training model used as the real-run side of the
+perf-vs-real validation (the reference validates against Megatron-LM runs:
+tools/b200/run_megatron_perf_real_pipeline.py; here the trainer is in-repo
+and ROCm-native).
+
+Uses the gfx950 HIP kernels for the fused hot ops (RMSNorm, RoPE, SwiGLU,
+flash attention, fused CE); plain linear layers go through hipBLASLt via
+torch.matmul. Activation-save behavior deliberately matches the
+simulator's per-leaf cache accounting (simumax_amd/ops/dense.py docstring).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+
+from ..core.config import ModelConfig
+from ..kernels import ops as K
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None):
+        super().__init__()
+        h = cfg.hidden_size
+        self.heads = cfg.head_num
+        self.kv_heads = cfg.kv_head_num
+        self.head_size = cfg.head_size
+        qkv_out = (cfg.head_num + 2 * cfg.kv_head_num) * cfg.head_size
+        self.input_norm = K.RMSNorm(h, dtype=dtype, device=device)
+        self.qkv_proj = nn.Linear(h, qkv_out, bias=False, dtype=dtype, device=device)
+        self.out_proj = nn.Linear(cfg.head_num * cfg.head_size, h, bias=False,
+                                  dtype=dtype, device=device)
+        self.pre_mlp_norm = K.RMSNorm(h, dtype=dtype, device=device)
+        assert cfg.use_swiglu
+        self.fc1 = nn.Linear(h, 2 * cfg.intermediate_size, bias=False, dtype=dtype, device=device)
+        self.fc2 = nn.Linear(cfg.intermediate_size, h, bias=False, dtype=dtype, device=device)
+
+    def forward(self, x, rope_cs, pos):
+        # x: [B, S, H]
+        B, S, H = x.shape
+        res = x
+        y = self.input_norm(x)
+        qkv = self.qkv_proj(y)
+        d = self.head_size
+        q, k, v = qkv.split(
+            [self.heads * d, self.kv_heads * d, self.kv_heads * d], dim=-1)
+        q = K.apply_rope(q.reshape(B * S, self.heads, d), rope_cs, pos)
+        k = K.apply_rope(k.reshape(B * S, self.kv_heads, d), rope_cs, pos)
+        q = q.view(B, S, self.heads, d)
+        k = k.view(B, S, self.kv_heads, d)
+        v = v.reshape(B, S, self.kv_heads, d)
+        ctx = K.flash_attention(q, k, v, causal=True)
+        x = res + self.out_proj(ctx.reshape(B, S, self.heads * d))
+        res = x
+        y = self.pre_mlp_norm(x)
+        y = self.fc2(K.swiglu(self.fc1(y)))
+        return res + y
+
+
+class LlamaForTraining(nn.Module):
+    def __init__(self, cfg: ModelConfig, seq_len: int, dtype=torch.bfloat16,
+                 rope_base=500000.0, device=None):
+        super().__init__()
+        self.cfg = cfg
+        self.seq_len = seq_len
+        self.embedding = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
+                                      dtype=dtype, device=device)
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(cfg, dtype, device) for _ in range(cfg.layer_num)])
+        self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False,
+                                 dtype=dtype, device=device)
+        cs = K.build_rope_cache(seq_len, cfg.head_size, base=rope_base,
+                                device=device or "cpu")
+        self.register_buffer("rope_cs", cs, persistent=False)
+
+    def forward(self, tokens, labels):
+        # tokens/labels: [B, S] int64
+        B, S = tokens.shape
+        pos = (torch.arange(S, device=tokens.device, dtype=torch.int32)
+               .repeat(B))
+        x = self.embedding(tokens)
+        for layer in self.layers:
+            x = layer(x, self.rope_cs, pos)
+        x = self.final_norm(x)
+        logits = self.lm_head(x)
+        loss = K.fused_cross_entropy(
+            logits.reshape(B * S, -1), labels.reshape(-1))
+        return loss.mean()
+
+    def num_params(self):
+        return sum(p.numel() for p in self.parameters())
