@@ -1,0 +1,82 @@
+"""Fold measured calibration tables (gpurun_out/calib/*.json) into the
+MI355X system config (reference parity: combine_efficiency.py +
+run_one_click_benchmark.py write-back)."""
+
+import json
+import os
+import statistics
+import sys
+
+REPO = os.path.normpath(os.path.join(os.path.dirname(__file__), "..", ".."))
+CALIB = os.path.join(REPO, "gpurun_out", "calib")
+SYSTEM = os.path.join(REPO, "configs", "system", "mi355x.json")
+
+
+def _load(name):
+    p = os.path.join(CALIB, name)
+    if os.path.exists(p):
+        with open(p) as f:
+            return json.load(f)
+    return {}
+
+
+def main():
+    with open(SYSTEM) as f:
+        sysc = json.load(f)
+    acc = sysc["accelerator"]
+
+    matmul = _load("matmul.json")
+    if matmul:
+        acc["op"]["matmul"]["accurate_efficient_factor"] = matmul
+        acc["op"]["matmul"]["efficient_factor"] = round(
+            statistics.median(matmul.values()), 4)
+        print(f"matmul: {len(matmul)} shapes, median eff "
+              f"{acc['op']['matmul']['efficient_factor']}")
+    for key in ("sdp_fwd", "sdp_bwd"):
+        tab = _load(f"{key}.json")
+        if tab:
+            acc["op"][key]["accurate_efficient_factor"] = tab
+            acc["op"][key]["efficient_factor"] = round(
+                statistics.median(tab.values()), 4)
+            print(f"{key}: {len(tab)} shapes, median eff "
+                  f"{acc['op'][key]['efficient_factor']}")
+    group = _load("group_matmul.json")
+    if group:
+        acc["op"]["group_matmul"]["accurate_efficient_factor"] = group
+        acc["op"]["group_matmul"]["efficient_factor"] = round(
+            statistics.median(group.values()), 4)
+        print(f"group_matmul: {len(group)} shapes")
+
+    bw = _load("bandwidth.json")
+    if bw:
+        if "default_eff" in bw:
+            acc["bandwidth"]["default"]["efficient_factor"] = round(
+                bw["default_eff"], 4)
+        if "launch_us" in bw:
+            for k in acc["bandwidth"]:
+                acc["bandwidth"][k]["latency_us"] = round(bw["launch_us"], 2)
+        if "ce_fusion_fwd_eff" in bw:
+            acc["bandwidth"]["ce_fusion"]["efficient_factor"] = round(
+                min(bw["ce_fusion_fwd_eff"], bw.get("ce_fusion_bwd_eff", 1) / 1.0),
+                4)
+            # unfused ce not implemented separately on MI355X: same kernel
+            acc["bandwidth"]["ce"]["efficient_factor"] = acc["bandwidth"][
+                "ce_fusion"]["efficient_factor"]
+        if "optimizer_eff" in bw:
+            acc["bandwidth"]["optimizer"] = {
+                "gbps": 8000.0,
+                "efficient_factor": round(bw["optimizer_eff"], 4),
+                "latency_us": round(bw.get("launch_us", 4.0), 2),
+            }
+        for k in ("rmsnorm_fwd_eff", "rmsnorm_bwd_eff", "swiglu_fwd_eff",
+                  "stream_gbps", "optimizer_gbps"):
+            if k in bw:
+                print(f"  {k}: {bw[k]:.4g}")
+
+    with open(SYSTEM, "w") as f:
+        json.dump(sysc, f, indent=1)
+    print(f"wrote {SYSTEM}")
+
+
+if __name__ == "__main__":
+    main()

@@ -383,7 +383,7 @@ class PerfLLM(PerfBase):
         # traffic: grads fp32 r (norm) + r (adam); master r+w; m r+w; v r+w;
         # bf16 param w; zero_grad w
         bytes_traffic = numel * (4 * 2 + 4 * 2 + 4 * 2 + 4 * 2 + 2 + 4)
-        return self.system.compute_mem_access_time("default", bytes_traffic)
+        return self.system.compute_mem_access_time("optimizer", bytes_traffic)
 
     # ---- cost ------------------------------------------------------------
     def analysis_cost(self) -> Result:

@@ -59,6 +59,10 @@ class MixedPrecisionAdam:
 
     @torch.no_grad()
     def step(self):
+        """In-place Adam: no fp32 temporaries beyond one per-param denom
+        (Megatron's fused optimizer behaves the same; the foreach_div path
+        would transiently allocate 2x the full fp32 state = +64 GiB on an
+        8B model)."""
         self.t += 1
         b1, b2 = self.cfg.adam_betas
         grads = [p.main_grad for p in self.params]
@@ -73,12 +77,11 @@ class MixedPrecisionAdam:
         torch._foreach_addcmul_(self.v, grads, grads, value=1 - b2)
         bc1 = 1 - b1 ** self.t
         bc2 = 1 - b2 ** self.t
-        denom = torch._foreach_sqrt(self.v)
-        torch._foreach_div_(denom, bc2 ** 0.5)
-        torch._foreach_add_(denom, self.cfg.adam_eps)
-        upd = torch._foreach_div(self.m, denom)
-        torch._foreach_add_(self.masters, upd, alpha=-self.cfg.lr / bc1)
-        for p, w in zip(self.params, self.masters):
+        step_size = self.cfg.lr / bc1
+        inv_sqrt_bc2 = 1.0 / (bc2 ** 0.5)
+        for w, m, v, p in zip(self.masters, self.m, self.v, self.params):
+            denom = v.sqrt().mul_(inv_sqrt_bc2).add_(self.cfg.adam_eps)
+            w.addcdiv_(m, denom, value=-step_size)
             p.data.copy_(w)
 
 
