@@ -87,13 +87,11 @@ def time_gemm(desc, device="cuda"):
         dout = torch.randn(k, m, device=device, dtype=dt)
         x = torch.randn(k, n, device=device, dtype=dt)
         if accumulate and out_dtype == "fp32":
+            # trainer path: fused GemmEx into fp32 main_grad (kernels/ops.py)
+            from simumax_amd.kernels.ops import ext
+            E = ext()
             main_grad = torch.zeros(m, n, device=device, dtype=torch.float32)
-
-            def fn():
-                g = torch.matmul(dout.t(), x)
-                main_grad.add_(g.float())
-        else:
-            fn = lambda: torch.matmul(dout.t(), x)
+            fn = lambda: E.wgrad_accum(dout.t().contiguous(), x, main_grad)
     else:
         raise ValueError(layout)
     iters = 10 if b * m * k * n < 2**40 else 4

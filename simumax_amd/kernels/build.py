@@ -30,6 +30,7 @@ def build(verbose: bool = True, force: bool = False) -> str:
     sources = [os.path.join(CSRC, s) for s in KERNEL_SOURCES
                if os.path.exists(os.path.join(CSRC, s))]
     binding = os.path.join(CSRC, "binding.cpp")
+    sources.append(os.path.join(CSRC, "wgrad.cpp"))
     all_src = sources + [binding, os.path.join(CSRC, "common.h")]
     if (not force and os.path.exists(OUT)
             and os.path.getmtime(OUT) > _newest_mtime(all_src)):
@@ -67,7 +68,7 @@ def build(verbose: bool = True, force: bool = False) -> str:
 
     link = ["hipcc", "-shared", "-o", OUT] + objs + [
         f"-L{torch_lib}", "-ltorch", "-ltorch_cpu", "-ltorch_hip", "-lc10",
-        "-lc10_hip", "-ltorch_python", f"-Wl,-rpath,{torch_lib}",
+        "-lc10_hip", "-ltorch_python", "-lhipblas", f"-Wl,-rpath,{torch_lib}",
     ]
     if verbose:
         print("[simumax_hip] linking", OUT)

@@ -187,7 +187,11 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
     return {dq.to(torch::kBFloat16), dk.contiguous(), dv.contiguous()};
 }
 
+void wgrad_accum(torch::Tensor dout, torch::Tensor x, torch::Tensor main_grad);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("wgrad_accum", &wgrad_accum,
+          "main_grad(fp32) += dout^T @ x (bf16 in, hipBLAS GemmEx)");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 fragment-layout probe");
     m.def("fa_fwd", &fa_fwd, "flash attention forward (gfx950 MFMA)");
     m.def("fa_bwd", &fa_bwd, "flash attention backward (gfx950 MFMA)");

@@ -202,6 +202,58 @@ def fused_cross_entropy(logits, labels):
 
 
 # --------------------------------------------------------------------------
+# Linear with fused fp32 wgrad accumulation (hipBLAS GemmEx, TE-style)
+# --------------------------------------------------------------------------
+_DUMMY_WGRADS = {}
+
+
+def _dummy_wgrad(shape, device, dtype):
+    """One shared placeholder grad per shape (TE get_dummy_wgrad analog —
+    the reference memory model tracks these as te_dummy_wgrad_shapes)."""
+    key = (tuple(shape), str(device), dtype)
+    if key not in _DUMMY_WGRADS:
+        _DUMMY_WGRADS[key] = torch.empty(shape, device=device, dtype=dtype)
+    return _DUMMY_WGRADS[key]
+
+
+class _FusedLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight):
+        ctx.save_for_backward(x, weight)
+        return torch.matmul(x, weight.t())
+
+    @staticmethod
+    def backward(ctx, dout):
+        x, weight = ctx.saved_tensors
+        dx = torch.matmul(dout, weight)
+        if x.is_cuda and hasattr(weight, "main_grad"):
+            d2 = dout.reshape(-1, dout.shape[-1]).contiguous()
+            x2 = x.reshape(-1, x.shape[-1]).contiguous()
+            ext().wgrad_accum(d2, x2, weight.main_grad)
+            dw = _dummy_wgrad(weight.shape, weight.device, weight.dtype)
+        else:
+            dw = torch.matmul(dout.reshape(-1, dout.shape[-1]).t(),
+                              x.reshape(-1, x.shape[-1]))
+        return dx, dw
+
+
+class FusedLinear(torch.nn.Module):
+    """nn.Linear(bias=False) whose wgrad accumulates straight into the fp32
+    main_grad buffer (single GemmEx, no convert-and-add pass)."""
+
+    def __init__(self, in_features, out_features, dtype=torch.bfloat16,
+                 device=None):
+        super().__init__()
+        w = torch.empty(out_features, in_features, dtype=dtype, device=device)
+        torch.nn.init.normal_(w, std=0.02)
+        self.weight = torch.nn.Parameter(w)
+        self.weight._fused_wgrad = True
+
+    def forward(self, x):
+        return _FusedLinearFn.apply(x, self.weight)
+
+
+# --------------------------------------------------------------------------
 # Flash attention (gfx950 HIP kernel; see csrc/attention.hip)
 # --------------------------------------------------------------------------
 class _FlashAttnFn(torch.autograd.Function):

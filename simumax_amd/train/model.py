@@ -14,7 +14,7 @@ from __future__ import annotations
 import math
 
 import torch
-import torch.nn as nn
+import torch.nn as nn  # noqa: F401 (Embedding)
 
 from ..core.config import ModelConfig
 from ..kernels import ops as K
@@ -29,13 +29,13 @@ class LlamaDecoderLayer(nn.Module):
         self.head_size = cfg.head_size
         qkv_out = (cfg.head_num + 2 * cfg.kv_head_num) * cfg.head_size
         self.input_norm = K.RMSNorm(h, dtype=dtype, device=device)
-        self.qkv_proj = nn.Linear(h, qkv_out, bias=False, dtype=dtype, device=device)
-        self.out_proj = nn.Linear(cfg.head_num * cfg.head_size, h, bias=False,
-                                  dtype=dtype, device=device)
+        self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype, device=device)
+        self.out_proj = K.FusedLinear(cfg.head_num * cfg.head_size, h,
+                                      dtype=dtype, device=device)
         self.pre_mlp_norm = K.RMSNorm(h, dtype=dtype, device=device)
         assert cfg.use_swiglu
-        self.fc1 = nn.Linear(h, 2 * cfg.intermediate_size, bias=False, dtype=dtype, device=device)
-        self.fc2 = nn.Linear(cfg.intermediate_size, h, bias=False, dtype=dtype, device=device)
+        self.fc1 = K.FusedLinear(h, 2 * cfg.intermediate_size, dtype=dtype, device=device)
+        self.fc2 = K.FusedLinear(cfg.intermediate_size, h, dtype=dtype, device=device)
 
     def forward(self, x, rope_cs, pos):
         # x: [B, S, H]
@@ -70,8 +70,8 @@ class LlamaForTraining(nn.Module):
         self.layers = nn.ModuleList(
             [LlamaDecoderLayer(cfg, dtype, device) for _ in range(cfg.layer_num)])
         self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
-        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False,
-                                 dtype=dtype, device=device)
+        self.lm_head = K.FusedLinear(cfg.hidden_size, cfg.vocab_size,
+                                     dtype=dtype, device=device)
         cs = K.build_rope_cache(seq_len, cfg.head_size, base=rope_base,
                                 device=device or "cpu")
         self.register_buffer("rope_cs", cs, persistent=False)

@@ -118,7 +118,8 @@ class DataParallelGradReducer:
 
     def _make_hook(self, bi, bucket, remaining):
         def hook(p):
-            p.main_grad.add_(p.grad.float())
+            if not getattr(p, "_fused_wgrad", False) or not p.is_cuda:
+                p.main_grad.add_(p.grad.float())
             p.grad = None
             remaining.discard(id(p))
             if not remaining:
@@ -147,7 +148,8 @@ class DataParallelGradReducer:
 def accumulate_main_grads(params):
     for p in params:
         if p.grad is not None:
-            p.main_grad.add_(p.grad.float())
+            if not getattr(p, "_fused_wgrad", False) or not p.is_cuda:
+                p.main_grad.add_(p.grad.float())
             p.grad = None
 
 

@@ -82,9 +82,11 @@ def test_net_zero_for_single_rank(mi355x_system):
 
 def test_mem_access_time(mi355x_system):
     one_gib = 1 << 30
+    cfg = mi355x_system.accelerator.bandwidth["default"]
     t = mi355x_system.compute_mem_access_time("default", one_gib)
-    bw = 8000 * 0.74  # GiB/s approx
-    assert t == pytest.approx(one_gib / (8000 * 1024**3 * 0.74) * 1e3 + 0.004)
+    assert t == pytest.approx(
+        one_gib / (cfg.gbps * 1024**3 * cfg.efficient_factor) * 1e3
+        + cfg.latency_us / 1e3)
 
 
 def test_stage_layers_uneven():
