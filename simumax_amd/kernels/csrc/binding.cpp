@@ -135,8 +135,8 @@ void mfma_probe_launch(const void *, const void *, void *, void *, int,
 void fa_fwd_launch(const void *, const void *, const void *, void *, void *,
                    int, int, int, int, int, hipStream_t);
 void fa_bwd_launch(const void *, const void *, const void *, const void *,
-                   const void *, const void *, void *, void *, int, int, int,
-                   int, int, hipStream_t);
+                   const void *, const void *, void *, void *, void *, int,
+                   int, int, int, int, hipStream_t);
 }
 
 std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B,
@@ -177,14 +177,15 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
     const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
     const int Hkv = k.size(2);
     auto f32 = q.options().dtype(torch::kFloat32);
-    auto dq = torch::zeros({B, S, Hq, D}, f32);
+    auto dq = torch::empty({B, S, Hq, D}, q.options());
     auto dkv = torch::zeros({B, S, Hkv, 2, D}, f32);
+    auto dsum = torch::empty({B, Hq, S}, f32);
     fa_bwd_launch(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                   o.data_ptr(), lse.data_ptr(), dq.data_ptr(), dkv.data_ptr(),
-                  B, S, Hq, Hkv, causal ? 1 : 0, cur_stream());
+                  dsum.data_ptr(), B, S, Hq, Hkv, causal ? 1 : 0, cur_stream());
     auto dk = dkv.select(3, 0).to(torch::kBFloat16);
     auto dv = dkv.select(3, 1).to(torch::kBFloat16);
-    return {dq.to(torch::kBFloat16), dk.contiguous(), dv.contiguous()};
+    return {dq, dk.contiguous(), dv.contiguous()};
 }
 
 void wgrad_accum(torch::Tensor dout, torch::Tensor x, torch::Tensor main_grad);
