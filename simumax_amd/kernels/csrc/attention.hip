@@ -304,9 +304,9 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
     extern __shared__ __attribute__((aligned(16))) char smem[];
     bf16raw *Q_lds = reinterpret_cast<bf16raw *>(smem);      // [32][QS2]
     bf16raw *dO_lds = Q_lds + BW_QT * QS2;                   // [32][QS2]
-    bf16raw *Qt_lds = dO_lds + BW_QT * QS2;                  // [128][TS2]
-    bf16raw *dOt_lds = Qt_lds + DHEAD * TS2;                 // [128][TS2]
-    float *lse_lds = reinterpret_cast<float *>(dOt_lds + DHEAD * TS2);
+    bf16raw *Q_img = dO_lds + BW_QT * QS2;                   // tr image 8*VSUB
+    bf16raw *dO_img = Q_img + 8 * VSUB;                      // tr image 8*VSUB
+    float *lse_lds = reinterpret_cast<float *>(dO_img + 8 * VSUB);
     float *D_lds = lse_lds + BW_QT;
     bf16raw *wbase = reinterpret_cast<bf16raw *>(D_lds + BW_QT);
     const int wave = threadIdx.x / WAVE;
@@ -369,11 +369,10 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
                 store8(Q_lds + r * QS2 + d0, qq);
                 bf16x8 dd = load8(dop + (long)src * q_row + d0);
                 store8(dO_lds + r * QS2 + d0, dd);
-#pragma unroll
-                for (int j = 0; j < 8; ++j) {
-                    Qt_lds[(d0 + j) * TS2 + r] = f2bf(qq.get(j));
-                    dOt_lds[(d0 + j) * TS2 + r] = f2bf(dd.get(j));
-                }
+                // tr images (conflict-free store8; read via ds_read_b64_tr_b16)
+                const int io = (d0 >> 4) * VSUB + v_img_row(r) * 16 + (d0 & 15);
+                store8(Q_img + io, qq);
+                store8(dO_img + io, dd);
             }
             for (int r = tid; r < BW_QT; r += FA_BLOCK) {
                 const int src = min(qt + r, S - 1);
@@ -415,10 +414,10 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
             bf16x8v a_dst = ld_frag(dSt_l + col * TS2 + kgrp * 8);
 #pragma unroll
             for (int dt = 0; dt < 8; ++dt) {
-                bf16x8v b_do = ld_frag(dOt_lds + (dt * 16 + col) * TS2 + kgrp * 8);
+                bf16x8v b_do = tr_frag(dO_img + dt * VSUB, lane);
                 dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     a_pt, b_do, dv_acc[dt], 0, 0, 0);
-                bf16x8v b_q = ld_frag(Qt_lds + (dt * 16 + col) * TS2 + kgrp * 8);
+                bf16x8v b_q = tr_frag(Q_img + dt * VSUB, lane);
                 dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     a_dst, b_q, dk_acc[dt], 0, 0, 0);
             }
@@ -639,7 +638,7 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
     // 2. dK/dV (kv-parallel)
     {
         dim3 grid(CDIV(S, WAVES * BKV), Hq, B);
-        size_t q_side = (2 * BW_QT * QS2 + 2 * DHEAD * TS2) * sizeof(bf16raw) +
+        size_t q_side = (2 * BW_QT * QS2 + 2 * 8 * VSUB) * sizeof(bf16raw) +
                         2 * BW_QT * sizeof(float);
         size_t per_wave = (BKV * KS + 2 * BKV * TS2) * sizeof(bf16raw);
         size_t smem = q_side + WAVES * per_wave;
