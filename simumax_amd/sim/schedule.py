@@ -1,0 +1,189 @@
+"""PpSchedule: build per-rank simulator job lists from the analytic chunks.
+
+Parity target: simumax/core/transformer/pipeline_schedule.py:30-959
+(PpSchedule.prefill_batch 1F1B, OptimizerSimulator) — jobs are derived
+from the SAME leaf modules/CommEvents the analytic coster priced, so
+perf() and simulate() agree by construction. Sync p2p semantics (the
+async batched-bundle VPP path is a later extension).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List
+
+from ..core.utils import get_pp_p2p_comm_size, get_pp_stage_representative_rank
+from .events import Job, MemDelta
+
+
+def _leaf_jobs_fwd(chunk, stage, mb, p2p=None):
+    jobs = []
+    segments = chunk._segments(chunk.leaf_modules())
+    for seg in segments:
+        for leaf in seg["leaves"]:
+            ci = leaf.get_cost_info()
+            ai = leaf.get_act_info()
+            cache = (leaf.input_info.total_bytes()
+                     if seg["recompute"] and leaf is seg["leaves"][0]
+                     else (0.0 if seg["recompute"] else ai.activation_mem_cache))
+            jobs.append(Job(
+                name=leaf.full_name, kind="fwd", dur=ci.fwd_compute_time,
+                mb=mb,
+                mem=MemDelta(alloc_bytes=cache,
+                             transient_bytes=ai.fwd_peak_mem_no_cache,
+                             token_key=f"mb{mb}.{leaf.full_name}"),
+                call_stack=leaf.full_name,
+            ))
+            for ev in leaf.comm_ops:
+                if ev.stage != "fwd":
+                    continue
+                jobs.append(Job(
+                    name=f"{leaf.full_name}.{ev.op_name}", kind="comm",
+                    dur=ev.time_ms, lane="comm", mb=mb,
+                    gid=f"mb{mb}-{leaf.full_name}-{ev.op_name}-{ev.comm_stage}",
+                    call_stack=leaf.full_name,
+                ))
+    return jobs
+
+
+def _leaf_jobs_bwd(chunk, stage, mb):
+    jobs = []
+    segments = chunk._segments(chunk.leaf_modules())
+    for seg in reversed(segments):
+        if seg["recompute"]:
+            # re-forward the segment (RecomputeBlockJob analog)
+            for leaf in seg["leaves"]:
+                ci = leaf.get_cost_info()
+                ai = leaf.get_act_info()
+                jobs.append(Job(
+                    name=f"{leaf.full_name}(recompute)", kind="recompute",
+                    dur=ci.recompute_compute_time, mb=mb,
+                    mem=MemDelta(alloc_bytes=ai.activation_mem_cache,
+                                 token_key=f"mb{mb}.rc.{leaf.full_name}"),
+                    call_stack=leaf.full_name,
+                ))
+        for leaf in reversed(seg["leaves"]):
+            ci = leaf.get_cost_info()
+            ai = leaf.get_act_info()
+            if seg["recompute"]:
+                free = ai.activation_mem_cache
+                key = f"mb{mb}.rc.{leaf.full_name}"
+            else:
+                free = ai.activation_mem_cache
+                key = f"mb{mb}.{leaf.full_name}"
+            jobs.append(Job(
+                name=f"{leaf.full_name}(bwd)", kind="bwd",
+                dur=ci.bwd_grad_act_time + ci.bwd_grad_w_time, mb=mb,
+                mem=MemDelta(free_bytes=free,
+                             transient_bytes=ai.bwd_peak_mem_no_cache,
+                             token_key=key),
+                call_stack=leaf.full_name,
+            ))
+            if seg["recompute"] and leaf is seg["leaves"][0]:
+                # release the segment-input tensor held since forward
+                jobs.append(Job(
+                    name=f"{leaf.full_name}(free segment input)", kind="bwd",
+                    dur=0.0, mb=mb,
+                    mem=MemDelta(free_bytes=leaf.input_info.total_bytes(),
+                                 token_key=f"mb{mb}.{leaf.full_name}"),
+                    call_stack=leaf.full_name,
+                ))
+            for ev in leaf.comm_ops:
+                if ev.stage not in ("bwd_act", "bwd_w"):
+                    continue
+                jobs.append(Job(
+                    name=f"{leaf.full_name}.{ev.op_name}(bwd)", kind="comm",
+                    dur=ev.time_ms, lane="comm", mb=mb,
+                    gid=f"mb{mb}-{leaf.full_name}-{ev.op_name}-{ev.stage}",
+                    call_stack=leaf.full_name,
+                ))
+    return jobs
+
+
+class PpSchedule:
+    """Builds the per-(simulated-)rank 1F1B job lists."""
+
+    def __init__(self, perf_model, merge_lanes=True):
+        self.perf = perf_model
+        self.strategy = perf_model.strategy
+        self.system = perf_model.system
+        self.merge_lanes = merge_lanes
+
+    def sim_ranks(self) -> List[int]:
+        s = self.strategy
+        if self.merge_lanes:
+            return [get_pp_stage_representative_rank(i, s)
+                    for i in range(s.pp_size)]
+        return list(range(s.world_size))
+
+    def build(self) -> Dict[int, List[Job]]:
+        s = self.perf.strategy
+        pp, mbc = s.pp_size, s.micro_batch_num
+        p2p_size = get_pp_p2p_comm_size(s, self.perf.model_config)
+        p2p_time = 0.0
+        if pp > 1:
+            p2p_time = self.system.compute_net_op_time(
+                "p2p", p2p_size, 2, net=s.pp_net, comm_stage="pp",
+                strategy=s)
+
+        ranks = self.sim_ranks()
+        jobs: Dict[int, List[Job]] = {r: [] for r in ranks}
+        per_stage_rank = {i: get_pp_stage_representative_rank(i, s)
+                          for i in range(pp)}
+
+        # per-rank 1F1B streams (same order as the analytic recurrence)
+        for stage in range(pp):
+            rep = per_stage_rank[stage]
+            stage_ranks = [r for r in ranks
+                           if r // (s.world_size // pp) == stage] or [rep]
+            chunk = self.perf.chunks[stage]
+            warm = min(pp - stage - 1, mbc)
+            stream = [("F", m) for m in range(warm)]
+            nf, nb = warm, 0
+            while nb < mbc:
+                if nf < mbc:
+                    stream.append(("F", nf)); nf += 1
+                stream.append(("B", nb)); nb += 1
+            for r in stage_ranks:
+                for kind, m in stream:
+                    if kind == "F":
+                        if stage > 0:
+                            jobs[r].append(Job(
+                                name=f"recv_fwd.mb{m}", kind="p2p",
+                                dur=p2p_time, lane="comm", mb=m,
+                                gid=f"p2p-f-mb{m}-{stage-1}-{stage}",
+                                peers=(per_stage_rank[stage - 1], r)))
+                        jobs[r].extend(_leaf_jobs_fwd(chunk, stage, m))
+                        if stage < pp - 1:
+                            jobs[r].append(Job(
+                                name=f"send_fwd.mb{m}", kind="p2p",
+                                dur=p2p_time, lane="comm", mb=m,
+                                gid=f"p2p-f-mb{m}-{stage}-{stage+1}",
+                                peers=(r, per_stage_rank[stage + 1])))
+                    else:
+                        if stage < pp - 1:
+                            jobs[r].append(Job(
+                                name=f"recv_bwd.mb{m}", kind="p2p",
+                                dur=p2p_time, lane="comm", mb=m,
+                                gid=f"p2p-b-mb{m}-{stage+1}-{stage}",
+                                peers=(per_stage_rank[stage + 1], r)))
+                        jobs[r].extend(_leaf_jobs_bwd(chunk, stage, m))
+                        if stage > 0:
+                            jobs[r].append(Job(
+                                name=f"send_bwd.mb{m}", kind="p2p",
+                                dur=p2p_time, lane="comm", mb=m,
+                                gid=f"p2p-b-mb{m}-{stage}-{stage-1}",
+                                peers=(r, per_stage_rank[stage - 1])))
+        # optimizer tail (DP collectives + adam traffic)
+        for stage in range(pp):
+            for r in ([per_stage_rank[stage]] if self.merge_lanes else
+                      [x for x in ranks
+                       if x // (s.world_size // pp) == stage]):
+                dp_t = self.perf._compute_dp_time(stage)
+                if dp_t > 0:
+                    jobs[r].append(Job(name="dp_grad_sync", kind="comm",
+                                       dur=dp_t, lane="comm",
+                                       gid=f"dp-{stage}", mb=-1))
+                jobs[r].append(Job(name="optimizer.adam", kind="optim",
+                                   dur=self.perf._compute_optim_time(stage),
+                                   mb=-1))
+        return jobs

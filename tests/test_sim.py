@@ -1,0 +1,106 @@
+"""Event-simulator tests: determinism, perf-vs-sim agreement, artifacts,
+memory-token balance, deadlock diagnostics."""
+
+import json
+
+import pytest
+
+from simumax_amd import (ModelConfig, PerfLLM, StrategyConfig, SystemConfig,
+                         get_simu_model_config, get_simu_strategy_config,
+                         get_simu_system_config)
+from simumax_amd.sim.events import DeadlockError, Job, SimuSystem
+
+
+def build(strategy="tp1_pp2_dp4_mbs1", model="llama2-tiny", **over):
+    p = PerfLLM()
+    st = StrategyConfig.init_from_config_file(get_simu_strategy_config(strategy))
+    for k, v in over.items():
+        setattr(st, k, v)
+    p.configure(
+        st,
+        ModelConfig.init_from_config_file(get_simu_model_config(model)),
+        SystemConfig.init_from_config_file(get_simu_system_config("mi355x")),
+    )
+    p.run_estimate()
+    return p
+
+
+def test_simulate_artifacts(tmp_path):
+    p = build()
+    res = p.simulate(str(tmp_path), merge_lanes=True)
+    for f in ("tracing_logs.json", "simu_memory_result.json",
+              "simu_memory_snapshot.json", "simu_memory_viz_snapshot.pickle"):
+        assert (tmp_path / f).exists(), f
+    tr = json.loads((tmp_path / "tracing_logs.json").read_text())
+    assert tr["displayTimeUnit"] == "ms"
+    slices = [e for e in tr["traceEvents"] if e.get("ph") == "X"]
+    assert len(slices) > 100
+    assert {e["pid"] for e in slices} == {"rank0", "rank4"}
+    assert any(e["cat"] == "comm" for e in slices)
+    snap = json.loads((tmp_path / "simu_memory_snapshot.json").read_text())
+    assert snap["schema"].startswith("simumax_amd.memory_snapshot")
+    allocs = [t for t in snap["cache_tokens"] if t["action"] == "alloc"]
+    frees = [t for t in snap["cache_tokens"] if t["action"] == "free"]
+    assert len(allocs) == len(frees)  # every cache token returned
+
+
+def test_sim_vs_perf_agreement(tmp_path):
+    """Simulated end-to-end within a few % of the analytic estimate
+    (reference's own validation strategy: docs/release_v1.2.md:33-35)."""
+    p = build()
+    cost = p.analysis_cost()
+    res = p.simulate(str(tmp_path))
+    analytic = (cost["pipeline_time"] / cost["straggler_ratio"]
+                + cost["dp_time"] + cost["optim_time"])
+    assert res["total_time"] == pytest.approx(analytic, rel=0.10)
+
+
+def test_sim_memory_matches_analytic(tmp_path):
+    p = build()
+    res = p.simulate(str(tmp_path))
+    mem = p.analysis_mem()
+    for stage, st in enumerate(mem["stages_raw"]):
+        rank = stage * (p.strategy.world_size // p.strategy.pp_size)
+        analytic_raw = st["peak_mem"] * p.strategy.mem_factor
+        assert res["peak_mem"][rank] == pytest.approx(analytic_raw, rel=0.02)
+
+
+def test_sim_determinism(tmp_path):
+    p1 = build()
+    r1 = p1.simulate(str(tmp_path / "a"))
+    p2 = build()
+    r2 = p2.simulate(str(tmp_path / "b"))
+    assert r1["total_time"] == r2["total_time"]
+
+
+def test_sim_with_recompute(tmp_path):
+    p = build(enable_recompute=True, recompute_granularity="full_block")
+    res = p.simulate(str(tmp_path))
+    tr = json.loads((tmp_path / "tracing_logs.json").read_text())
+    assert any("recompute" in e.get("name", "") for e in tr["traceEvents"])
+
+
+def test_deadlock_diagnostics():
+    # two ranks each waiting on a different rendezvous: must raise with
+    # per-rank blocked heads in the message
+    jobs = {
+        0: [Job(name="a", kind="comm", dur=1, lane="comm", gid="g1",
+                peers=(0, 1))],
+        1: [Job(name="b", kind="comm", dur=1, lane="comm", gid="g2",
+                peers=(0, 1))],
+    }
+    with pytest.raises(DeadlockError) as exc:
+        SimuSystem(jobs).run()
+    assert "blocked heads" in str(exc.value)
+
+
+def test_export_analytic_schedule_trace(tmp_path):
+    from simumax_amd.sim.trace import export_pipeline_schedule_trace
+
+    p = build()
+    cost = p.analysis_cost()
+    assert p.schedule_records is not None
+    out = export_pipeline_schedule_trace(
+        p.schedule_records, cost["chunk_fwd_times"], cost["chunk_bwd_times"],
+        str(tmp_path / "sched.json"))
+    assert len(out["traceEvents"]) == 2 * p.strategy.pp_size * p.strategy.micro_batch_num
