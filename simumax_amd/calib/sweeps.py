@@ -216,7 +216,10 @@ def sweep_bandwidth(path):
     t = _timeit(lambda: E.ce_bwd(logits, labels, dl, rm, rs), iters=5)
     out["ce_fusion_bwd_eff"] = 2 * lb / (t / 1e3) / (HBM_PEAK_GBPS * GiB)
 
-    # optimizer: in-place Adam traffic model = 38 B/param
+    # optimizer: the trainer's exact step sequence; traffic model shared
+    # via core.consts.OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+    from simumax_amd.core.consts import OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+
     numel = 2 * GiB // 4
     master = torch.zeros(numel, device="cuda", dtype=torch.float32)
     m_ = torch.zeros_like(master)
@@ -225,15 +228,18 @@ def sweep_bandwidth(path):
     pb = torch.zeros(numel, device="cuda", dtype=torch.bfloat16)
 
     def adam():
-        m_.mul_(0.9).add_(g_, alpha=0.1)
-        v_.mul_(0.95)
-        v_.addcmul_(g_, g_, value=0.05)
+        g_.zero_()                       # zero_grad
+        _ = g_.norm(2)                   # grad-norm clip pass
+        torch._foreach_mul_([m_], 0.9)
+        torch._foreach_add_([m_], [g_], alpha=0.1)
+        torch._foreach_mul_([v_], 0.95)
+        torch._foreach_addcmul_([v_], [g_], [g_], value=0.05)
         denom = v_.sqrt().add_(1e-8)
         master.addcdiv_(m_, denom, value=-1e-4)
         pb.copy_(master)
 
     t = _timeit(adam, iters=5)
-    traffic = numel * 38
+    traffic = numel * OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
     out["optimizer_eff"] = traffic / (t / 1e3) / (HBM_PEAK_GBPS * GiB)
     out["optimizer_gbps"] = traffic / (t / 1e3) / GiB
 

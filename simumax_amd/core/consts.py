@@ -50,3 +50,11 @@ MI355X = {
     "xgmi_link_gbps": 153.0,       # per link, per direction
     "lds_kib_per_cu": 160,
 }
+
+# Mixed-precision Adam HBM traffic per parameter (bytes), matching the
+# trainer's exact step sequence (train/trainer.py MixedPrecisionAdam):
+# zero_grad(w4) + grad-norm(r4) + m update(r4 rw8) + v update(r4 rw8)
+# + denom sqrt/add (r4 w4 rw8) + addcdiv (r4 r4 rw8) + bf16 copy (r4 w2).
+# The calibration sweep times the same sequence, so the efficiency factor
+# is consistent by construction.
+OPTIMIZER_TRAFFIC_BYTES_PER_PARAM = 70

@@ -391,8 +391,25 @@ class CoreAttention(MetaModule):
 
     def _leaf_act_info(self, info):
         q, k, v, o, lse = self._sdp_bytes()
+        if self.cp > 1 and self.cp_a2a:
+            # Ulysses a2a transient buffers (reference parity:
+            # dense_module.py:1259-1338): async_cp posts q/k/v a2a together
+            # (input + output buffers all live), sync_cp moves one at a time
+            if self.strategy.cp_a2a_mode == "async_cp":
+                info.fwd_peak_mem_no_cache = q + k + v  # post-a2a copies
+                info.bwd_peak_mem_no_cache = q + k + v + o
+            else:
+                info.fwd_peak_mem_no_cache = 2 * max(q, k, v)
+                info.bwd_peak_mem_no_cache = 2 * max(q, k, v)
+            if self.strategy.te_cp_a2a_saves_pre_posta2a_output:
+                # bwd re-uses the saved pre-PostA2A O: only dO moves back
+                info.bwd_peak_mem_no_cache = max(
+                    0.0, info.bwd_peak_mem_no_cache - o)
         if self.use_flash:
             info.activation_mem_cache = q + k + v + lse
+            if (self.cp > 1 and self.cp_a2a
+                    and self.strategy.te_cp_a2a_saves_pre_posta2a_output):
+                info.activation_mem_cache += o
         else:
             b, _ = self._bsd
             s = self.full_seq

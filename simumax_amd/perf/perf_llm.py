@@ -380,9 +380,9 @@ class PerfLLM(PerfBase):
         numel = (mi.dense_weight_bytes / 2) / max(1, _zero_div_dense(s)) + (
             mi.moe_weight_bytes / 2
         ) / max(1, _zero_div_moe(s))
-        # traffic: grads fp32 r (norm) + r (adam); master r+w; m r+w; v r+w;
-        # bf16 param w; zero_grad w
-        bytes_traffic = numel * (4 * 2 + 4 * 2 + 4 * 2 + 4 * 2 + 2 + 4)
+        from ..core.consts import OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+
+        bytes_traffic = numel * OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
         return self.system.compute_mem_access_time("optimizer", bytes_traffic)
 
     # ---- cost ------------------------------------------------------------

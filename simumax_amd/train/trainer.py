@@ -77,10 +77,13 @@ class MixedPrecisionAdam:
         torch._foreach_addcmul_(self.v, grads, grads, value=1 - b2)
         bc1 = 1 - b1 ** self.t
         bc2 = 1 - b2 ** self.t
-        step_size = self.cfg.lr / bc1
-        inv_sqrt_bc2 = 1.0 / (bc2 ** 0.5)
+        # fold bias corrections: m/(sqrt(v)/sqrt(bc2)+eps)/bc1
+        #   = sqrt(bc2)/bc1 * m/(sqrt(v)+eps*sqrt(bc2))  (exact)
+        sqrt_bc2 = bc2 ** 0.5
+        step_size = self.cfg.lr * sqrt_bc2 / bc1
+        eps2 = self.cfg.adam_eps * sqrt_bc2
         for w, m, v, p in zip(self.masters, self.m, self.v, self.params):
-            denom = v.sqrt().mul_(inv_sqrt_bc2).add_(self.cfg.adam_eps)
+            denom = v.sqrt().add_(eps2)
             w.addcdiv_(m, denom, value=-step_size)
             p.data.copy_(w)
 

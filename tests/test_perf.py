@@ -139,3 +139,40 @@ def test_analysis_writes_artifacts(tmp_path):
     for f in ("compute_result.json", "mem_result.json", "base_info.json",
               "net_info.json", "model_arch"):
         assert (tmp_path / "out" / f).exists()
+
+
+def test_cp_a2a_long_context():
+    """CP a2a (Ulysses): full-seq SDP with heads/(tp*cp), 8 a2a per layer."""
+    p = build(strategy="tp2_pp1_dp4_mbs1", seq_len=32768, cp_size=4,
+              tp_size=2, micro_batch_num=2)
+    chunk = p.chunks[0]
+    sdp = [l for l in chunk.leaf_modules()
+           if type(l).__name__ == "CoreAttention"][0]
+    assert "seq_len=32768" in sdp.get_input_shapes_desc("fwd")
+    assert sdp.sdp_head_num == 32 // 2 // 4
+    evs = [e for e in sdp.comm_ops if e.comm_stage == "cp"]
+    assert len(evs) == 8  # q,k,v,o fwd + mirrored bwd
+    assert all(e.op_name == "all2all" for e in evs)
+    cost = p.analysis_cost()
+    mem = p.analysis_mem()
+    assert cost["mfu"] > 0.02 and not mem["oom"]
+
+
+def test_cp_sync_mode_lower_peak():
+    kw = dict(strategy="tp2_pp1_dp4_mbs1", seq_len=32768, cp_size=4,
+              tp_size=2, micro_batch_num=2)
+    p_async = build(cp_a2a_mode="async_cp", **kw)
+    p_sync = build(cp_a2a_mode="sync_cp", **kw)
+    a = p_async.analysis_mem()["max_peak_mem"]
+    s = p_sync.analysis_mem()["max_peak_mem"]
+    assert s <= a
+
+
+def test_cp_all_gather_mode():
+    p = build(strategy="tp2_pp1_dp4_mbs1", seq_len=32768, cp_size=4,
+              tp_size=2, micro_batch_num=2, cp_comm_type="all_gather")
+    sdp = [l for l in p.chunks[0].leaf_modules()
+           if type(l).__name__ == "CoreAttention"][0]
+    kinds = [(e.stage, e.op_name) for e in sdp.comm_ops]
+    assert ("fwd", "all_gather") in kinds
+    assert ("bwd_act", "reduce_scatter") in kinds
