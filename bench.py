@@ -64,8 +64,7 @@ def predict(model_cfg, world, args):
     sysc = SystemConfig.init_from_config_file(get_simu_system_config("mi355x"))
     import copy
 
-    mc = copy.deepcopy(model_cfg)
-    mc.padded_vocab_size = False  # trainer uses raw vocab
+    mc = copy.deepcopy(model_cfg)  # trainer pads the vocab the same way
     p.configure(st, mc, sysc)
     p.run_estimate()
     cost = p.analysis_cost()

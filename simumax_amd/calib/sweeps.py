@@ -91,7 +91,8 @@ def time_gemm(desc, device="cuda"):
             from simumax_amd.kernels.ops import ext
             E = ext()
             main_grad = torch.zeros(m, n, device=device, dtype=torch.float32)
-            fn = lambda: E.wgrad_accum(dout.t().contiguous(), x, main_grad)
+            # dout [tokens=k, out=m], x [tokens=k, in=n] -> main_grad [m, n]
+            fn = lambda: E.wgrad_accum(dout, x, main_grad)
     else:
         raise ValueError(layout)
     iters = 10 if b * m * k * n < 2**40 else 4

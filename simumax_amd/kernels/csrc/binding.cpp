@@ -105,6 +105,7 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor logits, torch::Tensor labels) {
     CHECK_IN(labels);
     TORCH_CHECK(labels.scalar_type() == torch::kInt64);
     const int V = logits.size(-1);
+    TORCH_CHECK(V % 8 == 0, "fused CE requires vocab % 8 == 0 (pad the vocab)");
     const long rows = logits.numel() / V;
     auto opts = logits.options().dtype(torch::kFloat32);
     auto loss = torch::empty({rows}, opts);

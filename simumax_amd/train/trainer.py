@@ -156,8 +156,12 @@ def accumulate_main_grads(params):
             p.grad = None
 
 
-def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda"):
+def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
+                  tp_size: int = 1):
     torch.manual_seed(1234)
+    # Megatron-style vocab padding (keeps CE vocab a GPU-friendly multiple
+    # and makes GEMM shape keys match the calibration tables)
+    model_cfg.maybe_pad_vocab_size(tp_size)
     model = LlamaForTraining(model_cfg, cfg.seq_len, device=device)
     opt = MixedPrecisionAdam(model.parameters(), cfg)
     reducer = DataParallelGradReducer(list(model.parameters()),
