@@ -91,8 +91,15 @@ class MixedPrecisionAdam:
         # fold bias corrections: m/(sqrt(v)/sqrt(bc2)+eps)/bc1
         #   = sqrt(bc2)/bc1 * m/(sqrt(v)+eps*sqrt(bc2))  (exact)
         sqrt_bc2 = bc2 ** 0.5
-        denom = self.v.sqrt().add_(self.cfg.adam_eps * sqrt_bc2)
-        self.master.addcdiv_(self.m, denom, value=-self.cfg.lr * sqrt_bc2 / bc1)
+        eps2 = self.cfg.adam_eps * sqrt_bc2
+        step_size = -self.cfg.lr * sqrt_bc2 / bc1
+        # chunked denom: bounds the fp32 sqrt temporary to CHUNK elements
+        # (a flat v.sqrt() would transiently allocate the full 4B/param)
+        CHUNK = 1 << 29  # 512M elems = 2 GiB fp32
+        for off in range(0, self.v.numel(), CHUNK):
+            sl = slice(off, min(off + CHUNK, self.v.numel()))
+            denom = self.v[sl].sqrt().add_(eps2)
+            self.master[sl].addcdiv_(self.m[sl], denom, value=step_size)
         self.flat_param.copy_(self.master)
 
 
