@@ -211,6 +211,7 @@ class PerfLLM(PerfBase):
         s, m = self.strategy, self.model_config
         self.stage_layer_counts = stage_layers(s, m)
         self.chunks = []
+        self.vchunks = None
         first_idx = 0
         for stage in range(s.pp_size):
             layer_num = self.stage_layer_counts[stage]
@@ -224,6 +225,27 @@ class PerfLLM(PerfBase):
             )
             first_idx += layer_num
             self.chunks.append(chunk)
+        vp = max(1, s.interleaving_size)
+        if vp > 1:
+            # virtual-chunk models: layers split over pp*vp virtual stages
+            # in chunk-major order (Megatron interleaving; reference parity
+            # perf_llm.py:786-835)
+            assert m.layer_num % (s.pp_size * vp) == 0, (
+                "interleaving requires layer_num % (pp*vp) == 0")
+            per_v = m.layer_num // (s.pp_size * vp)
+            self.vchunks = []
+            for stage in range(s.pp_size):
+                row = []
+                for c in range(vp):
+                    v = c * s.pp_size + stage
+                    row.append(LLMModel(
+                        m, s, self.system, layer_num=per_v,
+                        with_embedding=(v == 0),
+                        with_loss=(v == s.pp_size * vp - 1),
+                        first_layer_idx=v * per_v,
+                        name=f"stage{stage}.chunk{c}",
+                    ))
+                self.vchunks.append(row)
 
     def _input_info_for_stage(self, stage: int) -> InputOutputInfo:
         s, m = self.strategy, self.model_config
@@ -247,6 +269,15 @@ class PerfLLM(PerfBase):
             # recompute flags were applied after the call; refresh cost-deps
             self._refresh_recompute_costs(chunk)
             chunk.compute_activations()
+        if self.vchunks is not None:
+            for stage, row in enumerate(self.vchunks):
+                for c, chunk in enumerate(row):
+                    v = c * self.strategy.pp_size + stage
+                    chunk(self._input_info_for_stage(0 if v == 0 else stage + 1),
+                          self.debug_ctx)
+                    chunk.apply_recompute()
+                    self._refresh_recompute_costs(chunk)
+                    chunk.compute_activations()
         self._estimated = True
 
     @staticmethod
@@ -298,11 +329,28 @@ class PerfLLM(PerfBase):
         s = self.strategy
         out = Result()
         stages = []
+        vp = max(1, s.interleaving_size)
         for stage, chunk in enumerate(self.chunks):
             model_info = chunk.get_model_info()
             pp_point = chunk.peak_point
             inflight = self._inflight_microbatches(stage)
             cache_per_mb = pp_point.cache_mem
+            if vp > 1 and self.vchunks is not None:
+                # stage holds its vp virtual chunks; in-flight fwd acts are
+                # per-VIRTUAL-chunk caches (reference: perf_llm.py:1801-1828)
+                from .vpp import interleaved_inflight_microbatches
+
+                model_info = self.vchunks[stage][0].get_model_info()
+                for c in range(1, vp):
+                    model_info = model_info + self.vchunks[stage][c].get_model_info()
+                caches = [self.vchunks[stage][c].peak_point.cache_mem
+                          for c in range(vp)]
+                inflight = interleaved_inflight_microbatches(
+                    s.pp_size, vp, s.micro_batch_num, stage)
+                cache_per_mb = max(caches)
+                pp_point = max((self.vchunks[stage][c].peak_point
+                                for c in range(vp)),
+                               key=lambda x: x.peak_mem)
             peak = (
                 model_info.all_bytes
                 + (inflight - 1) * cache_per_mb
@@ -431,14 +479,23 @@ class PerfLLM(PerfBase):
             ideal = max(mbc * (f + b) for f, b in zip(fwd, bwd))
             bubble_time = pipeline_time - ideal
         else:
-            # interleaved sync-VPP analytic estimate (Megatron formula)
-            f_c = [f / 1 for f in fwd]  # per virtual chunk times ~ equal split
-            chunk_f = sum(fwd) / (pp * vp)
-            chunk_b = sum(bwd) / (pp * vp)
-            ideal = mbc * (sum(fwd) + sum(bwd)) / pp
-            bubble_time = (pp - 1) * (chunk_f + chunk_b + 2 * p2p_time)
-            pipeline_time = ideal + bubble_time
-            self.schedule_records = None
+            # exact interleaved sync-VPP schedule (Megatron schedule table)
+            from .vpp import schedule_interleaved
+
+            vf = [[0.0] * vp for _ in range(pp)]
+            vb = [[0.0] * vp for _ in range(pp)]
+            for stage in range(pp):
+                for c in range(vp):
+                    ci = self.vchunks[stage][c].get_cost_info()
+                    vf[stage][c] = ci.fwd_compute_time + ci.fwd_net_exposed_time
+                    vb[stage][c] = (ci.bwd_compute_time + ci.bwd_net_exposed_time
+                                    + ci.recompute_compute_time
+                                    + ci.recompute_net_exposed_time)
+            pipeline_time, records = schedule_interleaved(
+                pp, vp, mbc, vf, vb, p2p_time)
+            self.schedule_records = records
+            ideal = max(mbc * (sum(vf[s_]) + sum(vb[s_])) for s_ in range(pp))
+            bubble_time = pipeline_time - ideal
 
         # straggler
         n = get_effective_straggler_sample_count(s, self.system.num_per_node)
