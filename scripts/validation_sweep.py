@@ -1,0 +1,111 @@
+"""Predicted-vs-measured validation sweep (1 GPU): several Llama configs
+in one process. Writes JSON lines to gpurun_out/validation.jsonl.
+
+The in-repo analog of the reference's perf-vs-real pipeline
+(tools/b200/run_megatron_perf_real_pipeline.py) with the in-repo trainer
+standing in for Megatron-LM.
+"""
+import gc
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+from simumax_amd import ModelConfig, get_simu_model_config
+from simumax_amd.train.trainer import (TrainConfig, build_trainer,
+                                       make_synthetic_batch, train_step)
+
+CASES = [
+    # (name, model, seq, mbs, mbc, steps)
+    ("8b_seq4096_mbc4", "llama3-8b", 4096, 1, 4, 3),
+    ("8b_seq4096_mbc8", "llama3-8b", 4096, 1, 8, 2),
+    ("8b_seq2048_mbs2", "llama3-8b", 2048, 2, 4, 3),
+    ("8b_seq8192_mbc2", "llama3-8b", 8192, 1, 2, 3),
+    ("70b_l12_seq4096", "llama3-70b-l12", 4096, 1, 2, 3),
+]
+
+OUT = "gpurun_out/validation.jsonl"
+
+
+def predict(model_cfg, seq, mbs, mbc):
+    import copy
+
+    from simumax_amd import (PerfLLM, StrategyConfig, SystemConfig,
+                             get_simu_system_config)
+
+    st = StrategyConfig(
+        seq_len=seq, micro_batch_size=mbs, micro_batch_num=mbc,
+        world_size=1, tp_size=1, pp_size=1,
+        enable_sequence_parallel=False, zero_state=0,
+        use_fp32_accum_grad=True, enable_recompute=False,
+        cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
+        mem_factor=1.0)
+    p = PerfLLM()
+    p.configure(st, copy.deepcopy(model_cfg),
+                SystemConfig.init_from_config_file(
+                    get_simu_system_config("mi355x")))
+    p.run_estimate()
+    return p.analysis_cost(), p.analysis_mem()
+
+
+def run_case(name, model, seq, mbs, mbc, steps, warmup=1):
+    mc = ModelConfig.init_from_config_file(get_simu_model_config(model))
+    tc = TrainConfig(seq_len=seq, micro_batch_size=mbs, micro_batch_num=mbc)
+    t0 = time.time()
+    m, opt, red = build_trainer(mc, tc, "cuda:0")
+    toks, labels = make_synthetic_batch(mc.vocab_size, mbc, mbs, seq, "cuda:0")
+    for _ in range(warmup):
+        train_step(m, opt, red, toks, labels, mbc)
+    torch.cuda.reset_peak_memory_stats()
+    torch.cuda.synchronize()
+    t1 = time.time()
+    for _ in range(steps):
+        train_step(m, opt, red, toks, labels, mbc)
+    torch.cuda.synchronize()
+    ms = (time.time() - t1) / steps * 1e3
+    peak = torch.cuda.max_memory_allocated()
+    cost, mem = predict(mc, seq, mbs, mbc)
+    row = dict(
+        case=name, model=model, seq=seq, mbs=mbs, mbc=mbc,
+        measured_ms=round(ms, 2), predicted_ms=round(cost["iter_time"], 2),
+        timing_err_pct=round((cost["iter_time"] - ms) / ms * 100, 2),
+        measured_gib=round(peak / 2**30, 2),
+        predicted_gib=round(mem["max_peak_mem"] / 2**30, 2),
+        mem_err_pct=round((mem["max_peak_mem"] - peak) / peak * 100, 2),
+        measured_mfu=round(
+            mc.flops_per_token(seq) * mbs * mbc * seq / (ms / 1e3) / 2.5e15, 4),
+        predicted_mfu=round(cost["mfu"], 4),
+        build_s=round(t1 - t0, 1),
+    )
+    print(json.dumps(row), flush=True)
+    with open(OUT, "a") as f:
+        f.write(json.dumps(row) + "\n")
+    # free everything before the next case
+    del m, opt, red, toks, labels
+    gc.collect()
+    torch.cuda.empty_cache()
+    return row
+
+
+def main():
+    os.makedirs("gpurun_out", exist_ok=True)
+    only = sys.argv[1] if len(sys.argv) > 1 else None
+    for case in CASES:
+        if only and only not in case[0]:
+            continue
+        try:
+            run_case(*case)
+        except torch.cuda.OutOfMemoryError:
+            print(json.dumps({"case": case[0], "error": "OOM"}), flush=True)
+            torch.cuda.empty_cache()
+        except Exception as e:  # keep sweeping
+            print(json.dumps({"case": case[0], "error": str(e)[:200]}),
+                  flush=True)
+
+
+if __name__ == "__main__":
+    main()

@@ -284,6 +284,14 @@ def enumerate_shapes(cases):
 DEFAULT_CASES = [
     ("llama3-8b", dict(tp_size=1, pp_size=1, enable_sequence_parallel=False,
                        micro_batch_num=4, zero_state=0)),
+    # single-GPU validation-sweep shapes
+    ("llama3-8b", dict(world_size=1, tp_size=1, pp_size=1, seq_len=2048,
+                       micro_batch_size=2, enable_sequence_parallel=False,
+                       zero_state=0)),
+    ("llama3-8b", dict(world_size=1, tp_size=1, pp_size=1, seq_len=8192,
+                       enable_sequence_parallel=False, zero_state=0)),
+    ("llama3-70b-l12", dict(world_size=1, tp_size=1, pp_size=1,
+                            enable_sequence_parallel=False, zero_state=0)),
     ("llama3-8b", dict(tp_size=2)),
     ("llama3-8b", dict(tp_size=4)),
     ("llama3-8b", dict(tp_size=8)),
