@@ -88,6 +88,10 @@ class Embedding(MetaModule, ParamMixin):
     def _leaf_act_info(self, info):
         # bwd scatter-add needs the token ids
         info.activation_mem_cache = self.input_info.tensors[0].mem_bytes()
+        # autograd materializes a DENSE bf16 weight grad, converted to fp32
+        # for main_grad accumulation (2+4 B/elem transient)
+        local_numel = self.vocab_size // self.strategy.tp_size * self.hidden_size
+        info.bwd_peak_mem_no_cache = local_numel * (self.element_size + 4)
 
     def _leaf_compute_info(self, info):
         out_full = self.output_info.first.mem_bytes()
@@ -532,6 +536,8 @@ class ParallelCE(MetaModule):
         if not self.strategy.cross_entropy_loss_fusion:
             # unfused keeps fp32 softmax copy transiently
             info.fwd_peak_mem_no_cache = t.numel() * FP32
+        # bwd materializes dlogits (same size as logits)
+        info.bwd_peak_mem_no_cache = t.mem_bytes()
 
     def _leaf_compute_info(self, info):
         t = self.input_info.first

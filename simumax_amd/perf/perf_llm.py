@@ -351,8 +351,17 @@ class PerfLLM(PerfBase):
                 pp_point = max((self.vchunks[stage][c].peak_point
                                 for c in range(vp)),
                                key=lambda x: x.peak_mem)
+            dummy_wgrad = 0.0
+            if s.use_fused_grad_accumulation:
+                from ..core.module import LinearBase
+
+                shapes = {(l.input_size, l.output_size)
+                          for l in chunk.leaf_modules()
+                          if isinstance(l, LinearBase)}
+                dummy_wgrad = sum(i * o * 2 for i, o in shapes)
             peak = (
                 model_info.all_bytes
+                + dummy_wgrad
                 + (inflight - 1) * cache_per_mb
                 + pp_point.peak_mem
             )
@@ -360,6 +369,7 @@ class PerfLLM(PerfBase):
             stages.append(
                 dict(
                     stage=stage,
+                    dummy_wgrad_mem=dummy_wgrad,
                     weight_mem=model_info.weight_bytes,
                     grad_mem=model_info.grad_bytes,
                     state_mem=model_info.state_bytes,
