@@ -512,7 +512,14 @@ class PerfLLM(PerfBase):
         straggler = estimate_straggler_increase_ratio(n) if s.enable_straggler_model else 1.0
         pipeline_time *= straggler
 
-        dp_time = max(self._compute_dp_time(i) for i in range(pp))
+        dp_time_raw = max(self._compute_dp_time(i) for i in range(pp))
+        if s.overlap_grad_reduce and dp_time_raw > 0:
+            # bucketed grad reduce rides the LAST microbatch's backward
+            # (Megatron no_sync semantics); only the tail is exposed
+            bwd_last = max(bwd) if bwd else 0.0
+            dp_time = max(0.0, dp_time_raw - bwd_last)
+        else:
+            dp_time = dp_time_raw
         optim_time = max(self._compute_optim_time(i) for i in range(pp))
         iter_time = pipeline_time + dp_time + optim_time
 
@@ -532,6 +539,7 @@ class PerfLLM(PerfBase):
                 bubble_time=bubble_time,
                 p2p_time_per_hop=p2p_time,
                 dp_time=dp_time,
+                dp_time_raw=dp_time_raw,
                 optim_time=optim_time,
                 straggler_ratio=straggler,
                 mfu=mfu,

@@ -179,6 +179,11 @@ class PpSchedule:
                       [x for x in ranks
                        if x // (s.world_size // pp) == stage]):
                 dp_t = self.perf._compute_dp_time(stage)
+                if s.overlap_grad_reduce and dp_t > 0:
+                    # exposed tail only (reduce overlaps last-mb backward)
+                    ci = self.perf.chunks[stage].get_cost_info()
+                    dp_t = max(0.0, dp_t - (ci.bwd_compute_time
+                                            + ci.bwd_net_exposed_time))
                 if dp_t > 0:
                     jobs[r].append(Job(name="dp_grad_sync", kind="comm",
                                        dur=dp_t, lane="comm",
