@@ -599,6 +599,43 @@ class Float8Quantizer(MetaModule):
         info.bwd_grad_act_accessed_mem = in_b + out_b
 
 
+class QuantizedColLinear(MetaModule):
+    """Float8Quantizer -> LinearCol with fp8 op keys (reference parity:
+    dense_module.py:2365-2453). The quantizer's fp8 copies are the extra
+    activation the fp8 path caches."""
+
+    def __init__(self, input_size, output_size, strategy, system,
+                 name="quant_linear_col", **kw):
+        super().__init__(strategy, system, name)
+        assert strategy.fp8, "QuantizedColLinear requires strategy.fp8"
+        self.quantizer = Float8Quantizer(strategy, system, "quantize")
+        self.linear = LinearCol(input_size, output_size, strategy, system,
+                                "linear", **kw)
+
+    def forward(self, input_info):
+        q = self.quantizer(input_info, self.path_debug_context)
+        # the GEMM consumes the fp8 tensor but keys/caches track bf16 I/O
+        return self.linear(InputOutputInfo([t.to(self.strategy.dtype)
+                                            for t in q.tensors]),
+                           self.path_debug_context)
+
+
+class QuantizedRowLinear(MetaModule):
+    def __init__(self, input_size, output_size, strategy, system,
+                 name="quant_linear_row", **kw):
+        super().__init__(strategy, system, name)
+        assert strategy.fp8, "QuantizedRowLinear requires strategy.fp8"
+        self.quantizer = Float8Quantizer(strategy, system, "quantize")
+        self.linear = LinearRow(input_size, output_size, strategy, system,
+                                "linear", **kw)
+
+    def forward(self, input_info):
+        q = self.quantizer(input_info, self.path_debug_context)
+        return self.linear(InputOutputInfo([t.to(self.strategy.dtype)
+                                            for t in q.tensors]),
+                           self.path_debug_context)
+
+
 class Add(MetaModule):
     """Residual add: 2 reads + 1 write, nothing cached (linear)."""
 

@@ -189,3 +189,43 @@ def test_cp_all_gather_mode():
     kinds = [(e.stage, e.op_name) for e in sdp.comm_ops]
     assert ("fwd", "all_gather") in kinds
     assert ("bwd_act", "reduce_scatter") in kinds
+
+
+def test_fp8_quantized_linear():
+    from simumax_amd.core.records import InputOutputInfo
+    from simumax_amd.core.tensor import TensorSize
+    from simumax_amd.ops.dense import QuantizedColLinear
+    from tests.test_ops import make_strategy
+
+    from simumax_amd import SystemConfig, get_simu_system_config
+
+    sysc = SystemConfig.init_from_config_file(get_simu_system_config("mi355x"))
+    s = make_strategy(fp8=True)
+    q = QuantizedColLinear(4096, 4096, s, sysc)
+    q(InputOutputInfo([TensorSize([1, 4096, 4096])]))
+    assert q.linear.fwd_op == "fp8_matmul"
+    # fp8 path caches the quantizer's row+col fp8 copies on top of inputs
+    assert q.get_act_info().activation_mem_cache > 4096 * 4096 * 2
+
+
+def test_golden_comparator():
+    from simumax_amd.testing.base_test_tool import ResultCheck
+
+    golden = {"iter_time": 100.0, "mem": "1.00 GB", "nested": {"x": [1, 2]}}
+    got_ok = {"iter_time": 100.05, "mem": str(1.0005 * 2**30) + " B",
+              "nested": {"x": [1, 2]}, "extra": 5}
+    rc = ResultCheck(rel_tol=1e-2)
+    assert rc.check(got_ok, golden), rc.report()
+    rc2 = ResultCheck(rel_tol=1e-3)
+    assert not rc2.check({"iter_time": 105.0}, {"iter_time": 100.0})
+    assert "iter_time" in rc2.report()
+
+
+def test_dualpp_speedup():
+    from simumax_amd.perf.dualpp import duration_1f1b, perf_dualpp
+
+    r = perf_dualpp(pp=8, mbc=32, f=10.0, b=20.0, flops_per_mb=1e12,
+                    peak_flops=2.5e15)
+    assert r["speedup"] > 1.0
+    assert r["mfu_dualpp"] > r["mfu_1f1b"]
+    assert duration_1f1b(8, 32, 10, 20) == (32 + 7) * 30
