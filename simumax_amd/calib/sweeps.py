@@ -165,6 +165,36 @@ def sweep_sdp(shape_keys, fwd_path, bwd_path, sparse_ratio=0.5):
     return fwd_tab, bwd_tab
 
 
+def parse_group_key(desc):
+    m = re.match(r"ng=(\d+), M=(\d+), N=(\d+), K=(\d+), dtype=(\w+)", desc)
+    assert m, desc
+    return int(m.group(1)), int(m.group(2)), int(m.group(3)), int(m.group(4))
+
+
+def sweep_grouped(shape_keys, path):
+    """Time ng per-expert GEMMs (hipBLASLt via torch) — the trainer-side
+    grouped-GEMM realization on ROCm ("parallel" group_linear_mode)."""
+    tab = _load(path)
+    for desc in shape_keys:
+        if desc in tab and not OVERWRITE:
+            continue
+        ng, m, n, k = parse_group_key(desc)
+        try:
+            x = torch.randn(ng, m, k, device="cuda", dtype=torch.bfloat16)
+            w = torch.randn(ng, k, n, device="cuda", dtype=torch.bfloat16)
+            fn = lambda: torch.bmm(x, w)
+            t_ms = _timeit(fn, iters=8)
+        except torch.cuda.OutOfMemoryError:
+            torch.cuda.empty_cache()
+            print(f"[group] OOM {desc}", flush=True)
+            continue
+        flops = 2 * ng * m * k * n
+        tab[desc] = flops / (t_ms / 1e3) / PEAK_BF16
+        print(f"[group] {desc} -> eff {tab[desc]:.4f} ({t_ms:.3f} ms)", flush=True)
+        _save(path, tab)
+    return tab
+
+
 def sweep_bandwidth(path):
     """HBM stream + fused-op bandwidth efficiencies + optimizer traffic."""
     from simumax_amd.kernels.ops import ext
@@ -317,6 +347,8 @@ def main():
     if which in ("all", "sdp"):
         sweep_sdp(sdp_keys, os.path.join(OUT_DIR, "sdp_fwd.json"),
                   os.path.join(OUT_DIR, "sdp_bwd.json"))
+    if which in ("all", "group"):
+        sweep_grouped(group_keys, os.path.join(OUT_DIR, "group_matmul.json"))
     print("[calib] done", flush=True)
 
 

@@ -209,7 +209,9 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
                     float p1 = (s1[j] == -INFINITY) ? 0.f : __expf(s1[j] - mn);
                     s0[j] = p0;
                     s1[j] = p1;
-                    l[qs][j] = l[qs][j] * alpha[j] + group16_sum(p0 + p1);
+                    // l kept as PER-LANE partials (this lane's 2 columns);
+                    // the cross-lane reduction happens once in the epilogue
+                    l[qs][j] = l[qs][j] * alpha[j] + p0 + p1;
                 }
 #pragma unroll
                 for (int dt = 0; dt < 8; ++dt)
@@ -222,11 +224,13 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
                     pw[(kgrp * 4 + j) * PS + 16 + col] = f2bf(s1[j]);
                 }
                 bf16x8v a_p = ld_frag(pw + col * PS + kgrp * 8);
+                __builtin_amdgcn_s_setprio(1);
 #pragma unroll
                 for (int dt = 0; dt < 8; ++dt) {
                     bf16x8v b_v = tr_frag(V_img + dt * VSUB, lane);
                     acc[qs][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, b_v, acc[qs][dt], 0, 0, 0);
                 }
+                __builtin_amdgcn_s_setprio(0);
             }
         }
         stage_write(t + 1);
@@ -235,6 +239,12 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     }
 
     bf16raw *op = o + ((long)b * S) * q_row + (long)h * DHEAD;
+    // reduce the per-lane l partials across the 16-lane column group
+    // exactly once (the main loop keeps l lane-local)
+#pragma unroll
+    for (int qs = 0; qs < 2; ++qs)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) l[qs][j] = group16_sum(l[qs][j]);
 #pragma unroll
     for (int qs = 0; qs < 2; ++qs)
 #pragma unroll
