@@ -257,7 +257,37 @@ class MetaModule:
                 bwd_time=self._cost_info.bwd_grad_act_time,
                 wgrad_time=self._cost_info.bwd_grad_w_time,
             )
+            self._dump_debug_point()
         return out
+
+    def _dump_debug_point(self):
+        """Append this path's F/B/W costs to TMP_PATH/cost_log.json
+        (reference parity: base_struct.py:771-800 debug_points)."""
+        import json
+        import os
+
+        from .consts import TMP_PATH
+
+        os.makedirs(TMP_PATH, exist_ok=True)
+        path_file = os.path.join(TMP_PATH, "cost_log.json")
+        data = {}
+        if os.path.exists(path_file):
+            try:
+                with open(path_file, encoding="utf-8") as f:
+                    data = json.load(f)
+            except json.JSONDecodeError:
+                data = {}
+        ci = self._cost_info
+        data[self.full_name] = {
+            "cost_F": ci.fwd_compute_time,
+            "cost_B": ci.bwd_grad_act_time,
+            "cost_W": ci.bwd_grad_w_time,
+            "recompute_F": ci.recompute_compute_time,
+            "net_F": ci.fwd_net_time,
+            "net_B": ci.bwd_net_time,
+        }
+        with open(path_file, "w", encoding="utf-8") as f:
+            json.dump(data, f, indent=4, ensure_ascii=False)
 
     # ---- recompute DFS ---------------------------------------------------
     def set_recompute(self, flag: bool = True):

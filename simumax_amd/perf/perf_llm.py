@@ -588,6 +588,10 @@ class PerfLLM(PerfBase):
                 json.dump(res["base_info"], f, indent=2, default=str)
             with open(os.path.join(save_path, "net_info.json"), "w") as f:
                 json.dump(self.system.real_comm_bw, f, indent=2, default=str)
+            with open(os.path.join(save_path, "efficiency_coverage.json"), "w") as f:
+                json.dump({"miss_efficiency": self.system.miss_efficiency,
+                           "hit_efficiency": self.system.hit_efficiency},
+                          f, indent=2, default=str)
             with open(os.path.join(save_path, "model_arch"), "w") as f:
                 f.write(repr(self.chunks[0]))
         return res

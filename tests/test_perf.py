@@ -229,3 +229,21 @@ def test_dualpp_speedup():
     assert r["speedup"] > 1.0
     assert r["mfu_dualpp"] > r["mfu_1f1b"]
     assert duration_1f1b(8, 32, 10, 20) == (32 + 7) * 30
+
+
+def test_debug_points_cost_log(tmp_path, monkeypatch):
+    import simumax_amd.core.consts as consts
+
+    monkeypatch.setattr(consts, "TMP_PATH", str(tmp_path))
+    p = PerfLLM()
+    p.debug_points = ["stage0.layer0.mlp.fc1"]
+    st = StrategyConfig.init_from_config_file(
+        get_simu_strategy_config("tp1_pp1_dp8_mbs1"))
+    p.configure(st,
+                ModelConfig.init_from_config_file(get_simu_model_config("llama2-tiny")),
+                SystemConfig.init_from_config_file(get_simu_system_config("mi355x")))
+    p.run_estimate()
+    import json as _json
+    log = _json.loads((tmp_path / "cost_log.json").read_text())
+    assert "stage0.layer0.mlp.fc1" in log
+    assert log["stage0.layer0.mlp.fc1"]["cost_F"] > 0
