@@ -103,3 +103,20 @@ def test_human_readable_roundtrip():
     s = HumanReadableSize.format_bytes(66.44 * 1024**3)
     assert s == "66.44 GB"
     assert HumanReadableSize.from_string(s) == pytest.approx(66.44 * 1024**3)
+
+
+def test_rccl_fit_bw_latency():
+    """alpha-beta fit recovers synthetic bw/latency (CPU-only)."""
+    from simumax_amd.calib.rccl_sweep import fit_bw_latency
+
+    bw_true = 900 * 1024**3          # bytes/s
+    lat_true_ms = 0.02
+    n, scale, offset = 8, 2, -1
+    rows = []
+    for size in [2**i for i in range(24, 33)]:
+        actual = size * scale + size * scale / n * offset
+        t_ms = actual / bw_true * 1e3 + lat_true_ms
+        rows.append((size, t_ms))
+    bw, lat = fit_bw_latency(rows, scale, offset, n)
+    assert bw == pytest.approx(900, rel=0.01)
+    assert lat == pytest.approx(lat_true_ms, rel=0.05)
