@@ -62,6 +62,10 @@ def main():
             # unfused ce not implemented separately on MI355X: same kernel
             acc["bandwidth"]["ce"]["efficient_factor"] = acc["bandwidth"][
                 "ce_fusion"]["efficient_factor"]
+        for key in ("permute_fwd", "permute_bwd"):
+            if f"{key}_eff" in bw:
+                acc["bandwidth"][key]["efficient_factor"] = round(
+                    bw[f"{key}_eff"], 4)
         if "optimizer_eff" in bw:
             acc["bandwidth"]["optimizer"] = {
                 "gbps": 8000.0,
