@@ -135,8 +135,8 @@ def sweep_sdp(shape_keys, fwd_path, bwd_path, sparse_ratio=0.5):
         if desc in fwd_tab and desc in bwd_tab and not OVERWRITE:
             continue
         b, s, hq, hkv, dqk, dv = parse_sdp_key(desc)
-        if dqk != 128 or dv != 128:
-            print(f"[sdp] skip non-D128 shape {desc}", flush=True)
+        if dqk not in (128, 192) or dv != 128:
+            print(f"[sdp] skip unsupported shape {desc}", flush=True)
             continue
         dt = torch.bfloat16
         try:

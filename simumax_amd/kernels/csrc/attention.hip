@@ -77,14 +77,21 @@ DEV bf16x8v tr_frag(const bf16raw *sub_base, int lane) {
     return r;
 }
 
-extern "C" __global__ __launch_bounds__(FA_BLOCK, 2)
+// DQK: q/k head dim (128 dense/GQA, 192 MLA); v/o stay 128 (DHEAD).
+// NQS: 16-row q subtiles per wave (2 at DQK=128; 1 at 192 for registers).
+template <int DQK, int NQS>
+__global__ __launch_bounds__(FA_BLOCK, 2)
 void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
                    const bf16raw *__restrict__ v, bf16raw *__restrict__ o,
                    float *__restrict__ lse, int B, int S, int Hq, int Hkv,
                    int causal) {
+    constexpr int KC = DQK / 32;          // 32-deep k chunks of QK^T
+    constexpr int KS_T = DQK + 8;         // padded K row stride
+    constexpr int QT = NQS * 16;          // q rows per wave
+    constexpr int BUF_ELEMS = KVTILE * KS_T + 8 * VSUB;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     bf16raw *buf0 = reinterpret_cast<bf16raw *>(smem);
-    bf16raw *P_all = buf0 + 2 * FWD_BUF_ELEMS;
+    bf16raw *P_all = buf0 + 2 * BUF_ELEMS;
 
     const int qblk = blockIdx.x;            // 128-row q block
     const int h = blockIdx.y;
@@ -95,22 +102,24 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     const int col = lane & 15;
     const int kgrp = lane >> 4;
 
-    const int qbase = qblk * (WAVES * QTILE) + wave * QTILE;
-    const float scale = rsqrtf((float)DHEAD);
+    const int qbase = qblk * (WAVES * QT) + wave * QT;
+    const float scale = rsqrtf((float)DQK);
 
-    const long q_row = (long)Hq * DHEAD;
-    const long kv_row = (long)Hkv * DHEAD;
-    const bf16raw *qp = q + ((long)b * S) * q_row + (long)h * DHEAD;
-    const bf16raw *kp = k + ((long)b * S) * kv_row + (long)hkv * DHEAD;
-    const bf16raw *vp = v + ((long)b * S) * kv_row + (long)hkv * DHEAD;
-    bf16raw *Pw = P_all + wave * QTILE * PS;
+    const long q_row = (long)Hq * DQK;        // q layout [B,S,Hq,DQK]
+    const long k_row = (long)Hkv * DQK;       // k layout [B,S,Hkv,DQK]
+    const long v_row = (long)Hkv * DHEAD;     // v layout [B,S,Hkv,128]
+    const long o_row = (long)Hq * DHEAD;      // o layout [B,S,Hq,128]
+    const bf16raw *qp = q + ((long)b * S) * q_row + (long)h * DQK;
+    const bf16raw *kp = k + ((long)b * S) * k_row + (long)hkv * DQK;
+    const bf16raw *vp = v + ((long)b * S) * v_row + (long)hkv * DHEAD;
+    bf16raw *Pw = P_all + wave * QT * PS;
 
-    bf16x8v a_q[2][4];
+    bf16x8v a_q[NQS][KC];
 #pragma unroll
-    for (int qs = 0; qs < 2; ++qs) {
+    for (int qs = 0; qs < NQS; ++qs) {
         const int qrow = qbase + qs * 16 + col;
 #pragma unroll
-        for (int kc = 0; kc < 4; ++kc) {
+        for (int kc = 0; kc < KC; ++kc) {
             bf16x8 raw = load8(qp + (long)min(qrow, S - 1) * q_row + kc * 32 + kgrp * 8);
             bf16x8 sc;
 #pragma unroll
@@ -119,50 +128,62 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
         }
     }
 
-    float m[2][4], l[2][4];
-    f32x4 acc[2][8];
+    float m[NQS][4], l[NQS][4];
+    f32x4 acc[NQS][8];
 #pragma unroll
-    for (int qs = 0; qs < 2; ++qs)
+    for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
             m[qs][j] = -INFINITY;
             l[qs][j] = 0.f;
         }
 #pragma unroll
-    for (int qs = 0; qs < 2; ++qs)
+    for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
         for (int dt = 0; dt < 8; ++dt) acc[qs][dt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    const int blk_rows = WAVES * QTILE;
+    const int blk_rows = WAVES * QT;
     const int kv_limit = causal ? min(S, qblk * blk_rows + blk_rows) : S;
     const int n_tiles = CDIV(kv_limit, KVTILE);
 
     // staging: T14 split — issue global loads into registers early, write
     // to LDS after compute. Per thread: 2 chunks of (K 8 elems + V 8 elems).
-    bf16x8 st_k[2], st_v[2];
+    constexpr int KCHUNKS = KVTILE * DQK / (FA_BLOCK * 8);   // per-thread K pieces
+    bf16x8 st_k[KCHUNKS], st_v[2];
     auto stage_load = [&](int t) {
         if (t >= n_tiles) return;
         const int kv = t * KVTILE;
         const int tid = threadIdx.x;
 #pragma unroll
-        for (int c = 0; c < 2; ++c) {
+        for (int c = 0; c < KCHUNKS; ++c) {
             const int e = tid * 8 + c * FA_BLOCK * 8;
-            const int kvr = e / DHEAD, d0 = e % DHEAD;
+            const int kvr = e / DQK, d0 = e % DQK;
             const int src = min(kv + kvr, S - 1);
-            st_k[c] = load8(kp + (long)src * kv_row + d0);
-            st_v[c] = load8(vp + (long)src * kv_row + d0);
+            st_k[c] = load8(kp + (long)src * k_row + d0);
         }
-    };
-    auto stage_write = [&](int t) {
-        if (t >= n_tiles) return;
-        bf16raw *K_lds = buf0 + (t & 1) * FWD_BUF_ELEMS;
-        bf16raw *V_img = K_lds + KVTILE * KS;
-        const int tid = threadIdx.x;
 #pragma unroll
         for (int c = 0; c < 2; ++c) {
             const int e = tid * 8 + c * FA_BLOCK * 8;
             const int kvr = e / DHEAD, d0 = e % DHEAD;
-            store8(K_lds + kvr * KS + d0, st_k[c]);
+            const int src = min(kv + kvr, S - 1);
+            st_v[c] = load8(vp + (long)src * v_row + d0);
+        }
+    };
+    auto stage_write = [&](int t) {
+        if (t >= n_tiles) return;
+        bf16raw *K_lds = buf0 + (t & 1) * BUF_ELEMS;
+        bf16raw *V_img = K_lds + KVTILE * KS_T;
+        const int tid = threadIdx.x;
+#pragma unroll
+        for (int c = 0; c < KCHUNKS; ++c) {
+            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int kvr = e / DQK, d0 = e % DQK;
+            store8(K_lds + kvr * KS_T + d0, st_k[c]);
+        }
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int kvr = e / DHEAD, d0 = e % DHEAD;
             // V image: subtile d0/16, row perm(kvr), col d0%16
             store8(V_img + (d0 >> 4) * VSUB + v_img_row(kvr) * 16 + (d0 & 15),
                    st_v[c]);
@@ -176,20 +197,33 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
 
     for (int t = 0; t < n_tiles; ++t) {
         const int kv = t * KVTILE;
-        bf16raw *K_lds = buf0 + (t & 1) * FWD_BUF_ELEMS;
-        bf16raw *V_img = K_lds + KVTILE * KS;
+        bf16raw *K_lds = buf0 + (t & 1) * BUF_ELEMS;
+        bf16raw *V_img = K_lds + KVTILE * KS_T;
 
-        if (!(causal && kv >= qbase + QTILE)) {
+        if (!(causal && kv >= qbase + QT)) {
+            // QK^T for ALL q-subtiles per K fragment: each B-fragment is
+            // read once and feeds NQS independent MFMA chains
+            f32x4 sq[NQS][2];
 #pragma unroll
-            for (int qs = 0; qs < 2; ++qs) {
-                f32x4 s0 = f32x4{0, 0, 0, 0}, s1 = f32x4{0, 0, 0, 0};
+            for (int qs = 0; qs < NQS; ++qs) {
+                sq[qs][0] = f32x4{0, 0, 0, 0};
+                sq[qs][1] = f32x4{0, 0, 0, 0};
+            }
 #pragma unroll
-                for (int kc = 0; kc < 4; ++kc) {
-                    bf16x8v b0 = ld_frag(K_lds + col * KS + kc * 32 + kgrp * 8);
-                    bf16x8v b1 = ld_frag(K_lds + (16 + col) * KS + kc * 32 + kgrp * 8);
-                    s0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], b0, s0, 0, 0, 0);
-                    s1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], b1, s1, 0, 0, 0);
+            for (int kc = 0; kc < KC; ++kc) {
+                bf16x8v b0 = ld_frag(K_lds + col * KS_T + kc * 32 + kgrp * 8);
+                bf16x8v b1 = ld_frag(K_lds + (16 + col) * KS_T + kc * 32 + kgrp * 8);
+#pragma unroll
+                for (int qs = 0; qs < NQS; ++qs) {
+                    sq[qs][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], b0, sq[qs][0], 0, 0, 0);
+                    sq[qs][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], b1, sq[qs][1], 0, 0, 0);
                 }
+            }
+            // softmax per subtile, then P->LDS
+            bf16x8v a_p[NQS];
+#pragma unroll
+            for (int qs = 0; qs < NQS; ++qs) {
+                f32x4 s0 = sq[qs][0], s1 = sq[qs][1];
                 float tile_max[4];
 #pragma unroll
                 for (int j = 0; j < 4; ++j) {
@@ -223,30 +257,33 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
                     pw[(kgrp * 4 + j) * PS + col] = f2bf(s0[j]);
                     pw[(kgrp * 4 + j) * PS + 16 + col] = f2bf(s1[j]);
                 }
-                bf16x8v a_p = ld_frag(pw + col * PS + kgrp * 8);
-                __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-                for (int dt = 0; dt < 8; ++dt) {
-                    bf16x8v b_v = tr_frag(V_img + dt * VSUB, lane);
-                    acc[qs][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p, b_v, acc[qs][dt], 0, 0, 0);
-                }
-                __builtin_amdgcn_s_setprio(0);
+                a_p[qs] = ld_frag(pw + col * PS + kgrp * 8);
             }
+            // PV for all subtiles per V fragment (one tr-read pair each)
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int dt = 0; dt < 8; ++dt) {
+                bf16x8v b_v = tr_frag(V_img + dt * VSUB, lane);
+#pragma unroll
+                for (int qs = 0; qs < NQS; ++qs)
+                    acc[qs][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p[qs], b_v, acc[qs][dt], 0, 0, 0);
+            }
+            __builtin_amdgcn_s_setprio(0);
         }
         stage_write(t + 1);
         __syncthreads();
         stage_load(t + 2);
     }
 
-    bf16raw *op = o + ((long)b * S) * q_row + (long)h * DHEAD;
+    bf16raw *op = o + ((long)b * S) * o_row + (long)h * DHEAD;
     // reduce the per-lane l partials across the 16-lane column group
     // exactly once (the main loop keeps l lane-local)
 #pragma unroll
-    for (int qs = 0; qs < 2; ++qs)
+    for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
         for (int j = 0; j < 4; ++j) l[qs][j] = group16_sum(l[qs][j]);
 #pragma unroll
-    for (int qs = 0; qs < 2; ++qs)
+    for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
             const int row = qbase + qs * 16 + kgrp * 4 + j;
@@ -254,7 +291,7 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
             const float inv = (l[qs][j] > 0.f) ? 1.f / l[qs][j] : 0.f;
 #pragma unroll
             for (int dt = 0; dt < 8; ++dt)
-                op[(long)row * q_row + dt * 16 + col] = f2bf(acc[qs][dt][j] * inv);
+                op[(long)row * o_row + dt * 16 + col] = f2bf(acc[qs][dt][j] * inv);
             if (col == 0)
                 lse[((long)b * Hq + h) * S + row] =
                     (l[qs][j] > 0.f) ? m[qs][j] + __logf(l[qs][j]) : -INFINITY;
@@ -289,20 +326,19 @@ void fa_bwd_pre_kernel(const bf16raw *__restrict__ dout,
         const long h = bh % Hq;
         const long b = bh / Hq;
         const long base = (((long)b * S + s_) * Hq + h) * DHEAD;
-        float2 d2 = *reinterpret_cast<const float2 *>(&dout[base + lane * 2]);
         // load 2 bf16 from dout and o (4 bytes each)
         const unsigned du = *reinterpret_cast<const unsigned *>(&dout[base + lane * 2]);
         const unsigned ou = *reinterpret_cast<const unsigned *>(&o[base + lane * 2]);
         float acc = bf2f((bf16raw)(du & 0xffff)) * bf2f((bf16raw)(ou & 0xffff)) +
                     bf2f((bf16raw)(du >> 16)) * bf2f((bf16raw)(ou >> 16));
-        (void)d2;
         acc = wave_sum(acc);
         if (lane == 0) D[r] = acc;
     }
 }
 
 // ---- dK/dV: wave owns BKV keys; q-side tiles staged per WG ----
-extern "C" __global__ __launch_bounds__(FA_BLOCK)
+template <int DQK>
+__global__ __launch_bounds__(FA_BLOCK)
 void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
                        const bf16raw *__restrict__ q,
                        const bf16raw *__restrict__ k,
@@ -311,11 +347,15 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
                        const float *__restrict__ Dsum,
                        float *__restrict__ dkv,  // [B,S,Hkv,2,D] fp32
                        int B, int S, int Hq, int Hkv, int causal) {
+    constexpr int KC = DQK / 32;
+    constexpr int KS_T = DQK + 8;
+    constexpr int QSUB = DQK / 16;     // q-side tr-image d subtiles for Q
+    constexpr int QS_T = DQK + 8;      // padded Q row stride
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    bf16raw *Q_lds = reinterpret_cast<bf16raw *>(smem);      // [32][QS2]
-    bf16raw *dO_lds = Q_lds + BW_QT * QS2;                   // [32][QS2]
-    bf16raw *Q_img = dO_lds + BW_QT * QS2;                   // tr image 8*VSUB
-    bf16raw *dO_img = Q_img + 8 * VSUB;                      // tr image 8*VSUB
+    bf16raw *Q_lds = reinterpret_cast<bf16raw *>(smem);      // [32][QS_T]
+    bf16raw *dO_lds = Q_lds + BW_QT * QS_T;                  // [32][QS2]
+    bf16raw *Q_img = dO_lds + BW_QT * QS2;                   // tr image QSUB*VSUB
+    bf16raw *dO_img = Q_img + QSUB * VSUB;                   // tr image 8*VSUB
     float *lse_lds = reinterpret_cast<float *>(dO_img + 8 * VSUB);
     float *D_lds = lse_lds + BW_QT;
     bf16raw *wbase = reinterpret_cast<bf16raw *>(D_lds + BW_QT);
@@ -323,9 +363,9 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
     const int lane = threadIdx.x % WAVE;
     const int col = lane & 15;
     const int kgrp = lane >> 4;
-    const int WSZ = BKV * KS + 2 * BKV * TS2;
-    bf16raw *K_l = wbase + wave * WSZ;         // [16][KS] scaled
-    bf16raw *Pt_l = K_l + BKV * KS;            // [16k][TS2]
+    const int WSZ = BKV * KS_T + 2 * BKV * TS2;
+    bf16raw *K_l = wbase + wave * WSZ;         // [16][KS_T] scaled
+    bf16raw *Pt_l = K_l + BKV * KS_T;          // [16k][TS2]
     bf16raw *dSt_l = Pt_l + BKV * TS2;         // [16k][TS2]
 
     const int kvblk = blockIdx.x;
@@ -334,55 +374,60 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
     const int hkv = h / (Hq / Hkv);
     const int kvbase = kvblk * (WAVES * BKV) + wave * BKV;
 
-    const long q_row = (long)Hq * DHEAD;
-    const long kv_row = (long)Hkv * DHEAD;
-    const bf16raw *qp = q + ((long)b * S) * q_row + (long)h * DHEAD;
-    const bf16raw *kp = k + ((long)b * S) * kv_row + (long)hkv * DHEAD;
-    const bf16raw *vp = v + ((long)b * S) * kv_row + (long)hkv * DHEAD;
-    const bf16raw *dop = dout + ((long)b * S) * q_row + (long)h * DHEAD;
-    const float scale = rsqrtf((float)DHEAD);
+    const long q_row = (long)Hq * DQK;
+    const long k_row = (long)Hkv * DQK;
+    const long v_row = (long)Hkv * DHEAD;
+    const long o_row = (long)Hq * DHEAD;
+    const bf16raw *qp = q + ((long)b * S) * q_row + (long)h * DQK;
+    const bf16raw *kp = k + ((long)b * S) * k_row + (long)hkv * DQK;
+    const bf16raw *vp = v + ((long)b * S) * v_row + (long)hkv * DHEAD;
+    const bf16raw *dop = dout + ((long)b * S) * o_row + (long)h * DHEAD;
+    const float scale = rsqrtf((float)DQK);
 
     // stage this wave's scaled K; V as B-fragments in registers
     bf16x8v b_v[4];
     {
-        for (int e = lane * 8; e < BKV * DHEAD; e += WAVE * 8) {
-            const int kvr = e / DHEAD, d0 = e % DHEAD;
+        for (int e = lane * 8; e < BKV * DQK; e += WAVE * 8) {
+            const int kvr = e / DQK, d0 = e % DQK;
             const int src = min(kvbase + kvr, S - 1);
-            bf16x8 kk = load8(kp + (long)src * kv_row + d0);
+            bf16x8 kk = load8(kp + (long)src * k_row + d0);
             bf16x8 ks;
 #pragma unroll
             for (int j = 0; j < 8; ++j) ks.set(j, kk.get(j) * scale);
-            store8(K_l + kvr * KS + d0, ks);
+            store8(K_l + kvr * KS_T + d0, ks);
         }
         const int src = min(kvbase + col, S - 1);
 #pragma unroll
         for (int kc = 0; kc < 4; ++kc)
-            b_v[kc] = ld_frag(vp + (long)src * kv_row + kc * 32 + kgrp * 8);
+            b_v[kc] = ld_frag(vp + (long)src * v_row + kc * 32 + kgrp * 8);
     }
 
-    f32x4 dv_acc[8], dk_acc[8];
+    f32x4 dv_acc[8], dk_acc[QSUB];
 #pragma unroll
-    for (int dt = 0; dt < 8; ++dt) {
-        dv_acc[dt] = f32x4{0, 0, 0, 0};
-        dk_acc[dt] = f32x4{0, 0, 0, 0};
-    }
+    for (int dt = 0; dt < 8; ++dt) dv_acc[dt] = f32x4{0, 0, 0, 0};
+#pragma unroll
+    for (int dt = 0; dt < QSUB; ++dt) dk_acc[dt] = f32x4{0, 0, 0, 0};
 
     const int q_start = causal
         ? (kvblk * (WAVES * BKV) / BW_QT) * BW_QT : 0;
     for (int qt = q_start; qt < S; qt += BW_QT) {
         {
             const int tid = threadIdx.x;
+            for (int e = tid * 8; e < BW_QT * DQK; e += FA_BLOCK * 8) {
+                const int r = e / DQK, d0 = e % DQK;
+                const int src = min(qt + r, S - 1);
+                bf16x8 qq = load8(qp + (long)src * q_row + d0);
+                store8(Q_lds + r * QS_T + d0, qq);
+                store8(Q_img + (d0 >> 4) * VSUB + v_img_row(r) * 16 + (d0 & 15),
+                       qq);
+            }
             for (int e = tid * 8; e < BW_QT * DHEAD; e += FA_BLOCK * 8) {
                 const int r = e / DHEAD, d0 = e % DHEAD;
                 const int src = min(qt + r, S - 1);
-                bf16x8 qq = load8(qp + (long)src * q_row + d0);
-                store8(Q_lds + r * QS2 + d0, qq);
-                bf16x8 dd = load8(dop + (long)src * q_row + d0);
+                bf16x8 dd = load8(dop + (long)src * o_row + d0);
                 store8(dO_lds + r * QS2 + d0, dd);
-                // tr images (conflict-free store8; read via ds_read_b64_tr_b16)
-                const int io = (d0 >> 4) * VSUB + v_img_row(r) * 16 + (d0 & 15);
-                store8(Q_img + io, qq);
-                store8(dO_img + io, dd);
+                store8(dO_img + (d0 >> 4) * VSUB + v_img_row(r) * 16 + (d0 & 15),
+                       dd);
             }
             for (int r = tid; r < BW_QT; r += FA_BLOCK) {
                 const int src = min(qt + r, S - 1);
@@ -397,10 +442,13 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
             f32x4 s0 = f32x4{0, 0, 0, 0};
             f32x4 dp0 = f32x4{0, 0, 0, 0};
 #pragma unroll
-            for (int kc = 0; kc < 4; ++kc) {
-                bf16x8v a_qf = ld_frag(Q_lds + (qs * 16 + col) * QS2 + kc * 32 + kgrp * 8);
-                bf16x8v b0 = ld_frag(K_l + col * KS + kc * 32 + kgrp * 8);
+            for (int kc = 0; kc < KC; ++kc) {
+                bf16x8v a_qf = ld_frag(Q_lds + (qs * 16 + col) * QS_T + kc * 32 + kgrp * 8);
+                bf16x8v b0 = ld_frag(K_l + col * KS_T + kc * 32 + kgrp * 8);
                 s0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_qf, b0, s0, 0, 0, 0);
+            }
+#pragma unroll
+            for (int kc = 0; kc < 4; ++kc) {
                 bf16x8v a_do = ld_frag(dO_lds + (qs * 16 + col) * QS2 + kc * 32 + kgrp * 8);
                 dp0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do, b_v[kc], dp0, 0, 0, 0);
             }
@@ -418,7 +466,7 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
                 dSt_l[col * TS2 + qrow] = f2bf(ds0);
             }
         }
-        // dV += P^T @ dO ; dK += dS^T @ Q(unscaled via Qt)
+        // dV += P^T @ dO ; dK += dS^T @ Q(unscaled via Q_img)
         {
             bf16x8v a_pt = ld_frag(Pt_l + col * TS2 + kgrp * 8);
             bf16x8v a_dst = ld_frag(dSt_l + col * TS2 + kgrp * 8);
@@ -427,6 +475,9 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
                 bf16x8v b_do = tr_frag(dO_img + dt * VSUB, lane);
                 dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     a_pt, b_do, dv_acc[dt], 0, 0, 0);
+            }
+#pragma unroll
+            for (int dt = 0; dt < QSUB; ++dt) {
                 bf16x8v b_q = tr_frag(Q_img + dt * VSUB, lane);
                 dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                     a_dst, b_q, dk_acc[dt], 0, 0, 0);
@@ -435,23 +486,24 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
         __syncthreads();
     }
 
+    // dkv layout: [B, S, Hkv, DQK + DHEAD] fp32 (dk then dv)
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
         const int krow = kvbase + kgrp * 4 + j;
         if (krow >= S) continue;
-        const long base = (((long)b * S + krow) * Hkv + hkv) * 2 * DHEAD;
+        const long base = (((long)b * S + krow) * Hkv + hkv) * (DQK + DHEAD);
 #pragma unroll
-        for (int dt = 0; dt < 8; ++dt) {
+        for (int dt = 0; dt < QSUB; ++dt)
             atomicAdd(&dkv[base + dt * 16 + col], dk_acc[dt][j]);
-            atomicAdd(&dkv[base + DHEAD + dt * 16 + col], dv_acc[dt][j]);
-        }
+#pragma unroll
+        for (int dt = 0; dt < 8; ++dt)
+            atomicAdd(&dkv[base + DQK + dt * 16 + col], dv_acc[dt][j]);
     }
 }
 
 // ---- dQ: q-parallel, fwd-like; K/V staged double-buffered + K tr-image ----
-#define DQ_BUF_ELEMS (2 * KVTILE * KS + 8 * VSUB)
-
-extern "C" __global__ __launch_bounds__(FA_BLOCK, 2)
+template <int DQK, int NQS>
+__global__ __launch_bounds__(FA_BLOCK, 2)
 void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
                       const bf16raw *__restrict__ q,
                       const bf16raw *__restrict__ k,
@@ -460,9 +512,14 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
                       const float *__restrict__ Dsum,
                       bf16raw *__restrict__ dq,
                       int B, int S, int Hq, int Hkv, int causal) {
+    constexpr int KC = DQK / 32;
+    constexpr int KS_T = DQK + 8;
+    constexpr int QSUB = DQK / 16;
+    constexpr int QT = NQS * 16;
+    constexpr int BUF = KVTILE * KS_T + KVTILE * KS + QSUB * VSUB;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     bf16raw *buf0 = reinterpret_cast<bf16raw *>(smem);
-    bf16raw *dS_all = buf0 + 2 * DQ_BUF_ELEMS;
+    bf16raw *dS_all = buf0 + 2 * BUF;
 
     const int qblk = blockIdx.x;
     const int h = blockIdx.y;
@@ -472,26 +529,28 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
     const int lane = threadIdx.x % WAVE;
     const int col = lane & 15;
     const int kgrp = lane >> 4;
-    const int qbase = qblk * (WAVES * QTILE) + wave * QTILE;
-    const float scale = rsqrtf((float)DHEAD);
+    const int qbase = qblk * (WAVES * QT) + wave * QT;
+    const float scale = rsqrtf((float)DQK);
 
-    const long q_row = (long)Hq * DHEAD;
-    const long kv_row = (long)Hkv * DHEAD;
-    const bf16raw *qp = q + ((long)b * S) * q_row + (long)h * DHEAD;
-    const bf16raw *kp = k + ((long)b * S) * kv_row + (long)hkv * DHEAD;
-    const bf16raw *vp = v + ((long)b * S) * kv_row + (long)hkv * DHEAD;
-    const bf16raw *dop = dout + ((long)b * S) * q_row + (long)h * DHEAD;
-    bf16raw *dSw = dS_all + wave * QTILE * PS;
+    const long q_row = (long)Hq * DQK;
+    const long k_row = (long)Hkv * DQK;
+    const long v_row = (long)Hkv * DHEAD;
+    const long o_row = (long)Hq * DHEAD;
+    const bf16raw *qp = q + ((long)b * S) * q_row + (long)h * DQK;
+    const bf16raw *kp = k + ((long)b * S) * k_row + (long)hkv * DQK;
+    const bf16raw *vp = v + ((long)b * S) * v_row + (long)hkv * DHEAD;
+    const bf16raw *dop = dout + ((long)b * S) * o_row + (long)h * DHEAD;
+    bf16raw *dSw = dS_all + wave * QT * PS;
 
     // per-wave q-side registers: scaled Q frags, lse, D; dO fragments are
     // re-read from global per tile (the WG's dO tile stays L2-resident)
-    bf16x8v a_q[2][4];
-    float lse_r[2][4], D_r[2][4];
+    bf16x8v a_q[NQS][KC];
+    float lse_r[NQS][4], D_r[NQS][4];
 #pragma unroll
-    for (int qs = 0; qs < 2; ++qs) {
+    for (int qs = 0; qs < NQS; ++qs) {
         const int qrow = qbase + qs * 16 + col;
 #pragma unroll
-        for (int kc = 0; kc < 4; ++kc) {
+        for (int kc = 0; kc < KC; ++kc) {
             bf16x8 raw = load8(qp + (long)min(qrow, S - 1) * q_row + kc * 32 + kgrp * 8);
             bf16x8 sc;
 #pragma unroll
@@ -507,45 +566,57 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
         }
     }
 
-    f32x4 dq_acc[2][8];
+    f32x4 dq_acc[NQS][QSUB];
 #pragma unroll
-    for (int qs = 0; qs < 2; ++qs)
+    for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
-        for (int dt = 0; dt < 8; ++dt) dq_acc[qs][dt] = f32x4{0, 0, 0, 0};
+        for (int dt = 0; dt < QSUB; ++dt) dq_acc[qs][dt] = f32x4{0, 0, 0, 0};
 
-    const int blk_rows = WAVES * QTILE;
+    const int blk_rows = WAVES * QT;
     const int kv_limit = causal ? min(S, qblk * blk_rows + blk_rows) : S;
     const int n_tiles = CDIV(kv_limit, KVTILE);
 
-    bf16x8 st_k[2], st_v[2];
+    constexpr int KCHUNKS = KVTILE * DQK / (FA_BLOCK * 8);
+    bf16x8 st_k[KCHUNKS], st_v[2];
     auto stage_load = [&](int t) {
         if (t >= n_tiles) return;
         const int kv = t * KVTILE;
         const int tid = threadIdx.x;
 #pragma unroll
-        for (int c = 0; c < 2; ++c) {
+        for (int c = 0; c < KCHUNKS; ++c) {
             const int e = tid * 8 + c * FA_BLOCK * 8;
-            const int kvr = e / DHEAD, d0 = e % DHEAD;
+            const int kvr = e / DQK, d0 = e % DQK;
             const int src = min(kv + kvr, S - 1);
-            st_k[c] = load8(kp + (long)src * kv_row + d0);
-            st_v[c] = load8(vp + (long)src * kv_row + d0);
+            st_k[c] = load8(kp + (long)src * k_row + d0);
         }
-    };
-    auto stage_write = [&](int t) {
-        if (t >= n_tiles) return;
-        bf16raw *K_lds = buf0 + (t & 1) * DQ_BUF_ELEMS;      // raw K rows
-        bf16raw *V_lds = K_lds + KVTILE * KS;                // V rows
-        bf16raw *K_img = V_lds + KVTILE * KS;                // tr image (raw)
-        const int tid = threadIdx.x;
 #pragma unroll
         for (int c = 0; c < 2; ++c) {
             const int e = tid * 8 + c * FA_BLOCK * 8;
             const int kvr = e / DHEAD, d0 = e % DHEAD;
+            const int src = min(kv + kvr, S - 1);
+            st_v[c] = load8(vp + (long)src * v_row + d0);
+        }
+    };
+    auto stage_write = [&](int t) {
+        if (t >= n_tiles) return;
+        bf16raw *K_lds = buf0 + (t & 1) * BUF;               // raw K rows
+        bf16raw *V_lds = K_lds + KVTILE * KS_T;              // V rows [32][KS]
+        bf16raw *K_img = V_lds + KVTILE * KS;                // tr image (raw)
+        const int tid = threadIdx.x;
+#pragma unroll
+        for (int c = 0; c < KCHUNKS; ++c) {
+            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int kvr = e / DQK, d0 = e % DQK;
             // K kept raw: S reuses the pre-scaled Q fragments (a_q)
-            store8(K_lds + kvr * KS + d0, st_k[c]);
-            store8(V_lds + kvr * KS + d0, st_v[c]);
+            store8(K_lds + kvr * KS_T + d0, st_k[c]);
             store8(K_img + (d0 >> 4) * VSUB + v_img_row(kvr) * 16 + (d0 & 15),
                    st_k[c]);
+        }
+#pragma unroll
+        for (int c = 0; c < 2; ++c) {
+            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int kvr = e / DHEAD, d0 = e % DHEAD;
+            store8(V_lds + kvr * KS + d0, st_v[c]);
         }
     };
 
@@ -556,24 +627,27 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
 
     for (int t = 0; t < n_tiles; ++t) {
         const int kv = t * KVTILE;
-        bf16raw *K_lds = buf0 + (t & 1) * DQ_BUF_ELEMS;
-        bf16raw *V_lds = K_lds + KVTILE * KS;
+        bf16raw *K_lds = buf0 + (t & 1) * BUF;
+        bf16raw *V_lds = K_lds + KVTILE * KS_T;
         bf16raw *K_img = V_lds + KVTILE * KS;
 
-        if (!(causal && kv >= qbase + QTILE)) {
+        if (!(causal && kv >= qbase + QT)) {
 #pragma unroll
-            for (int qs = 0; qs < 2; ++qs) {
+            for (int qs = 0; qs < NQS; ++qs) {
                 f32x4 s0 = f32x4{0, 0, 0, 0}, s1 = f32x4{0, 0, 0, 0};
                 f32x4 dp0 = f32x4{0, 0, 0, 0}, dp1 = f32x4{0, 0, 0, 0};
 #pragma unroll
-                for (int kc = 0; kc < 4; ++kc) {
-                    bf16x8v bk0 = ld_frag(K_lds + col * KS + kc * 32 + kgrp * 8);
-                    bf16x8v bk1 = ld_frag(K_lds + (16 + col) * KS + kc * 32 + kgrp * 8);
+                for (int kc = 0; kc < KC; ++kc) {
+                    bf16x8v bk0 = ld_frag(K_lds + col * KS_T + kc * 32 + kgrp * 8);
+                    bf16x8v bk1 = ld_frag(K_lds + (16 + col) * KS_T + kc * 32 + kgrp * 8);
                     s0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], bk0, s0, 0, 0, 0);
                     s1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], bk1, s1, 0, 0, 0);
+                }
+#pragma unroll
+                for (int kc = 0; kc < 4; ++kc) {
                     bf16x8v bv0 = ld_frag(V_lds + col * KS + kc * 32 + kgrp * 8);
                     bf16x8v bv1 = ld_frag(V_lds + (16 + col) * KS + kc * 32 + kgrp * 8);
-                    bf16x8v a_do = ld_frag(dop + (long)min(qbase + qs * 16 + col, S - 1) * q_row + kc * 32 + kgrp * 8);
+                    bf16x8v a_do = ld_frag(dop + (long)min(qbase + qs * 16 + col, S - 1) * o_row + kc * 32 + kgrp * 8);
                     dp0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do, bv0, dp0, 0, 0, 0);
                     dp1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do, bv1, dp1, 0, 0, 0);
                 }
@@ -595,7 +669,7 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
                 }
                 bf16x8v a_ds = ld_frag(dSw + qs * 16 * PS + col * PS + kgrp * 8);
 #pragma unroll
-                for (int dt = 0; dt < 8; ++dt) {
+                for (int dt = 0; dt < QSUB; ++dt) {
                     bf16x8v b_k = tr_frag(K_img + dt * VSUB, lane);
                     dq_acc[qs][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                         a_ds, b_k, dq_acc[qs][dt], 0, 0, 0);
@@ -607,35 +681,51 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
         stage_load(t + 2);
     }
 
-    bf16raw *dqp = dq + ((long)b * S) * q_row + (long)h * DHEAD;
+    bf16raw *dqp = dq + ((long)b * S) * q_row + (long)h * DQK;
 #pragma unroll
-    for (int qs = 0; qs < 2; ++qs)
+    for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
             const int row = qbase + qs * 16 + kgrp * 4 + j;
             if (row >= S) continue;
 #pragma unroll
-            for (int dt = 0; dt < 8; ++dt)
+            for (int dt = 0; dt < QSUB; ++dt)
                 dqp[(long)row * q_row + dt * 16 + col] = f2bf(dq_acc[qs][dt][j]);
         }
 }
 
 extern "C" void fa_fwd_launch(const void *q, const void *k, const void *v,
                               void *o, void *lse, int B, int S, int Hq,
-                              int Hkv, int causal, hipStream_t stream) {
-    dim3 grid(CDIV(S, WAVES * QTILE), Hq, B);
-    size_t smem = (2 * FWD_BUF_ELEMS + WAVES * QTILE * PS) * sizeof(bf16raw);
-    hipLaunchKernelGGL(fa_fwd_kernel, grid, dim3(FA_BLOCK), smem, stream,
-                       (const bf16raw *)q, (const bf16raw *)k,
-                       (const bf16raw *)v, (bf16raw *)o, (float *)lse, B, S,
-                       Hq, Hkv, causal);
+                              int Hkv, int dqk, int causal,
+                              hipStream_t stream) {
+    if (dqk == 128) {
+        constexpr int QT = 32;
+        dim3 grid(CDIV(S, WAVES * QT), Hq, B);
+        size_t smem = (2 * (KVTILE * (128 + 8) + 8 * VSUB) + WAVES * QT * PS)
+                      * sizeof(bf16raw);
+        hipLaunchKernelGGL((fa_fwd_kernel<128, 2>), grid, dim3(FA_BLOCK),
+                           smem, stream, (const bf16raw *)q,
+                           (const bf16raw *)k, (const bf16raw *)v,
+                           (bf16raw *)o, (float *)lse, B, S, Hq, Hkv, causal);
+    } else if (dqk == 192) {
+        constexpr int QT = 16;
+        dim3 grid(CDIV(S, WAVES * QT), Hq, B);
+        size_t smem = (2 * (KVTILE * (192 + 8) + 8 * VSUB) + WAVES * QT * PS)
+                      * sizeof(bf16raw);
+        hipLaunchKernelGGL((fa_fwd_kernel<192, 1>), grid, dim3(FA_BLOCK),
+                           smem, stream, (const bf16raw *)q,
+                           (const bf16raw *)k, (const bf16raw *)v,
+                           (bf16raw *)o, (float *)lse, B, S, Hq, Hkv, causal);
+    } else {
+        abort();
+    }
 }
 
 extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
                               const void *v, const void *o, const void *lse,
                               void *dq_bf16, void *dkv, void *dsum,
-                              int B, int S, int Hq, int Hkv, int causal,
-                              hipStream_t stream) {
+                              int B, int S, int Hq, int Hkv, int dqk,
+                              int causal, hipStream_t stream) {
     // 1. D = rowsum(dO*O)
     {
         long rows = (long)B * Hq * S;
@@ -648,24 +738,55 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
     // 2. dK/dV (kv-parallel)
     {
         dim3 grid(CDIV(S, WAVES * BKV), Hq, B);
-        size_t q_side = (2 * BW_QT * QS2 + 2 * 8 * VSUB) * sizeof(bf16raw) +
-                        2 * BW_QT * sizeof(float);
-        size_t per_wave = (BKV * KS + 2 * BKV * TS2) * sizeof(bf16raw);
-        size_t smem = q_side + WAVES * per_wave;
-        hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid, dim3(FA_BLOCK), smem,
-                           stream, (const bf16raw *)dout, (const bf16raw *)q,
-                           (const bf16raw *)k, (const bf16raw *)v,
-                           (const float *)lse, (const float *)dsum,
-                           (float *)dkv, B, S, Hq, Hkv, causal);
+        size_t smem;
+        if (dqk == 128) {
+            smem = (BW_QT * (128 + 8) + BW_QT * QS2 + (128 / 16 + 8) * VSUB)
+                       * sizeof(bf16raw)
+                   + 2 * BW_QT * sizeof(float)
+                   + WAVES * (BKV * (128 + 8) + 2 * BKV * TS2) * sizeof(bf16raw);
+            hipLaunchKernelGGL((fa_bwd_dkv_kernel<128>), grid, dim3(FA_BLOCK),
+                               smem, stream, (const bf16raw *)dout,
+                               (const bf16raw *)q, (const bf16raw *)k,
+                               (const bf16raw *)v, (const float *)lse,
+                               (const float *)dsum, (float *)dkv, B, S, Hq,
+                               Hkv, causal);
+        } else if (dqk == 192) {
+            smem = (BW_QT * (192 + 8) + BW_QT * QS2 + (192 / 16 + 8) * VSUB)
+                       * sizeof(bf16raw)
+                   + 2 * BW_QT * sizeof(float)
+                   + WAVES * (BKV * (192 + 8) + 2 * BKV * TS2) * sizeof(bf16raw);
+            hipLaunchKernelGGL((fa_bwd_dkv_kernel<192>), grid, dim3(FA_BLOCK),
+                               smem, stream, (const bf16raw *)dout,
+                               (const bf16raw *)q, (const bf16raw *)k,
+                               (const bf16raw *)v, (const float *)lse,
+                               (const float *)dsum, (float *)dkv, B, S, Hq,
+                               Hkv, causal);
+        } else {
+            abort();
+        }
     }
     // 3. dQ (q-parallel)
-    {
-        dim3 grid(CDIV(S, WAVES * QTILE), Hq, B);
-        size_t smem = (2 * DQ_BUF_ELEMS + WAVES * QTILE * PS) * sizeof(bf16raw);
-        hipLaunchKernelGGL(fa_bwd_dq_kernel, grid, dim3(FA_BLOCK), smem,
-                           stream, (const bf16raw *)dout, (const bf16raw *)q,
-                           (const bf16raw *)k, (const bf16raw *)v,
-                           (const float *)lse, (const float *)dsum,
-                           (bf16raw *)dq_bf16, B, S, Hq, Hkv, causal);
+    if (dqk == 128) {
+        constexpr int QT = 32;
+        dim3 grid(CDIV(S, WAVES * QT), Hq, B);
+        size_t smem = (2 * (KVTILE * (128 + 8) + KVTILE * KS + 8 * VSUB)
+                       + WAVES * QT * PS) * sizeof(bf16raw);
+        hipLaunchKernelGGL((fa_bwd_dq_kernel<128, 2>), grid, dim3(FA_BLOCK),
+                           smem, stream, (const bf16raw *)dout,
+                           (const bf16raw *)q, (const bf16raw *)k,
+                           (const bf16raw *)v, (const float *)lse,
+                           (const float *)dsum, (bf16raw *)dq_bf16, B, S, Hq,
+                           Hkv, causal);
+    } else {
+        constexpr int QT = 16;
+        dim3 grid(CDIV(S, WAVES * QT), Hq, B);
+        size_t smem = (2 * (KVTILE * (192 + 8) + KVTILE * KS + 12 * VSUB)
+                       + WAVES * QT * PS) * sizeof(bf16raw);
+        hipLaunchKernelGGL((fa_bwd_dq_kernel<192, 1>), grid, dim3(FA_BLOCK),
+                           smem, stream, (const bf16raw *)dout,
+                           (const bf16raw *)q, (const bf16raw *)k,
+                           (const bf16raw *)v, (const float *)lse,
+                           (const float *)dsum, (bf16raw *)dq_bf16, B, S, Hq,
+                           Hkv, causal);
     }
 }

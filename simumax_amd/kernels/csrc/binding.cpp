@@ -134,10 +134,10 @@ extern "C" {
 void mfma_probe_launch(const void *, const void *, void *, void *, int,
                        hipStream_t);
 void fa_fwd_launch(const void *, const void *, const void *, void *, void *,
-                   int, int, int, int, int, hipStream_t);
+                   int, int, int, int, int, int, hipStream_t);
 void fa_bwd_launch(const void *, const void *, const void *, const void *,
                    const void *, const void *, void *, void *, void *, int,
-                   int, int, int, int, hipStream_t);
+                   int, int, int, int, int, hipStream_t);
 }
 
 std::vector<torch::Tensor> mfma_probe(torch::Tensor A, torch::Tensor B,
@@ -157,15 +157,17 @@ std::vector<torch::Tensor> fa_fwd(torch::Tensor q, torch::Tensor k,
     CHECK_IN(q);
     CHECK_IN(k);
     CHECK_IN(v);
-    TORCH_CHECK(q.dim() == 4, "q must be [B,S,Hq,D]");
-    const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
-    const int Hkv = k.size(2);
-    TORCH_CHECK(D == 128, "fa kernel supports D=128");
+    TORCH_CHECK(q.dim() == 4, "q must be [B,S,Hq,Dqk]");
+    const int B = q.size(0), S = q.size(1), Hq = q.size(2), Dqk = q.size(3);
+    const int Hkv = k.size(2), Dv = v.size(3);
+    TORCH_CHECK(Dqk == 128 || Dqk == 192, "fa kernel supports Dqk in {128,192}");
+    TORCH_CHECK(Dv == 128, "fa kernel supports Dv=128");
     TORCH_CHECK(Hq % Hkv == 0);
-    auto o = torch::empty_like(q);
+    auto o = torch::empty({B, S, Hq, Dv}, q.options());
     auto lse = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat32));
     fa_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                  lse.data_ptr(), B, S, Hq, Hkv, causal ? 1 : 0, cur_stream());
+                  lse.data_ptr(), B, S, Hq, Hkv, Dqk, causal ? 1 : 0,
+                  cur_stream());
     return {o, lse};
 }
 
@@ -175,17 +177,18 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
                                   bool causal) {
     CHECK_IN(dout);
     CHECK_IN(q);
-    const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
-    const int Hkv = k.size(2);
+    const int B = q.size(0), S = q.size(1), Hq = q.size(2), Dqk = q.size(3);
+    const int Hkv = k.size(2), Dv = v.size(3);
     auto f32 = q.options().dtype(torch::kFloat32);
-    auto dq = torch::empty({B, S, Hq, D}, q.options());
-    auto dkv = torch::zeros({B, S, Hkv, 2, D}, f32);
+    auto dq = torch::empty({B, S, Hq, Dqk}, q.options());
+    auto dkv = torch::zeros({B, S, Hkv, Dqk + Dv}, f32);
     auto dsum = torch::empty({B, Hq, S}, f32);
     fa_bwd_launch(dout.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                   o.data_ptr(), lse.data_ptr(), dq.data_ptr(), dkv.data_ptr(),
-                  dsum.data_ptr(), B, S, Hq, Hkv, causal ? 1 : 0, cur_stream());
-    auto dk = dkv.select(3, 0).to(torch::kBFloat16);
-    auto dv = dkv.select(3, 1).to(torch::kBFloat16);
+                  dsum.data_ptr(), B, S, Hq, Hkv, Dqk, causal ? 1 : 0,
+                  cur_stream());
+    auto dk = dkv.narrow(3, 0, Dqk).to(torch::kBFloat16);
+    auto dv = dkv.narrow(3, Dqk, Dv).to(torch::kBFloat16);
     return {dq, dk.contiguous(), dv.contiguous()};
 }
 

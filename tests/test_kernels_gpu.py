@@ -150,3 +150,28 @@ def test_trainer_smoke_and_loss_decreases():
               for _ in range(8)]
     assert all(l == l for l in losses), "NaN loss"
     assert losses[-1] < losses[0], f"loss did not decrease: {losses}"
+
+
+def test_flash_attention_mla_shapes():
+    """MLA asymmetric head dims: qk 192, v/o 128."""
+    B, S, H, Dqk, Dv = 1, 512, 16, 192, 128
+    torch.manual_seed(11)
+    q = torch.randn(B, S, H, Dqk, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, S, H, Dqk, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, S, H, Dv, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = K.flash_attention(q, k, v, causal=True)
+    oref = K._sdpa_torch(q.detach(), k.detach(), v.detach(), True)
+    assert relerr(o, oref) < 3e-2, f"mla fwd {relerr(o, oref)}"
+    do = torch.randn_like(o)
+    o.backward(do)
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    v2 = v.detach().float().requires_grad_(True)
+    o2 = K._sdpa_torch(q2, k2, v2, True)
+    o2.backward(do.float())
+    assert relerr(q.grad, q2.grad) < 5e-2, f"mla dq {relerr(q.grad, q2.grad)}"
+    assert relerr(k.grad, k2.grad) < 5e-2, f"mla dk {relerr(k.grad, k2.grad)}"
+    assert relerr(v.grad, v2.grad) < 5e-2, f"mla dv {relerr(v.grad, v2.grad)}"
