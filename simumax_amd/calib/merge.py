@@ -40,6 +40,13 @@ def main():
                 statistics.median(tab.values()), 4)
             print(f"{key}: {len(tab)} shapes, median eff "
                   f"{acc['op'][key]['efficient_factor']}")
+    fp8 = _load("fp8_matmul.json")
+    if fp8:
+        acc["op"]["fp8_matmul"]["accurate_efficient_factor"] = fp8
+        acc["op"]["fp8_matmul"]["efficient_factor"] = round(
+            statistics.median(fp8.values()), 4)
+        print(f"fp8_matmul: {len(fp8)} shapes, median eff "
+              f"{acc['op']['fp8_matmul']['efficient_factor']}")
     group = _load("group_matmul.json")
     if group:
         acc["op"]["group_matmul"]["accurate_efficient_factor"] = group

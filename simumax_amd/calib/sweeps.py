@@ -165,6 +165,48 @@ def sweep_sdp(shape_keys, fwd_path, bwd_path, sparse_ratio=0.5):
     return fwd_tab, bwd_tab
 
 
+def time_fp8_gemm(desc, device="cuda"):
+    """fp8 e4m3 GEMM via torch._scaled_mm (hipBLASLt fp8 path on gfx950);
+    NT-accumulate wgrad shapes are skipped (grads stay bf16 on this stack)."""
+    b, m, k, n, layout, accumulate, out_dtype = parse_gemm_key(desc)
+    if layout == "NT" or b != 1:
+        return None
+    f8 = torch.float8_e4m3fn
+    if layout == "TN":     # x[m,k] @ w[n,k]^T
+        a = torch.randn(m, k, device=device).to(f8)
+        w = torch.randn(n, k, device=device).to(f8)
+        bmat = w.t()
+    else:                  # NN: d[m,k] @ w[k,n]
+        a = torch.randn(m, k, device=device).to(f8)
+        bmat = torch.randn(n, k, device=device).to(f8).t()  # column-major [k,n]
+    sa = torch.ones((), device=device)
+    sb = torch.ones((), device=device)
+    fn = lambda: torch._scaled_mm(a, bmat, scale_a=sa, scale_b=sb,
+                                  out_dtype=torch.bfloat16)
+    t_ms = _timeit(fn, iters=10)
+    flops = 2 * m * k * n
+    return flops / (t_ms / 1e3) / (2 * PEAK_BF16), t_ms  # vs 5 PF fp8 peak
+
+
+def sweep_fp8_gemms(shape_keys, path):
+    table = _load(path)
+    for desc in shape_keys:
+        if desc in table and not OVERWRITE:
+            continue
+        try:
+            r = time_fp8_gemm(desc)
+        except (RuntimeError, torch.cuda.OutOfMemoryError) as e:
+            print(f"[fp8] skip {desc}: {str(e)[:80]}", flush=True)
+            torch.cuda.empty_cache()
+            continue
+        if r is None:
+            continue
+        table[desc] = r[0]
+        print(f"[fp8] {desc} -> eff {r[0]:.4f} ({r[1]:.3f} ms)", flush=True)
+        _save(path, table)
+    return table
+
+
 def parse_group_key(desc):
     m = re.match(r"ng=(\d+), M=(\d+), N=(\d+), K=(\d+), dtype=(\w+)", desc)
     assert m, desc
@@ -349,6 +391,8 @@ def main():
                   os.path.join(OUT_DIR, "sdp_bwd.json"))
     if which in ("all", "group"):
         sweep_grouped(group_keys, os.path.join(OUT_DIR, "group_matmul.json"))
+    if which in ("all", "fp8"):
+        sweep_fp8_gemms(gemm_keys, os.path.join(OUT_DIR, "fp8_matmul.json"))
     print("[calib] done", flush=True)
 
 
