@@ -117,6 +117,8 @@ class PpSchedule:
 
     def build(self) -> Dict[int, List[Job]]:
         s = self.perf.strategy
+        if max(1, s.interleaving_size) > 1:
+            return self.build_interleaved()
         pp, mbc = s.pp_size, s.micro_batch_num
         p2p_size = get_pp_p2p_comm_size(s, self.perf.model_config)
         p2p_time = 0.0
@@ -173,7 +175,78 @@ class PpSchedule:
                                 dur=p2p_time, lane="comm", mb=m,
                                 gid=f"p2p-b-mb{m}-{stage}-{stage-1}",
                                 peers=(r, per_stage_rank[stage - 1])))
+        self._append_optimizer(jobs, ranks, per_stage_rank)
+        return jobs
+
+    def build_interleaved(self) -> Dict[int, List[Job]]:
+        """Sync-VPP replay: Megatron interleaved schedule table over the
+        per-virtual-chunk models (reference parity:
+        pipeline_schedule.py:97-715)."""
+        from ..perf.vpp import chunk_id_of, mb_id_of
+
+        s = self.perf.strategy
+        pp, mbc = s.pp_size, s.micro_batch_num
+        vp = s.interleaving_size
+        assert self.perf.vchunks is not None
+        p2p_size = get_pp_p2p_comm_size(s, self.perf.model_config)
+        p2p_time = self.system.compute_net_op_time(
+            "p2p", p2p_size, 2, net=s.pp_net, comm_stage="pp", strategy=s)
+        ranks = self.sim_ranks()
+        jobs: Dict[int, List[Job]] = {r: [] for r in ranks}
+        per_stage_rank = {i: get_pp_stage_representative_rank(i, s)
+                          for i in range(pp)}
+        nv = pp * vp
+        total = mbc * vp
+        for stage in range(pp):
+            r = per_stage_rank[stage]
+            warm = min((pp - stage - 1) * 2 + (vp - 1) * pp, total)
+            stream = [("F", k) for k in range(warm)]
+            nf, nb = warm, 0
+            while nb < total:
+                if nf < total:
+                    stream.append(("F", nf)); nf += 1
+                stream.append(("B", nb)); nb += 1
+            for kind, k in stream:
+                fwd = kind == "F"
+                c = chunk_id_of(k, pp, vp, fwd)
+                m = mb_id_of(k, pp, vp)
+                v = c * pp + stage
+                chunk = self.perf.vchunks[stage][c]
+                if fwd:
+                    if v > 0:
+                        src = per_stage_rank[stage - 1 if stage > 0 else pp - 1]
+                        jobs[r].append(Job(
+                            name=f"recv_fwd.v{v}.mb{m}", kind="p2p",
+                            dur=p2p_time, lane="comm", mb=m,
+                            gid=f"p2p-f-mb{m}-v{v-1}-v{v}", peers=(src, r)))
+                    jobs[r].extend(_leaf_jobs_fwd(chunk, stage, m * vp + c))
+                    if v < nv - 1:
+                        dst = per_stage_rank[stage + 1 if stage < pp - 1 else 0]
+                        jobs[r].append(Job(
+                            name=f"send_fwd.v{v}.mb{m}", kind="p2p",
+                            dur=p2p_time, lane="comm", mb=m,
+                            gid=f"p2p-f-mb{m}-v{v}-v{v+1}", peers=(r, dst)))
+                else:
+                    if v < nv - 1:
+                        src = per_stage_rank[stage + 1 if stage < pp - 1 else 0]
+                        jobs[r].append(Job(
+                            name=f"recv_bwd.v{v}.mb{m}", kind="p2p",
+                            dur=p2p_time, lane="comm", mb=m,
+                            gid=f"p2p-b-mb{m}-v{v+1}-v{v}", peers=(src, r)))
+                    jobs[r].extend(_leaf_jobs_bwd(chunk, stage, m * vp + c))
+                    if v > 0:
+                        dst = per_stage_rank[stage - 1 if stage > 0 else pp - 1]
+                        jobs[r].append(Job(
+                            name=f"send_bwd.v{v}.mb{m}", kind="p2p",
+                            dur=p2p_time, lane="comm", mb=m,
+                            gid=f"p2p-b-mb{m}-v{v}-v{v-1}", peers=(r, dst)))
+        self._append_optimizer(jobs, ranks, per_stage_rank)
+        return jobs
+
+    def _append_optimizer(self, jobs, ranks, per_stage_rank):
+        s = self.perf.strategy
         # optimizer tail (DP collectives + adam traffic)
+        pp = s.pp_size
         for stage in range(pp):
             for r in ([per_stage_rank[stage]] if self.merge_lanes else
                       [x for x in ranks
@@ -191,4 +264,3 @@ class PpSchedule:
                 jobs[r].append(Job(name="optimizer.adam", kind="optim",
                                    dur=self.perf._compute_optim_time(stage),
                                    mb=-1))
-        return jobs

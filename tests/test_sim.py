@@ -104,3 +104,17 @@ def test_export_analytic_schedule_trace(tmp_path):
         p.schedule_records, cost["chunk_fwd_times"], cost["chunk_bwd_times"],
         str(tmp_path / "sched.json"))
     assert len(out["traceEvents"]) == 2 * p.strategy.pp_size * p.strategy.micro_batch_num
+
+
+def test_sim_interleaved_vpp(tmp_path):
+    p = build(strategy="tp1_pp4_vp2_sync_mbs1_mbc8", model="llama3-8b")
+    # shrink for test speed
+    cost = p.analysis_cost()
+    res = p.simulate(str(tmp_path))
+    # simulated total within 15% of the exact analytic interleaved schedule
+    analytic = (cost["pipeline_time"] / cost["straggler_ratio"]
+                + cost["dp_time"] + cost["optim_time"])
+    assert res["total_time"] == pytest.approx(analytic, rel=0.15)
+    tr = json.loads((tmp_path / "tracing_logs.json").read_text())
+    names = {e.get("name", "") for e in tr["traceEvents"]}
+    assert any("chunk1" in n for n in names)  # second virtual chunk ran
