@@ -162,10 +162,14 @@ class LinearCol(LinearBase, ParamMixin):
     def _leaf_act_info(self, info):
         # caches the (sharded under SP) input; the gathered copy is transient
         info.activation_mem_cache = self.input_info.first.mem_bytes()
+        # bwd transient: autograd's clone of the shared dummy wgrad
+        # (freed by the post-accumulate hook right after the wgrad GEMM)
+        info.bwd_peak_mem_no_cache = (
+            self.input_size * self.output_size * self.element_size)
         if self.sp:
             gathered = self.input_info.first.mem_bytes() * self.tp
-            info.fwd_peak_mem_no_cache = gathered
-            info.bwd_peak_mem_no_cache = gathered
+            info.fwd_peak_mem_no_cache = max(info.fwd_peak_mem_no_cache, gathered)
+            info.bwd_peak_mem_no_cache += gathered
 
     def _leaf_compute_info(self, info):
         k = self.get_gemm_bmnk("fwd")
@@ -231,6 +235,8 @@ class LinearRow(LinearBase, ParamMixin):
 
     def _leaf_act_info(self, info):
         info.activation_mem_cache = self.input_info.first.mem_bytes()
+        info.bwd_peak_mem_no_cache = (
+            self.input_size * self.output_size * self.element_size)
         # full (pre-scatter) output is transient under SP
         if self.sp:
             full_out = self.output_info.first.mem_bytes() * self.tp

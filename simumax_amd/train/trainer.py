@@ -123,26 +123,30 @@ class DataParallelGradReducer:
         for p in self.params:
             offs[id(p)] = (off, off + p.numel())
             off += p.numel()
-        if self.overlap:
-            # reverse order (grads become ready back-to-front); reversed
-            # consecutive params are a contiguous flat slice
-            buckets, cur, cur_bytes = [], [], 0
-            for p in reversed(self.params):
-                cur.append(p)
-                cur_bytes += p.numel() * 4
-                if cur_bytes >= bucket_bytes:
-                    buckets.append(cur)
-                    cur, cur_bytes = [], 0
-            if cur:
+        # Hooks are registered UNCONDITIONALLY: releasing p.grad right after
+        # each accumulation is a memory-correctness requirement — autograd
+        # CLONES the shared dummy wgrad (its use_count is too high to
+        # steal), and without the hook those weight-sized clones pile up in
+        # p.grad until backward ends (~25 GiB on llama3-70b-l12).
+        # reverse order (grads become ready back-to-front); reversed
+        # consecutive params are a contiguous flat slice
+        buckets, cur, cur_bytes = [], [], 0
+        for p in reversed(self.params):
+            cur.append(p)
+            cur_bytes += p.numel() * 4
+            if cur_bytes >= bucket_bytes:
                 buckets.append(cur)
-            for bucket in buckets:
-                lo = min(offs[id(p)][0] for p in bucket)
-                hi = max(offs[id(p)][1] for p in bucket)
-                remaining = {id(p) for p in bucket}
-                for p in bucket:
-                    p.register_post_accumulate_grad_hook(
-                        self._make_hook((lo, hi), remaining))
-            self._buckets = buckets
+                cur, cur_bytes = [], 0
+        if cur:
+            buckets.append(cur)
+        for bucket in buckets:
+            lo = min(offs[id(p)][0] for p in bucket)
+            hi = max(offs[id(p)][1] for p in bucket)
+            remaining = {id(p) for p in bucket}
+            for p in bucket:
+                p.register_post_accumulate_grad_hook(
+                    self._make_hook((lo, hi), remaining))
+        self._buckets = buckets
 
     def _make_hook(self, span, remaining):
         def hook(p):
@@ -151,7 +155,7 @@ class DataParallelGradReducer:
             p.grad = None
             remaining.discard(id(p))
             if not remaining:
-                if self.reduce_this_pass:
+                if self.overlap and self.reduce_this_pass:
                     sl = self.flat_grad[span[0]:span[1]]
                     sl.div_(dist.get_world_size())
                     self.handles.append(dist.all_reduce(sl, async_op=True))
@@ -207,9 +211,7 @@ def train_step(model, opt, reducer, tokens, labels, micro_batch_num=1):
     for mb in range(micro_batch_num):
         reducer.reduce_this_pass = mb == micro_batch_num - 1
         loss = model(tokens[mb], labels[mb])
-        loss.backward()
-        if not reducer.overlap:
-            accumulate_main_grads(opt.params)
+        loss.backward()   # per-param hooks accumulate + free grads inline
         total_loss += loss.item()
     reducer.finalize()
     opt.step()
