@@ -84,7 +84,8 @@ def run_case(name, model, seq, mbs, mbc, steps, warmup=1):
     print(json.dumps(row), flush=True)
     with open(OUT, "a") as f:
         f.write(json.dumps(row) + "\n")
-    # free everything before the next case
+    # free everything before the next case (hooks pin the model otherwise)
+    red.remove_hooks()
     del m, opt, red, toks, labels
     gc.collect()
     torch.cuda.empty_cache()

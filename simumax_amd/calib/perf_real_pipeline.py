@@ -98,6 +98,7 @@ def real_run(case: dict) -> dict:
     ms = (time.time() - t0) / case["steps"] * 1e3
     out = dict(measured_ms=ms,
                measured_bytes=float(torch.cuda.max_memory_allocated()))
+    red.remove_hooks()
     del model, opt, red, toks, labels
     import gc
 

@@ -139,14 +139,24 @@ class DataParallelGradReducer:
                 cur, cur_bytes = [], 0
         if cur:
             buckets.append(cur)
+        self._hook_handles = []
         for bucket in buckets:
             lo = min(offs[id(p)][0] for p in bucket)
             hi = max(offs[id(p)][1] for p in bucket)
             remaining = {id(p) for p in bucket}
             for p in bucket:
-                p.register_post_accumulate_grad_hook(
-                    self._make_hook((lo, hi), remaining))
+                self._hook_handles.append(p.register_post_accumulate_grad_hook(
+                    self._make_hook((lo, hi), remaining)))
         self._buckets = buckets
+
+    def remove_hooks(self):
+        """Detach the post-accumulate hooks. The hook closures reference the
+        reducer (and through it every parameter) from C++-side hook storage
+        that the cyclic GC cannot traverse — a trainer is only freed after
+        calling this."""
+        for h in self._hook_handles:
+            h.remove()
+        self._hook_handles.clear()
 
     def _make_hook(self, span, remaining):
         def hook(p):
