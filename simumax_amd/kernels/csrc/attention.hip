@@ -17,6 +17,7 @@
 // Reference behavior target: sdp_fwd/sdp_bwd priced by
 // simumax/core/transformer/dense_module.py:1061-1605.
 #include "common.h"
+#include <cstdlib>
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
@@ -699,11 +700,12 @@ extern "C" void fa_fwd_launch(const void *q, const void *k, const void *v,
                               int Hkv, int dqk, int causal,
                               hipStream_t stream) {
     if (dqk == 128) {
-        constexpr int QT = 32;
+        // NQS=1 -> 3 waves/SIMD: measured +18-30% over NQS=2
+        constexpr int QT = 16;
         dim3 grid(CDIV(S, WAVES * QT), Hq, B);
         size_t smem = (2 * (KVTILE * (128 + 8) + 8 * VSUB) + WAVES * QT * PS)
                       * sizeof(bf16raw);
-        hipLaunchKernelGGL((fa_fwd_kernel<128, 2>), grid, dim3(FA_BLOCK),
+        hipLaunchKernelGGL((fa_fwd_kernel<128, 1>), grid, dim3(FA_BLOCK),
                            smem, stream, (const bf16raw *)q,
                            (const bf16raw *)k, (const bf16raw *)v,
                            (bf16raw *)o, (float *)lse, B, S, Hq, Hkv, causal);
@@ -767,11 +769,11 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
     }
     // 3. dQ (q-parallel)
     if (dqk == 128) {
-        constexpr int QT = 32;
+        constexpr int QT = 16;   // NQS=1: occupancy over per-wave work
         dim3 grid(CDIV(S, WAVES * QT), Hq, B);
         size_t smem = (2 * (KVTILE * (128 + 8) + KVTILE * KS + 8 * VSUB)
                        + WAVES * QT * PS) * sizeof(bf16raw);
-        hipLaunchKernelGGL((fa_bwd_dq_kernel<128, 2>), grid, dim3(FA_BLOCK),
+        hipLaunchKernelGGL((fa_bwd_dq_kernel<128, 1>), grid, dim3(FA_BLOCK),
                            smem, stream, (const bf16raw *)dout,
                            (const bf16raw *)q, (const bf16raw *)k,
                            (const bf16raw *)v, (const float *)lse,
