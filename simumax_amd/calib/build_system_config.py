@@ -11,7 +11,6 @@ Usage (8-GPU node):   python -m torch.distributed.run --standalone \
 """
 
 import os
-import subprocess
 import sys
 
 

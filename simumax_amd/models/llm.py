@@ -14,12 +14,11 @@ budget the strategy search sizes against.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 from ..core.module import MetaModule
 from ..core.records import InputOutputInfo
-from ..core.tensor import TensorSize
 from ..ops.dense import (
     Add,
     Attention,

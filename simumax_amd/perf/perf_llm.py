@@ -20,7 +20,7 @@ import json
 import math
 import os
 from copy import deepcopy
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 from ..core.config import ModelConfig, StrategyConfig, SystemConfig

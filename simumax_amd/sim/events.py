@@ -13,7 +13,7 @@ and sim.memory (allocator timeline).
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 

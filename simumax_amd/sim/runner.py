@@ -8,7 +8,6 @@ simu_memory_snapshot.json, simu_memory_viz_snapshot.pickle.
 from __future__ import annotations
 
 import os
-from typing import Dict
 
 from ..core.records import Result
 from .events import SimuSystem

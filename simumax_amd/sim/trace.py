@@ -11,7 +11,6 @@ wrapper {"traceEvents": [...], "displayTimeUnit": "ms"}.
 from __future__ import annotations
 
 import json
-from typing import List
 
 _LANE_ORDER = ["fwd compute", "bwd compute", "recompute", "optimizer",
                "comm", "wait", "pp"]

@@ -11,7 +11,6 @@ simulator's per-leaf cache accounting (simumax_amd/ops/dense.py docstring).
 
 from __future__ import annotations
 
-import math
 
 import torch
 import torch.nn as nn  # noqa: F401 (Embedding)

@@ -14,10 +14,7 @@ with zero_state=0 for the same strategy config.
 
 from __future__ import annotations
 
-import os
-import time
 from dataclasses import dataclass
-from typing import Optional
 
 import torch
 import torch.distributed as dist
