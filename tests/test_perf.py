@@ -247,3 +247,29 @@ def test_debug_points_cost_log(tmp_path, monkeypatch):
     log = _json.loads((tmp_path / "cost_log.json").read_text())
     assert "stage0.layer0.mlp.fc1" in log
     assert log["stage0.layer0.mlp.fc1"]["cost_F"] > 0
+
+
+def test_fast_estimator_matches_exact():
+    """FastEstimator (layer-profile cache) ranks within ~10% of the exact
+    estimate and reuses profiles across pp values."""
+    from simumax_amd.tuning.profile_cache import FastEstimator
+
+    mc = ModelConfig.init_from_config_file(get_simu_model_config("llama3-8b"))
+    sysc = SystemConfig.init_from_config_file(get_simu_system_config("mi355x"))
+    fe = FastEstimator(mc, sysc)
+    for pp in (1, 2, 4):
+        st = StrategyConfig.init_from_config_file(
+            get_simu_strategy_config("tp1_pp2_dp4_mbs1"))
+        st.pp_size = pp
+        st.micro_batch_num = 8
+        fast = fe.estimate(st)
+        exact = PerfLLM()
+        exact.configure(st, mc, sysc)
+        exact.run_estimate()
+        cost = exact.analysis_cost()
+        mem = exact.analysis_mem()
+        assert fast["iter_time"] == pytest.approx(cost["iter_time"], rel=0.12), pp
+        assert fast["peak_mem"] == pytest.approx(mem["max_peak_mem"], rel=0.15), pp
+    # pp sweep shares ONE profile
+    assert fe.cache.misses == 1
+    assert fe.cache.hits == 2

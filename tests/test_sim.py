@@ -118,3 +118,19 @@ def test_sim_interleaved_vpp(tmp_path):
     tr = json.loads((tmp_path / "tracing_logs.json").read_text())
     names = {e.get("name", "") for e in tr["traceEvents"]}
     assert any("chunk1" in n for n in names)  # second virtual chunk ran
+
+
+def test_sim_all_world_ranks(tmp_path):
+    """merge_lanes=False: one simulated lane per world rank."""
+    p = build()
+    res = p.simulate(str(tmp_path), merge_lanes=False)
+    assert len(res["ranks"]) == p.strategy.world_size
+    tr = json.loads((tmp_path / "tracing_logs.json").read_text())
+    pids = {e["pid"] for e in tr["traceEvents"] if e.get("ph") == "X"}
+    assert len(pids) == 8
+    # per-stage peaks must agree across same-stage ranks
+    per_stage = p.strategy.world_size // p.strategy.pp_size
+    for stage in range(p.strategy.pp_size):
+        peaks = [res["peak_mem"][r] for r in res["ranks"]
+                 if r // per_stage == stage]
+        assert max(peaks) - min(peaks) < 1.0
