@@ -26,6 +26,7 @@ CASES = [
     ("8b_seq2048_mbs2", "llama3-8b", 2048, 2, 4, 3),
     ("8b_seq8192_mbc2", "llama3-8b", 8192, 1, 2, 3),
     ("70b_l12_seq4096", "llama3-70b-l12", 4096, 1, 2, 3),
+    ("mixtral_l8_moe", "mixtral-8x7b-l8", 4096, 1, 2, 3),
 ]
 
 OUT = "gpurun_out/validation.jsonl"

@@ -371,6 +371,8 @@ DEFAULT_CASES = [
     ("llama3-70b-l12", dict(tp_size=2, pp_size=2, world_size=8)),
     ("llama3-70b-l12", dict(tp_size=8)),
     ("deepseekv2-l4", dict(ep_size=8, enable_sequence_parallel=False)),
+    ("mixtral-8x7b-l8", dict(world_size=1, tp_size=1, pp_size=1, ep_size=1,
+                             enable_sequence_parallel=False, zero_state=0)),
     ("deepseekv2-l4", dict(ep_size=4, pp_size=2,
                            enable_sequence_parallel=False)),
 ]

@@ -33,8 +33,14 @@ class LlamaDecoderLayer(nn.Module):
                                       dtype=dtype, device=device)
         self.pre_mlp_norm = K.RMSNorm(h, dtype=dtype, device=device)
         assert cfg.use_swiglu
-        self.fc1 = K.FusedLinear(h, 2 * cfg.intermediate_size, dtype=dtype, device=device)
-        self.fc2 = K.FusedLinear(cfg.intermediate_size, h, dtype=dtype, device=device)
+        self.use_moe = cfg.model_type == "moe"
+        if self.use_moe:
+            from .moe import MoEMLP
+
+            self.moe_mlp = MoEMLP(cfg, dtype=dtype, device=device)
+        else:
+            self.fc1 = K.FusedLinear(h, 2 * cfg.intermediate_size, dtype=dtype, device=device)
+            self.fc2 = K.FusedLinear(cfg.intermediate_size, h, dtype=dtype, device=device)
 
     def forward(self, x, rope_cs, pos):
         # x: [B, S, H]
@@ -54,7 +60,10 @@ class LlamaDecoderLayer(nn.Module):
         x = res + self.out_proj(ctx.reshape(B, S, self.heads * d))
         res = x
         y = self.pre_mlp_norm(x)
-        y = self.fc2(K.swiglu(self.fc1(y)))
+        if self.use_moe:
+            y = self.moe_mlp(y)
+        else:
+            y = self.fc2(K.swiglu(self.fc1(y)))
         return res + y
 
 

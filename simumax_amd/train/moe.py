@@ -1,0 +1,86 @@
+"""MoE MLP for the reference trainer (mixtral-style top-k routing with
+capacity dispatch), matching the simulator's MoE cost/memory accounting:
+
+* router GEMM -> softmax -> top-k (Router)
+* capacity-padded dispatch: cap = ceil(tokens*topk/E * capacity); tokens
+  beyond an expert's capacity are dropped (Permutation with
+  moe_pad_expert_input_to_capacity)
+* per-expert GEMMs as ONE torch.bmm over [E, cap, *] — the exact op the
+  group_matmul calibration sweep times (calib/sweeps.sweep_grouped)
+* weighted combine scatter (UnPermutation)
+
+Single-rank experts (EP1); the EP all-to-all path is simulator-side this
+round (multi-GPU EP training is a round-2 item).
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+
+
+class MoEMLP(nn.Module):
+    def __init__(self, cfg, dtype=torch.bfloat16, device=None):
+        super().__init__()
+        h = cfg.hidden_size
+        self.E = cfg.expert_num
+        self.topk = cfg.topk
+        self.I = cfg.moe_ffn_hidden_size
+        self.capacity = getattr(cfg, "capacity", 1) or 1
+        self.router = nn.Linear(h, self.E, bias=False, dtype=dtype,
+                                device=device)
+        # grouped weights: [E, 2I, h] (fc1, swiglu) and [E, h, I] (fc2)
+        self.w1 = nn.Parameter(torch.empty(self.E, 2 * self.I, h, dtype=dtype,
+                                           device=device))
+        self.w2 = nn.Parameter(torch.empty(self.E, h, self.I, dtype=dtype,
+                                           device=device))
+        nn.init.normal_(self.w1, std=0.02)
+        nn.init.normal_(self.w2, std=0.02)
+
+    def forward(self, x):
+        from ..kernels import ops as K
+
+        B, S, H = x.shape
+        N = B * S
+        xf = x.reshape(N, H)
+        logits = self.router(xf).float()                     # [N, E]
+        probs = torch.softmax(logits, dim=-1)
+        weight, idx = probs.topk(self.topk, dim=-1)          # [N, k]
+        weight = weight / weight.sum(-1, keepdim=True)
+
+        cap = int(math.ceil(N * self.topk / self.E * self.capacity))
+        flat_expert = idx.reshape(-1)                        # [N*k]
+        flat_token = (torch.arange(N, device=x.device)
+                      .repeat_interleave(self.topk))         # [N*k]
+        # position of each (token, expert) slot within its expert's queue
+        order = torch.argsort(flat_expert, stable=True)
+        counts = torch.bincount(flat_expert, minlength=self.E)
+        # rank within expert for sorted order
+        offs = torch.cumsum(counts, 0) - counts
+        rank_sorted = (torch.arange(N * self.topk, device=x.device)
+                       - offs[flat_expert[order]])
+        keep = rank_sorted < cap                             # capacity drop
+        src_tok = flat_token[order][keep]
+        dst_exp = flat_expert[order][keep]
+        dst_slot = rank_sorted[keep]
+        slot_index = dst_exp * cap + dst_slot                # [M]
+        w_kept = weight.reshape(-1)[order][keep].to(x.dtype)
+
+        # dispatch: [E*cap, H] padded buffer (Permutation)
+        xp = torch.zeros(self.E * cap, H, dtype=x.dtype, device=x.device)
+        xp.index_copy_(0, slot_index, xf.index_select(0, src_tok))
+        xp = xp.view(self.E, cap, H)
+
+        # grouped GEMMs (the calibrated bmm path) + fused swiglu
+        h1 = torch.bmm(xp, self.w1.transpose(1, 2))          # [E, cap, 2I]
+        a = K.swiglu(h1.reshape(-1, 2 * self.I)).reshape(self.E, cap, self.I)
+        y = torch.bmm(a, self.w2.transpose(1, 2))            # [E, cap, H]
+
+        # combine: weighted scatter back (UnPermutation)
+        y_flat = y.reshape(self.E * cap, H)
+        out = torch.zeros_like(xf)
+        out.index_add_(0, src_tok,
+                       y_flat.index_select(0, slot_index) * w_kept[:, None])
+        return out.view(B, S, H)
