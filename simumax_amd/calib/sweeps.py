@@ -214,8 +214,12 @@ def parse_group_key(desc):
 
 
 def sweep_grouped(shape_keys, path):
-    """Time ng per-expert GEMMs (hipBLASLt via torch) — the trainer-side
-    grouped-GEMM realization on ROCm ("parallel" group_linear_mode)."""
+    """Time the trainer's grouped-GEMM realization per stage: per-expert mm
+    loop for fwd/dgrad, fused wgrad_accum loop for wgrad (train/moe.py
+    _GroupedLinearFn; torch.bmm's backward faults on this stack)."""
+    from simumax_amd.kernels.ops import ext
+
+    E_ = ext()
     tab = _load(path)
     for desc in shape_keys:
         if desc in tab and not OVERWRITE:
@@ -224,7 +228,18 @@ def sweep_grouped(shape_keys, path):
         try:
             x = torch.randn(ng, m, k, device="cuda", dtype=torch.bfloat16)
             w = torch.randn(ng, k, n, device="cuda", dtype=torch.bfloat16)
-            fn = lambda: torch.bmm(x, w)
+            out = torch.empty(ng, m, n, device="cuda", dtype=torch.bfloat16)
+            if "stage=fwd" in desc or "stage=bwd_grad_act" in desc:
+                def fn():
+                    for e in range(ng):
+                        torch.mm(x[e], w[e], out=out[e])
+            else:  # wgrad: fused fp32 accumulate per expert
+                d = torch.randn(ng, m, n, device="cuda", dtype=torch.bfloat16)
+                mg = torch.zeros(ng, k, n, device="cuda", dtype=torch.float32)
+
+                def fn():
+                    for e in range(ng):
+                        E_.wgrad_accum(x[e], d[e], mg[e])
             t_ms = _timeit(fn, iters=8)
         except torch.cuda.OutOfMemoryError:
             torch.cuda.empty_cache()

@@ -21,6 +21,49 @@ import torch
 import torch.nn as nn
 
 
+class _GroupedLinearFn(torch.autograd.Function):
+    """Per-expert GEMM loop: out[e] = x[e] @ w[e] with fused fp32 wgrad
+    accumulation. (torch.bmm's BACKWARD memory-faults in hipBLASLt
+    strided-batched mode at these shapes on this stack, so grouped GEMMs
+    run as E plain GEMMs — the same op the group_matmul calibration
+    sweeps time.)"""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        E = x.shape[0]
+        out = torch.empty(E, x.shape[1], w.shape[2], dtype=x.dtype,
+                          device=x.device)
+        for e in range(E):
+            torch.mm(x[e], w[e], out=out[e])
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        from ..kernels.ops import _dummy_wgrad, ext
+
+        x, w = ctx.saved_tensors
+        E = x.shape[0]
+        dout = dout.contiguous()
+        dx = torch.empty_like(x)
+        fused = x.is_cuda and hasattr(w, "main_grad")
+        dw = None if fused else torch.empty_like(w)
+        for e in range(E):
+            torch.mm(dout[e], w[e].t(), out=dx[e])
+            if fused:
+                # main_grad[e] (K,N) += x[e]^T (K,M) @ dout[e] (M,N)
+                ext().wgrad_accum(x[e], dout[e], w.main_grad[e])
+            else:
+                torch.mm(x[e].t(), dout[e], out=dw[e])
+        if fused:
+            dw = _dummy_wgrad(w.shape, w.device, w.dtype)
+        return dx, dw
+
+
+def grouped_linear(x, w):
+    return _GroupedLinearFn.apply(x, w)
+
+
 class MoEMLP(nn.Module):
     def __init__(self, cfg, dtype=torch.bfloat16, device=None):
         super().__init__()
@@ -31,13 +74,17 @@ class MoEMLP(nn.Module):
         self.capacity = getattr(cfg, "capacity", 1) or 1
         self.router = nn.Linear(h, self.E, bias=False, dtype=dtype,
                                 device=device)
-        # grouped weights: [E, 2I, h] (fc1, swiglu) and [E, h, I] (fc2)
-        self.w1 = nn.Parameter(torch.empty(self.E, 2 * self.I, h, dtype=dtype,
+        # grouped weights stored PRE-TRANSPOSED for bmm ([E, K, N]) — a
+        # runtime .transpose(1,2) view fed to hipBLASLt strided-batched GEMM
+        # memory-faults on this stack
+        self.w1 = nn.Parameter(torch.empty(self.E, h, 2 * self.I, dtype=dtype,
                                            device=device))
-        self.w2 = nn.Parameter(torch.empty(self.E, h, self.I, dtype=dtype,
+        self.w2 = nn.Parameter(torch.empty(self.E, self.I, h, dtype=dtype,
                                            device=device))
         nn.init.normal_(self.w1, std=0.02)
         nn.init.normal_(self.w2, std=0.02)
+        self.w1._fused_wgrad = True
+        self.w2._fused_wgrad = True
 
     def forward(self, x):
         from ..kernels import ops as K
@@ -73,10 +120,10 @@ class MoEMLP(nn.Module):
         xp.index_copy_(0, slot_index, xf.index_select(0, src_tok))
         xp = xp.view(self.E, cap, H)
 
-        # grouped GEMMs (the calibrated bmm path) + fused swiglu
-        h1 = torch.bmm(xp, self.w1.transpose(1, 2))          # [E, cap, 2I]
+        # grouped GEMMs (per-expert mm loop, the calibrated path) + swiglu
+        h1 = grouped_linear(xp, self.w1)                     # [E, cap, 2I]
         a = K.swiglu(h1.reshape(-1, 2 * self.I)).reshape(self.E, cap, self.I)
-        y = torch.bmm(a, self.w2.transpose(1, 2))            # [E, cap, H]
+        y = grouped_linear(a, self.w2)                       # [E, cap, H]
 
         # combine: weighted scatter back (UnPermutation)
         y_flat = y.reshape(self.E * cap, H)
