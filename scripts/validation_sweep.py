@@ -86,7 +86,10 @@ def run_case(name, model, seq, mbs, mbc, steps, warmup=1):
     with open(OUT, "a") as f:
         f.write(json.dumps(row) + "\n")
     # free everything before the next case (hooks pin the model otherwise)
+    from simumax_amd.kernels.ops import clear_dummy_wgrads
+
     red.remove_hooks()
+    clear_dummy_wgrads()
     del m, opt, red, toks, labels
     gc.collect()
     torch.cuda.empty_cache()

@@ -98,7 +98,10 @@ def real_run(case: dict) -> dict:
     ms = (time.time() - t0) / case["steps"] * 1e3
     out = dict(measured_ms=ms,
                measured_bytes=float(torch.cuda.max_memory_allocated()))
+    from simumax_amd.kernels.ops import clear_dummy_wgrads
+
     red.remove_hooks()
+    clear_dummy_wgrads()
     del model, opt, red, toks, labels
     import gc
 

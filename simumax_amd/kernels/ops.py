@@ -207,6 +207,12 @@ def fused_cross_entropy(logits, labels):
 _DUMMY_WGRADS = {}
 
 
+def clear_dummy_wgrads():
+    """Release the shared dummy-wgrad buffers (benchmark sweeps that build
+    several models in one process must call this between models)."""
+    _DUMMY_WGRADS.clear()
+
+
 def _dummy_wgrad(shape, device, dtype):
     """One shared placeholder grad per shape (TE get_dummy_wgrad analog —
     the reference memory model tracks these as te_dummy_wgrad_shapes)."""
