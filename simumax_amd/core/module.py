@@ -177,23 +177,31 @@ class MetaModule:
         sysc = self.system
         comp = self._compute_info
 
-        def stage_time(op_name, stage, flops, mem):
+        def stage_time(op_name, stage, flops, mem, extra=0.0):
             c = sysc.compute_op_accuracy_time(
                 op_name, flops, shape_desc=self.get_input_shapes_desc(stage),
                 reture_detail=True,
             )
             io = sysc.compute_mem_access_time(op_name, mem, reture_detail=True)
-            return sysc.compute_end2end_time(c["compute_only_time"], io["io_time"])
+            t = sysc.compute_end2end_time(c["compute_only_time"], io["io_time"])
+            if extra > 0:
+                # glue kernels (layout copies, grad fan-in adds) run as
+                # separate launches at stream bandwidth, additive to the op
+                t += sysc.compute_mem_access_time("default", extra)
+            return t
 
         ci = self._cost_info
         ci.fwd_compute_time = stage_time(self.fwd_op, "fwd", comp.fwd_flops,
-                                         comp.fwd_accessed_mem)
+                                         comp.fwd_accessed_mem,
+                                         comp.fwd_extra_mem)
         ci.bwd_grad_act_time = stage_time(self.bwd_act_op, "bwd_grad_act",
                                           comp.bwd_grad_act_flops,
-                                          comp.bwd_grad_act_accessed_mem)
+                                          comp.bwd_grad_act_accessed_mem,
+                                          comp.bwd_grad_act_extra_mem)
         ci.bwd_grad_w_time = stage_time(self.bwd_w_op, "bwd_grad_w",
                                         comp.bwd_grad_w_flops,
-                                        comp.bwd_grad_w_accessed_mem)
+                                        comp.bwd_grad_w_accessed_mem,
+                                        comp.bwd_grad_w_extra_mem)
         self._price_comm()
         if self.enable_recompute and not self.is_variance_node:
             ci.recompute_compute_time = ci.fwd_compute_time

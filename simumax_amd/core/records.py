@@ -59,6 +59,12 @@ class ModuleComputeInfo:
     bwd_grad_act_accessed_mem: float = 0.0
     bwd_grad_w_accessed_mem: float = 0.0
     recompute_accessed_mem: float = 0.0
+    # bytes moved by SEPARATE glue kernels around the op (layout copies,
+    # autograd fan-in grad adds, main-grad cast+add hooks) — priced
+    # additively at stream bandwidth, not folded into the roofline
+    fwd_extra_mem: float = 0.0
+    bwd_grad_act_extra_mem: float = 0.0
+    bwd_grad_w_extra_mem: float = 0.0
 
     @property
     def bwd_flops(self):
@@ -78,6 +84,9 @@ class ModuleComputeInfo:
             self.bwd_grad_act_accessed_mem + other.bwd_grad_act_accessed_mem,
             self.bwd_grad_w_accessed_mem + other.bwd_grad_w_accessed_mem,
             self.recompute_accessed_mem + other.recompute_accessed_mem,
+            self.fwd_extra_mem + other.fwd_extra_mem,
+            self.bwd_grad_act_extra_mem + other.bwd_grad_act_extra_mem,
+            self.bwd_grad_w_extra_mem + other.bwd_grad_w_extra_mem,
         )
 
 

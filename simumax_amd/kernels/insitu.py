@@ -1,0 +1,124 @@
+"""In-situ operator calibration: hipEvent timing of the trainer's real
+op calls, keyed by the simulator's exact shape-key strings.
+
+Microbenchmarks (calib/sweeps.py) time 10 identical iterations with hot
+L2 and burst clocks; inside a training step the same op runs in a mixed
+kernel stream. Enabling this registry during real steps produces
+per-shape-key efficiency tables measured UNDER TRAINING CONDITIONS,
+which calib/merge.py merges with precedence over the microbench values
+(reference analog: SimuMax's benchmark suite measuring operator
+efficiency on the target model's own shapes).
+
+Usage (see scripts/insitu_calib.py):
+    from simumax_amd.kernels import insitu
+    insitu.enable()
+    ... run training steps ...
+    insitu.dump("gpurun_out/calib")        # *_insitu.json sweep files
+"""
+
+from __future__ import annotations
+
+import json
+import os
+from collections import defaultdict
+
+import torch
+
+ENABLED = False
+_RECORDS = defaultdict(list)  # (table, key) -> [(start_evt, end_evt), ...]
+
+PEAK_BF16_TFLOPS = 2500.0
+
+
+def enable():
+    global ENABLED
+    ENABLED = True
+    _RECORDS.clear()
+
+
+def disable():
+    global ENABLED
+    ENABLED = False
+
+
+def start(table: str, key: str):
+    """Record an event pair around the op the caller is about to launch.
+    Returns a closure to call right after the launch (same stream)."""
+    s = torch.cuda.Event(enable_timing=True)
+    e = torch.cuda.Event(enable_timing=True)
+    s.record()
+
+    def stop():
+        e.record()
+        _RECORDS[(table, key)].append((s, e))
+
+    return stop
+
+
+def gemm_key(b, m, k, n, layout, accumulate, out_dtype):
+    return (f"b={b}, m={m}, k={k}, n={n}, layout={layout}, "
+            f"accumulate={accumulate}, out_dtype={out_dtype}")
+
+
+def sdp_key(b, s, hq, hkv, dqk, dv, contiguous=True):
+    return (f"batch={b}, seq_len={s}, head_num={hq}, kv_head_num={hkv}, "
+            f"qk_head_dim={dqk}, v_head_dim={dv}, "
+            f"qkv_contiguous={contiguous}")
+
+
+def _gemm_flops(key):
+    import re
+
+    m = re.match(r"b=(\d+), m=(\d+), k=(\d+), n=(\d+)", key)
+    b, mm, k, n = (int(m.group(i)) for i in range(1, 5))
+    return 2 * b * mm * k * n
+
+
+def _sdp_flops(key, stage):
+    import re
+
+    m = re.match(
+        r"batch=(\d+), seq_len=(\d+), head_num=(\d+), kv_head_num=(\d+), "
+        r"qk_head_dim=(\d+), v_head_dim=(\d+)", key)
+    b, s, hq, _, dqk, dv = (int(m.group(i)) for i in range(1, 7))
+    qk = 2 * b * hq * s * s * dqk
+    pv = 2 * b * hq * s * s * dv
+    causal = 0.5
+    if stage == "fwd":
+        return (qk + pv) * causal
+    return (3 * qk + 2 * pv) * causal  # bwd recomputes QK^T
+
+
+def summarize():
+    """Sync and reduce recorded event pairs -> {table: {key: stats}}."""
+    torch.cuda.synchronize()
+    out = defaultdict(dict)
+    for (table, key), pairs in _RECORDS.items():
+        ts = sorted(s.elapsed_time(e) for s, e in pairs)
+        t = ts[len(ts) // 2]  # median instance
+        row = dict(t_ms=round(t, 5), n=len(ts))
+        if table == "matmul":
+            row["eff"] = _gemm_flops(key) / (t / 1e3) / (PEAK_BF16_TFLOPS * 1e12)
+        elif table in ("sdp_fwd", "sdp_bwd"):
+            stage = "fwd" if table == "sdp_fwd" else "bwd"
+            row["eff"] = _sdp_flops(key, stage) / (t / 1e3) / (PEAK_BF16_TFLOPS * 1e12)
+        out[table][key] = row
+    return out
+
+
+def dump(outdir):
+    os.makedirs(outdir, exist_ok=True)
+    summary = summarize()
+    # matmul_insitu.json matches the sweep-file schema {desc: {eff, t_ms}}
+    for table, rows in summary.items():
+        path = os.path.join(outdir, f"{table}_insitu.json")
+        existing = {}
+        if os.path.exists(path):
+            with open(path) as f:
+                existing = json.load(f)
+        existing.update(rows)
+        with open(path, "w") as f:
+            json.dump(existing, f, indent=1, sort_keys=True)
+        print(f"[insitu] {path}: {len(rows)} keys")
+    _RECORDS.clear()
+    return summary

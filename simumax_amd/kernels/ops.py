@@ -13,6 +13,8 @@ import os
 
 import torch
 
+from . import insitu
+
 _EXT = None
 _TRIED = False
 
@@ -226,17 +228,41 @@ class _FusedLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight):
         ctx.save_for_backward(x, weight)
+        if insitu.ENABLED and x.is_cuda:
+            b, m = (x.shape[0], x.shape[1]) if x.ndim == 3 else (1, x.shape[0])
+            stop = insitu.start("matmul", insitu.gemm_key(
+                b, m, x.shape[-1], weight.shape[0], "TN", False, "bf16"))
+            out = torch.matmul(x, weight.t())
+            stop()
+            return out
         return torch.matmul(x, weight.t())
 
     @staticmethod
     def backward(ctx, dout):
         x, weight = ctx.saved_tensors
+        timing = insitu.ENABLED and x.is_cuda
+        if timing:
+            b, m = (x.shape[0], x.shape[1]) if x.ndim == 3 else (1, x.shape[0])
+            stop = insitu.start("matmul", insitu.gemm_key(
+                b, m, weight.shape[0], weight.shape[1], "NN", False, "bf16"))
         dx = torch.matmul(dout, weight)
+        if timing:
+            stop()
         if x.is_cuda and hasattr(weight, "main_grad"):
             d2 = dout.reshape(-1, dout.shape[-1]).contiguous()
             x2 = x.reshape(-1, x.shape[-1]).contiguous()
+            if timing:
+                stop = insitu.start("matmul", insitu.gemm_key(
+                    1, weight.shape[0], x2.shape[0], weight.shape[1],
+                    "NT", True, "fp32"))
             ext().wgrad_accum(d2, x2, weight.main_grad)
-            dw = _dummy_wgrad(weight.shape, weight.device, weight.dtype)
+            if timing:
+                stop()
+            # fresh unreferenced buffer: autograd's AccumulateGrad STEALS it
+            # (no weight-sized clone kernel; the shared-dummy variant gets
+            # cloned because its use_count is too high). The DP reducer's
+            # post-accumulate hook frees it right away.
+            dw = torch.empty_like(weight)
         else:
             dw = torch.matmul(dout.reshape(-1, dout.shape[-1]).t(),
                               x.reshape(-1, x.shape[-1]))
@@ -266,7 +292,14 @@ class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal):
         if _use_hip(q):
-            o, lse = ext().fa_fwd(q, k, v, causal)
+            if insitu.ENABLED:
+                stop = insitu.start("sdp_fwd", insitu.sdp_key(
+                    q.shape[0], q.shape[1], q.shape[2], k.shape[2],
+                    q.shape[3], v.shape[3]))
+                o, lse = ext().fa_fwd(q, k, v, causal)
+                stop()
+            else:
+                o, lse = ext().fa_fwd(q, k, v, causal)
             ctx.save_for_backward(q, k, v, o, lse)
             ctx.causal = causal
             return o
@@ -279,8 +312,16 @@ class _FlashAttnFn(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         if _use_hip(q):
-            dq, dk, dv = ext().fa_bwd(do.contiguous(), q, k, v, o, lse,
-                                      ctx.causal)
+            if insitu.ENABLED:
+                stop = insitu.start("sdp_bwd", insitu.sdp_key(
+                    q.shape[0], q.shape[1], q.shape[2], k.shape[2],
+                    q.shape[3], v.shape[3]))
+                dq, dk, dv = ext().fa_bwd(do.contiguous(), q, k, v, o, lse,
+                                          ctx.causal)
+                stop()
+            else:
+                dq, dk, dv = ext().fa_bwd(do.contiguous(), q, k, v, o, lse,
+                                          ctx.causal)
             return dq, dk, dv, None
         # CPU fallback: autograd through the reference math
         with torch.enable_grad():

@@ -99,9 +99,14 @@ class Embedding(MetaModule, ParamMixin):
             out_full *= self.strategy.tp_size
         info.fwd_accessed_mem = out_full + self.input_info.tensors[0].mem_bytes()
         # bwd: scatter-add of grad into weight grad buffer
+        local_w = self.vocab_size // self.strategy.tp_size * self.hidden_size
         info.bwd_grad_w_accessed_mem = out_full + (
-            self.vocab_size // self.strategy.tp_size * self.hidden_size * self.grad_element_size
+            local_w * self.grad_element_size
         )
+        if self.strategy.use_fused_grad_accumulation:
+            # post-accumulate hook: p.grad.float() temp (r2+w4) then
+            # main_grad.add_ (r4+r4+w4) = 18 B/elem, per microbatch
+            info.bwd_grad_w_extra_mem = local_w * 18
 
     def _leaf_intra_net_info(self):
         tp = self.strategy.tp_size
@@ -162,8 +167,8 @@ class LinearCol(LinearBase, ParamMixin):
     def _leaf_act_info(self, info):
         # caches the (sharded under SP) input; the gathered copy is transient
         info.activation_mem_cache = self.input_info.first.mem_bytes()
-        # bwd transient: autograd's clone of the shared dummy wgrad
-        # (freed by the post-accumulate hook right after the wgrad GEMM)
+        # bwd transient: the freshly allocated placeholder wgrad that
+        # AccumulateGrad steals (freed by the post-accumulate hook)
         info.bwd_peak_mem_no_cache = (
             self.input_size * self.output_size * self.element_size)
         if self.sp:
@@ -302,6 +307,9 @@ class LayerNorm(MetaModule, ParamMixin):
         # bwd reads dout + input, writes din (+weight-grad reduction)
         info.bwd_grad_act_accessed_mem = 3 * b * passes
         info.bwd_grad_w_accessed_mem = 0
+        if self.strategy.use_fused_grad_accumulation:
+            # post-accumulate hook cast+add of the norm weight grad
+            info.bwd_grad_w_extra_mem = self.hidden_size * 18
 
 
 class RotaryEmbedding(MetaModule):
@@ -440,6 +448,13 @@ class CoreAttention(MetaModule):
         if self.use_flash:
             info.fwd_accessed_mem = q + k + v + o + lse
             info.bwd_grad_act_accessed_mem = 2 * (q + k + v) + 2 * o + lse
+            if self.qkv_contiguous:
+                # trainer parity: q/k/v are strided views of the fused qkv
+                # GEMM output and are materialized contiguous before the
+                # kernel (fwd); backward of the split cats dq/dk/dv into
+                # one dqkv buffer. Both are separate copy kernels.
+                info.fwd_extra_mem = 2 * (q + k + v)
+                info.bwd_grad_act_extra_mem = 2 * (q + k + v)
         else:
             scores = b * self.sdp_head_num * s * s * self.element_size
             info.fwd_accessed_mem = q + k + v + o + 4 * scores
@@ -652,7 +667,10 @@ class Add(MetaModule):
         b = self.input_info.tensors[0].mem_bytes()
         info.fwd_flops = self.input_info.tensors[0].numel()
         info.fwd_accessed_mem = 3 * b
-        # bwd of add is a no-op (gradient fan-out)
+        # bwd of add passes grads through, but the residual FAN-IN (the
+        # skip tensor is consumed twice) makes autograd accumulate two
+        # gradient paths: one elementwise add of the grad tensor
+        info.bwd_grad_act_extra_mem = 3 * b
 
 
 # ==========================================================================

@@ -351,14 +351,11 @@ class PerfLLM(PerfBase):
                 pp_point = max((self.vchunks[stage][c].peak_point
                                 for c in range(vp)),
                                key=lambda x: x.peak_mem)
+            # fused-wgrad placeholder grads are freshly allocated per
+            # backward and STOLEN by AccumulateGrad (kernels/ops.py), so
+            # they are transient (modeled per-linear in bwd_peak_mem) —
+            # no persistent shared-dummy term remains.
             dummy_wgrad = 0.0
-            if s.use_fused_grad_accumulation:
-                from ..core.module import LinearBase
-
-                shapes = {(l.input_size, l.output_size)
-                          for l in chunk.leaf_modules()
-                          if isinstance(l, LinearBase)}
-                dummy_wgrad = sum(i * o * 2 for i, o in shapes)
             peak = (
                 model_info.all_bytes
                 + dummy_wgrad

@@ -29,19 +29,11 @@ def run_simulation(perf_model, save_path: str, merge_lanes: bool = True) -> Resu
     # reference parity simu_artifacts.py:9-10)
     s = perf_model.strategy
     per_stage = s.world_size // s.pp_size
-    from ..core.module import LinearBase
-
     base = {}
     for r in jobs:
         stage = r // per_stage
         chunk = perf_model.chunks[stage]
         base[r] = chunk.get_model_info().all_bytes
-        if s.use_fused_grad_accumulation:
-            # shared dummy-wgrad buffers (static, one per weight shape)
-            shapes = {(l.input_size, l.output_size)
-                      for l in chunk.leaf_modules()
-                      if isinstance(l, LinearBase)}
-            base[r] += sum(i * o * 2 for i, o in shapes)
     tracker = SimuMemoryTracker(base)
     for e in sorted(system.log, key=lambda x: (x.start, x.rank)):
         tracker.record(e.rank, e.start, e.name, e.kind, e.mem)

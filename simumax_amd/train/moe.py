@@ -21,6 +21,24 @@ import torch
 import torch.nn as nn
 
 
+def _grouped_key(ng, m, k, n, stage):
+    """The simulator's group_matmul shape key (core/module.py
+    GroupLinearBase.get_input_shapes_desc) for the trainer's [E,M,K]@[E,K,N]
+    grouped op. NOTE: N/K in the key are the LAYER's output/input sizes."""
+    s = (f"ng={ng}, M={m}, N={n}, K={k}, dtype=bf16, out_dtype=bf16, "
+         f"main_grad_dtype=fp32")
+    if stage == "fwd":
+        s += (", stage=fwd, grad=False, accumulate=False, "
+              "use_split_accumulator=False, single_output=True")
+    elif stage == "bwd_grad_act":
+        s += (", stage=bwd_grad_act, grad=True, accumulate=False, "
+              "use_split_accumulator=True, single_output=False")
+    else:
+        s += (", stage=bwd_grad_w, grad=True, accumulate=True, "
+              "use_split_accumulator=True, single_output=False")
+    return s
+
+
 class _GroupedLinearFn(torch.autograd.Function):
     """Per-expert GEMM loop: out[e] = x[e] @ w[e] with fused fp32 wgrad
     accumulation. (torch.bmm's BACKWARD memory-faults in hipBLASLt
@@ -30,17 +48,26 @@ class _GroupedLinearFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w):
+        from ..kernels import insitu
+
         ctx.save_for_backward(x, w)
         E = x.shape[0]
         out = torch.empty(E, x.shape[1], w.shape[2], dtype=x.dtype,
                           device=x.device)
+        timing = insitu.ENABLED and x.is_cuda
+        if timing:
+            stop = insitu.start("group_matmul", _grouped_key(
+                E, x.shape[1], w.shape[1], w.shape[2], "fwd"))
         for e in range(E):
             torch.mm(x[e], w[e], out=out[e])
+        if timing:
+            stop()
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        from ..kernels.ops import _dummy_wgrad, ext
+        from ..kernels import insitu
+        from ..kernels.ops import ext
 
         x, w = ctx.saved_tensors
         E = x.shape[0]
@@ -48,15 +75,28 @@ class _GroupedLinearFn(torch.autograd.Function):
         dx = torch.empty_like(x)
         fused = x.is_cuda and hasattr(w, "main_grad")
         dw = None if fused else torch.empty_like(w)
+        timing = insitu.ENABLED and x.is_cuda
+        if timing:
+            stop_dx = insitu.start("group_matmul", _grouped_key(
+                E, x.shape[1], w.shape[1], w.shape[2], "bwd_grad_act"))
         for e in range(E):
             torch.mm(dout[e], w[e].t(), out=dx[e])
-            if fused:
+        if timing:
+            stop_dx()
+            stop_dw = insitu.start("group_matmul", _grouped_key(
+                E, x.shape[1], w.shape[1], w.shape[2], "bwd_grad_w"))
+        if fused:
+            for e in range(E):
                 # main_grad[e] (K,N) += x[e]^T (K,M) @ dout[e] (M,N)
                 ext().wgrad_accum(x[e], dout[e], w.main_grad[e])
-            else:
+        else:
+            for e in range(E):
                 torch.mm(x[e].t(), dout[e], out=dw[e])
+        if timing:
+            stop_dw()
         if fused:
-            dw = _dummy_wgrad(w.shape, w.device, w.dtype)
+            # fresh unreferenced buffer -> AccumulateGrad steals it (no clone)
+            dw = torch.empty_like(w)
         return dx, dw
 
 

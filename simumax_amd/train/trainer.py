@@ -121,10 +121,9 @@ class DataParallelGradReducer:
             offs[id(p)] = (off, off + p.numel())
             off += p.numel()
         # Hooks are registered UNCONDITIONALLY: releasing p.grad right after
-        # each accumulation is a memory-correctness requirement — autograd
-        # CLONES the shared dummy wgrad (its use_count is too high to
-        # steal), and without the hook those weight-sized clones pile up in
-        # p.grad until backward ends (~25 GiB on llama3-70b-l12).
+        # each accumulation is a memory-correctness requirement — without
+        # the hook the weight-sized placeholder wgrads pile up in p.grad
+        # until backward ends (~25 GiB on llama3-70b-l12).
         # reverse order (grads become ready back-to-front); reversed
         # consecutive params are a contiguous flat slice
         buckets, cur, cur_bytes = [], [], 0
