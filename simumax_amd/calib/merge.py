@@ -20,12 +20,28 @@ def _load(name):
     return {}
 
 
+def _overlay_insitu(table, name):
+    """Overlay in-situ measured efficiencies (kernels/insitu.py dump;
+    {desc: {eff, t_ms, n}}) with precedence over microbench values —
+    in-situ timing reflects training-condition clocks and cache state."""
+    ins = _load(name)
+    n = 0
+    for desc, row in ins.items():
+        if isinstance(row, dict) and "eff" in row:
+            table[desc] = round(row["eff"], 4)
+            n += 1
+    if n:
+        print(f"  overlaid {n} in-situ shapes from {name}")
+    return table
+
+
 def main():
     with open(SYSTEM) as f:
         sysc = json.load(f)
     acc = sysc["accelerator"]
 
     matmul = _load("matmul.json")
+    matmul = _overlay_insitu(matmul, "matmul_insitu.json")
     if matmul:
         acc["op"]["matmul"]["accurate_efficient_factor"] = matmul
         acc["op"]["matmul"]["efficient_factor"] = round(
@@ -34,6 +50,7 @@ def main():
               f"{acc['op']['matmul']['efficient_factor']}")
     for key in ("sdp_fwd", "sdp_bwd"):
         tab = _load(f"{key}.json")
+        tab = _overlay_insitu(tab, f"{key}_insitu.json")
         if tab:
             acc["op"][key]["accurate_efficient_factor"] = tab
             acc["op"][key]["efficient_factor"] = round(
@@ -48,6 +65,7 @@ def main():
         print(f"fp8_matmul: {len(fp8)} shapes, median eff "
               f"{acc['op']['fp8_matmul']['efficient_factor']}")
     group = _load("group_matmul.json")
+    group = _overlay_insitu(group, "group_matmul_insitu.json")
     if group:
         acc["op"]["group_matmul"]["accurate_efficient_factor"] = group
         acc["op"]["group_matmul"]["efficient_factor"] = round(
@@ -55,6 +73,7 @@ def main():
         print(f"group_matmul: {len(group)} shapes")
 
     bw = _load("bandwidth.json")
+    bw.update(_load("bandwidth_insitu.json"))
     if bw:
         if "default_eff" in bw:
             acc["bandwidth"]["default"]["efficient_factor"] = round(
