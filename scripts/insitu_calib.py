@@ -84,10 +84,15 @@ def main():
 
     # median optimizer efficiency across models
     effs = sorted(r["eff"] for r in opt_samples)
-    out = dict(optimizer_eff=effs[len(effs) // 2],
-               optimizer_samples=opt_samples)
+    path = os.path.join(OUTDIR, "bandwidth_insitu.json")
+    out = {}
+    if os.path.exists(path):
+        with open(path) as f:
+            out = json.load(f)
+    out["optimizer_eff"] = effs[len(effs) // 2]
+    out["optimizer_samples"] = opt_samples
     os.makedirs(OUTDIR, exist_ok=True)
-    with open(os.path.join(OUTDIR, "bandwidth_insitu.json"), "w") as f:
+    with open(path, "w") as f:
         json.dump(out, f, indent=1)
     print(json.dumps(out, indent=1))
 

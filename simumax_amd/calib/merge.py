@@ -88,8 +88,12 @@ def main():
             # unfused ce not implemented separately on MI355X: same kernel
             acc["bandwidth"]["ce"]["efficient_factor"] = acc["bandwidth"][
                 "ce_fusion"]["efficient_factor"]
-        for key in ("permute_fwd", "permute_bwd"):
+        for key in ("permute_fwd", "permute_bwd", "rmsnorm_fwd",
+                    "rmsnorm_bwd", "rope", "swiglu", "swiglu_bwd"):
             if f"{key}_eff" in bw:
+                acc["bandwidth"].setdefault(key, {
+                    "gbps": 8000.0, "efficient_factor": 0.55,
+                    "latency_us": round(bw.get("launch_us", 4.0), 2)})
                 acc["bandwidth"][key]["efficient_factor"] = round(
                     bw[f"{key}_eff"], 4)
         if "optimizer_eff" in bw:

@@ -177,12 +177,13 @@ class MetaModule:
         sysc = self.system
         comp = self._compute_info
 
-        def stage_time(op_name, stage, flops, mem, extra=0.0):
+        def stage_time(op_name, stage, flops, mem, extra=0.0, mem_op=None):
             c = sysc.compute_op_accuracy_time(
                 op_name, flops, shape_desc=self.get_input_shapes_desc(stage),
                 reture_detail=True,
             )
-            io = sysc.compute_mem_access_time(op_name, mem, reture_detail=True)
+            io = sysc.compute_mem_access_time(mem_op or op_name, mem,
+                                              reture_detail=True)
             t = sysc.compute_end2end_time(c["compute_only_time"], io["io_time"])
             if extra > 0:
                 # glue kernels (layout copies, grad fan-in adds) run as
@@ -193,11 +194,13 @@ class MetaModule:
         ci = self._cost_info
         ci.fwd_compute_time = stage_time(self.fwd_op, "fwd", comp.fwd_flops,
                                          comp.fwd_accessed_mem,
-                                         comp.fwd_extra_mem)
+                                         comp.fwd_extra_mem,
+                                         getattr(self, "fwd_mem_op", None))
         ci.bwd_grad_act_time = stage_time(self.bwd_act_op, "bwd_grad_act",
                                           comp.bwd_grad_act_flops,
                                           comp.bwd_grad_act_accessed_mem,
-                                          comp.bwd_grad_act_extra_mem)
+                                          comp.bwd_grad_act_extra_mem,
+                                          getattr(self, "bwd_act_mem_op", None))
         ci.bwd_grad_w_time = stage_time(self.bwd_w_op, "bwd_grad_w",
                                         comp.bwd_grad_w_flops,
                                         comp.bwd_grad_w_accessed_mem,

@@ -89,11 +89,24 @@ def _sdp_flops(key, stage):
     return (3 * qk + 2 * pv) * causal  # bwd recomputes QK^T
 
 
+HBM_PEAK_GBPS = 8000.0
+
+
 def summarize():
-    """Sync and reduce recorded event pairs -> {table: {key: stats}}."""
+    """Sync and reduce recorded event pairs -> {table: {key: stats}}.
+    `bw_<op>` tables carry modeled byte counts as the key; their summary
+    is one stream efficiency per op (total bytes / total time / peak)."""
     torch.cuda.synchronize()
     out = defaultdict(dict)
+    bw_acc = defaultdict(lambda: [0.0, 0.0, 0])  # op -> [bytes, ms, n]
     for (table, key), pairs in _RECORDS.items():
+        if table.startswith("bw_"):
+            acc = bw_acc[table[3:]]
+            for s, e in pairs:
+                acc[0] += float(key)
+                acc[1] += s.elapsed_time(e)
+                acc[2] += 1
+            continue
         ts = sorted(s.elapsed_time(e) for s, e in pairs)
         t = ts[len(ts) // 2]  # median instance
         row = dict(t_ms=round(t, 5), n=len(ts))
@@ -103,6 +116,12 @@ def summarize():
             stage = "fwd" if table == "sdp_fwd" else "bwd"
             row["eff"] = _sdp_flops(key, stage) / (t / 1e3) / (PEAK_BF16_TFLOPS * 1e12)
         out[table][key] = row
+    for op, (byt, ms, n) in bw_acc.items():
+        # subtract launch latency so the efficiency composes with the
+        # model's additive latency_us term
+        ms = max(ms - n * 0.004, 1e-6)
+        out["bandwidth"][f"{op}_eff"] = round(
+            byt / (ms / 1e3) / (HBM_PEAK_GBPS * 1024**3), 4)
     return out
 
 

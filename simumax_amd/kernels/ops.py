@@ -57,7 +57,12 @@ class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
         if _use_hip(x):
-            y, rstd = ext().rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps)
+            if insitu.ENABLED:
+                stop = insitu.start("bw_rmsnorm_fwd", str(2 * x.numel() * x.element_size()))
+                y, rstd = ext().rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps)
+                stop()
+            else:
+                y, rstd = ext().rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps)
         else:
             xf = x.float()
             rstd = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
@@ -70,7 +75,12 @@ class _RMSNormFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, rstd = ctx.saved_tensors
         if _use_hip(x):
-            dx, dw = ext().rmsnorm_bwd(dy.contiguous(), x, weight, rstd)
+            if insitu.ENABLED:
+                stop = insitu.start("bw_rmsnorm_bwd", str(3 * x.numel() * x.element_size()))
+                dx, dw = ext().rmsnorm_bwd(dy.contiguous(), x, weight, rstd)
+                stop()
+            else:
+                dx, dw = ext().rmsnorm_bwd(dy.contiguous(), x, weight, rstd)
         else:
             H = x.shape[-1]
             xf = x.float().reshape(-1, H)
@@ -111,6 +121,11 @@ class _RoPEFn(torch.autograd.Function):
     def forward(ctx, x, cs, pos):
         ctx.save_for_backward(cs, pos)
         if _use_hip(x):
+            if insitu.ENABLED:
+                stop = insitu.start("bw_rope", str(2 * x.numel() * x.element_size()))
+                y = ext().rope(x.contiguous(), cs, pos, 1.0)
+                stop()
+                return y
             return ext().rope(x.contiguous(), cs, pos, 1.0)
         return _rope_torch(x, cs, pos, 1.0)
 
@@ -118,6 +133,11 @@ class _RoPEFn(torch.autograd.Function):
     def backward(ctx, dy):
         cs, pos = ctx.saved_tensors
         if _use_hip(dy):
+            if insitu.ENABLED:
+                stop = insitu.start("bw_rope", str(2 * dy.numel() * dy.element_size()))
+                dx = ext().rope(dy.contiguous(), cs, pos, -1.0)
+                stop()
+                return dx, None, None
             return ext().rope(dy.contiguous(), cs, pos, -1.0), None, None
         return _rope_torch(dy, cs, pos, -1.0), None, None
 
@@ -146,6 +166,12 @@ class _SwigluFn(torch.autograd.Function):
     def forward(ctx, x):
         ctx.save_for_backward(x)
         if _use_hip(x):
+            if insitu.ENABLED:
+                b = x.numel() * x.element_size()
+                stop = insitu.start("bw_swiglu", str(b + b // 2))
+                y = ext().swiglu_fwd(x.contiguous())
+                stop()
+                return y
             return ext().swiglu_fwd(x.contiguous())
         g, u = x.float().chunk(2, dim=-1)
         return (torch.nn.functional.silu(g) * u).to(x.dtype)
@@ -154,6 +180,12 @@ class _SwigluFn(torch.autograd.Function):
     def backward(ctx, dy):
         (x,) = ctx.saved_tensors
         if _use_hip(x):
+            if insitu.ENABLED:
+                b = x.numel() * x.element_size()
+                stop = insitu.start("bw_swiglu_bwd", str(2 * b + b // 2))
+                dx = ext().swiglu_bwd(dy.contiguous(), x)
+                stop()
+                return dx
             return ext().swiglu_bwd(dy.contiguous(), x)
         g, u = x.float().chunk(2, dim=-1)
         dyf = dy.float()
@@ -175,7 +207,12 @@ class _CEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits, labels):
         if _use_hip(logits):
-            loss, row_max, row_sum = ext().ce_fwd(logits.contiguous(), labels)
+            if insitu.ENABLED:
+                stop = insitu.start("bw_ce_fusion", str(logits.numel() * logits.element_size()))
+                loss, row_max, row_sum = ext().ce_fwd(logits.contiguous(), labels)
+                stop()
+            else:
+                loss, row_max, row_sum = ext().ce_fwd(logits.contiguous(), labels)
         else:
             lf = logits.float()
             row_max = lf.max(-1).values
@@ -189,7 +226,12 @@ class _CEFn(torch.autograd.Function):
     def backward(ctx, dloss):
         logits, labels, row_max, row_sum = ctx.saved_tensors
         if _use_hip(logits):
-            d = ext().ce_bwd(logits, labels, dloss.contiguous(), row_max, row_sum)
+            if insitu.ENABLED:
+                stop = insitu.start("bw_ce_fusion_bwd", str(2 * logits.numel() * logits.element_size()))
+                d = ext().ce_bwd(logits, labels, dloss.contiguous(), row_max, row_sum)
+                stop()
+            else:
+                d = ext().ce_bwd(logits, labels, dloss.contiguous(), row_max, row_sum)
         else:
             p = (logits.float() - row_max[:, None]).exp() / row_sum[:, None]
             p.scatter_add_(1, labels[:, None],

@@ -282,6 +282,10 @@ class LayerNorm(MetaModule, ParamMixin):
     read-in/write-out); unfused = 2 extra passes. Reference:
     dense_module.py:784-995."""
 
+    # measured stream efficiencies of the shipped CDNA4 kernels
+    fwd_mem_op = "rmsnorm_fwd"
+    bwd_act_mem_op = "rmsnorm_bwd"
+
     def __init__(self, hidden_size, strategy, system, name="norm", norm_type="rms"):
         super().__init__(strategy, system, name)
         self.hidden_size = hidden_size
@@ -316,6 +320,9 @@ class RotaryEmbedding(MetaModule):
     """RoPE on q,k. Linear in x: bwd needs only cos/sin tables, so no cache.
     Fused CDNA4 kernel reads+writes q,k once. Reference:
     dense_module.py:1806-1873."""
+
+    fwd_mem_op = "rope"
+    bwd_act_mem_op = "rope"
 
     def __init__(self, strategy, system, name="rope"):
         super().__init__(strategy, system, name)
@@ -492,6 +499,9 @@ class Swiglu(MetaModule):
     """Fused SwiGLU: y = silu(x1) * x2 over the fc1 output's two halves.
     Caches its input (fc2 caches y itself). Reference:
     dense_module.py:1874-2096."""
+
+    fwd_mem_op = "swiglu"
+    bwd_act_mem_op = "swiglu_bwd"
 
     def __init__(self, strategy, system, name="swiglu", weighted=False):
         super().__init__(strategy, system, name)
