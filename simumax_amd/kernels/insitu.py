@@ -115,6 +115,12 @@ def summarize():
         elif table in ("sdp_fwd", "sdp_bwd"):
             stage = "fwd" if table == "sdp_fwd" else "bwd"
             row["eff"] = _sdp_flops(key, stage) / (t / 1e3) / (PEAK_BF16_TFLOPS * 1e12)
+        elif table == "group_matmul":
+            import re
+
+            m = re.match(r"ng=(\d+), M=(\d+), N=(\d+), K=(\d+)", key)
+            ng, mm, n, k = (int(m.group(i)) for i in range(1, 5))
+            row["eff"] = (2 * ng * mm * n * k) / (t / 1e3) / (PEAK_BF16_TFLOPS * 1e12)
         out[table][key] = row
     for op, (byt, ms, n) in bw_acc.items():
         # subtract launch latency so the efficiency composes with the
