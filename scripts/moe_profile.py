@@ -21,7 +21,7 @@ def main():
     toks, labels = make_synthetic_batch(mc.vocab_size, 2, 1, 4096, "cuda:0")
     train_step(m, opt, red, toks, labels, 2)
     torch.cuda.synchronize()
-    with profile(activities=[ProfilerActivity.CUDA],
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
                  record_shapes=True) as prof:
         train_step(m, opt, red, toks, labels, 2)
         torch.cuda.synchronize()
