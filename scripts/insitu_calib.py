@@ -38,6 +38,40 @@ CASES = [
 OUTDIR = "gpurun_out/calib"
 
 
+def measure_routing_chain(device="cuda:0"):
+    """Wall-clock the exact top-k routing chain of train/moe.py (softmax,
+    topk, argsort, bincount, cumsum, bookkeeping index ops) on the
+    mixtral shape: ~50 tiny host-launched kernels whose cost is launch-
+    bound, not bandwidth-bound. Returns ms per invocation."""
+    import math
+    import time
+
+    N, E, topk, cap_f = 4096, 8, 2, 1
+    logits = torch.randn(N, E, device=device, dtype=torch.float32)
+    torch.cuda.synchronize()
+    t0 = time.time()
+    ITER = 30
+    for _ in range(ITER):
+        probs = torch.softmax(logits, dim=-1)
+        weight, idx = probs.topk(topk, dim=-1)
+        weight = weight / weight.sum(-1, keepdim=True)
+        cap = int(math.ceil(N * topk / E * cap_f))
+        flat_expert = idx.reshape(-1)
+        flat_token = torch.arange(N, device=device).repeat_interleave(topk)
+        order = torch.argsort(flat_expert, stable=True)
+        counts = torch.bincount(flat_expert, minlength=E)
+        offs = torch.cumsum(counts, 0) - counts
+        rank_sorted = (torch.arange(N * topk, device=device)
+                       - offs[flat_expert[order]])
+        keep = rank_sorted < cap
+        src_tok = flat_token[order][keep]
+        dst_exp = flat_expert[order][keep]
+        slot_index = dst_exp * cap + rank_sorted[keep]
+        w_kept = weight.reshape(-1)[order][keep].to(torch.bfloat16)
+    torch.cuda.synchronize()
+    return (time.time() - t0) / ITER * 1e3
+
+
 def main():
     opt_samples = []
     for model, seq, mbs, mbc in sys.argv[1:] and [
@@ -91,6 +125,9 @@ def main():
             out = json.load(f)
     out["optimizer_eff"] = effs[len(effs) // 2]
     out["optimizer_samples"] = opt_samples
+    rt = measure_routing_chain()
+    out["moe_routing_ms"] = round(rt, 4)
+    print(f"[routing chain] {rt:.3f} ms per layer invocation")
     os.makedirs(OUTDIR, exist_ok=True)
     with open(path, "w") as f:
         json.dump(out, f, indent=1)

@@ -96,6 +96,12 @@ def main():
                     "latency_us": round(bw.get("launch_us", 4.0), 2)})
                 acc["bandwidth"][key]["efficient_factor"] = round(
                     bw[f"{key}_eff"], 4)
+        if "moe_routing_ms" in bw:
+            acc["bandwidth"]["moe_routing"] = {
+                "gbps": 8000.0,
+                "efficient_factor": 0.55,
+                "latency_us": round(bw["moe_routing_ms"] * 1e3, 1),
+            }
         if "optimizer_eff" in bw:
             acc["bandwidth"]["optimizer"] = {
                 "gbps": 8000.0,

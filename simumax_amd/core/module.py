@@ -177,7 +177,8 @@ class MetaModule:
         sysc = self.system
         comp = self._compute_info
 
-        def stage_time(op_name, stage, flops, mem, extra=0.0, mem_op=None):
+        def stage_time(op_name, stage, flops, mem, extra=0.0, mem_op=None,
+                       extra_op=None):
             c = sysc.compute_op_accuracy_time(
                 op_name, flops, shape_desc=self.get_input_shapes_desc(stage),
                 reture_detail=True,
@@ -186,21 +187,23 @@ class MetaModule:
                                               reture_detail=True)
             t = sysc.compute_end2end_time(c["compute_only_time"], io["io_time"])
             if extra > 0:
-                # glue kernels (layout copies, grad fan-in adds) run as
-                # separate launches at stream bandwidth, additive to the op
-                t += sysc.compute_mem_access_time("default", extra)
+                # glue kernels (layout copies, grad fan-in adds, host-bound
+                # small-launch chains) run as separate launches, additive
+                t += sysc.compute_mem_access_time(extra_op or "default", extra)
             return t
 
         ci = self._cost_info
         ci.fwd_compute_time = stage_time(self.fwd_op, "fwd", comp.fwd_flops,
                                          comp.fwd_accessed_mem,
                                          comp.fwd_extra_mem,
-                                         getattr(self, "fwd_mem_op", None))
+                                         getattr(self, "fwd_mem_op", None),
+                                         getattr(self, "fwd_extra_op", None))
         ci.bwd_grad_act_time = stage_time(self.bwd_act_op, "bwd_grad_act",
                                           comp.bwd_grad_act_flops,
                                           comp.bwd_grad_act_accessed_mem,
                                           comp.bwd_grad_act_extra_mem,
-                                          getattr(self, "bwd_act_mem_op", None))
+                                          getattr(self, "bwd_act_mem_op", None),
+                                          getattr(self, "bwd_act_extra_op", None))
         ci.bwd_grad_w_time = stage_time(self.bwd_w_op, "bwd_grad_w",
                                         comp.bwd_grad_w_flops,
                                         comp.bwd_grad_w_accessed_mem,

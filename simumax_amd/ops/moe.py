@@ -45,6 +45,9 @@ class Router(LinearBase, ParamMixin):
         logits = b * s * self.output_size * FP32
         info.activation_mem_cache = t.mem_bytes() + logits + b * s * self.topk * (FP32 + 4)
 
+    fwd_extra_op = "moe_routing"
+    bwd_act_extra_op = "moe_routing"
+
     def _leaf_compute_info(self, info):
         k = self.get_gemm_bmnk("fwd")
         flops = 2 * k["B"] * k["M"] * k["K"] * k["N"]
@@ -56,6 +59,12 @@ class Router(LinearBase, ParamMixin):
         info.fwd_accessed_mem = t.mem_bytes() + 3 * logits
         info.bwd_grad_act_accessed_mem = t.mem_bytes() + 3 * logits
         info.bwd_grad_w_accessed_mem = t.mem_bytes() + logits
+        # the routing chain (softmax/topk/argsort/bincount/cumsum + index
+        # bookkeeping) is a host-launch-bound run of ~50 tiny kernels; its
+        # calibrated per-layer latency lives in bandwidth["moe_routing"]
+        # (measured by scripts/insitu_calib.py). Priced once fwd, once bwd.
+        info.fwd_extra_mem = logits
+        info.bwd_grad_act_extra_mem = logits
 
 
 class Permutation(MetaModule):
@@ -100,8 +109,12 @@ class Permutation(MetaModule):
     def _leaf_compute_info(self, info):
         in_b = self.input_info.first.mem_bytes()
         out_b = self.output_info.first.mem_bytes()
-        info.fwd_accessed_mem = in_b * self.topk + out_b
-        info.bwd_grad_act_accessed_mem = in_b * self.topk + out_b
+        # trainer dispatch sequence (train/moe.py): zero-fill the padded
+        # buffer, index_select the kept tokens (r+w), index_copy into the
+        # capacity slots (r+w) -> ~5 passes over the expanded buffer
+        info.fwd_accessed_mem = 2 * out_b + 3 * max(in_b * self.topk, out_b)
+        # bwd: index_select the slot grads (r+w) + index_add into dX (r+r+w)
+        info.bwd_grad_act_accessed_mem = 2 * out_b + 3 * in_b
 
     def _comp_leaf_cost_info(self):
         sysc = self.system
@@ -160,8 +173,13 @@ class UnPermutation(MetaModule):
     def _leaf_compute_info(self, info):
         in_b = self.input_info.first.mem_bytes() // self.etp
         out_b = self.output_info.first.mem_bytes()
-        info.fwd_accessed_mem = in_b + out_b * self.topk
-        info.bwd_grad_act_accessed_mem = in_b + out_b * self.topk
+        # trainer combine (train/moe.py): index_select expert outputs (r+w),
+        # probs-weighted multiply (r+w), zero-fill + index_add into the
+        # token buffer (r+r+w)
+        info.fwd_accessed_mem = 4 * in_b + 3 * out_b
+        # bwd: index_select dout, multiply by probs, dprobs reduction
+        # (reads y and dout), index_add back to slot grads
+        info.bwd_grad_act_accessed_mem = 6 * in_b + 2 * out_b
         n = self.input_info.first.numel() // self.etp
         info.fwd_flops = 2 * n
         info.bwd_grad_act_flops = 2 * n
