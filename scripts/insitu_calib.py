@@ -33,6 +33,8 @@ CASES = [
     ("llama3-8b", 8192, 1, 2),
     ("llama3-70b-l12", 4096, 1, 2),
     ("mixtral-8x7b-l8", 4096, 1, 2),
+    ("llama3-8b", 16384, 1, 1),
+    ("qwen3-32b-l12", 4096, 1, 2),
 ]
 
 OUTDIR = "gpurun_out/calib"

@@ -27,6 +27,8 @@ CASES = [
     ("8b_seq8192_mbc2", "llama3-8b", 8192, 1, 2, 3),
     ("70b_l12_seq4096", "llama3-70b-l12", 4096, 1, 2, 3),
     ("mixtral_l8_moe", "mixtral-8x7b-l8", 4096, 1, 2, 3),
+    ("8b_seq16384_mbc1", "llama3-8b", 16384, 1, 1, 3),
+    ("qwen32b_l12_seq4096", "qwen3-32b-l12", 4096, 1, 2, 3),
 ]
 
 OUT = "gpurun_out/validation.jsonl"

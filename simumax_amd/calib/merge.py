@@ -13,11 +13,16 @@ SYSTEM = os.path.join(REPO, "configs", "system", "mi355x.json")
 
 
 def _load(name):
-    p = os.path.join(CALIB, name)
-    if os.path.exists(p):
-        with open(p) as f:
-            return json.load(f)
-    return {}
+    """Tracked baseline (calib_raw/, committed) overlaid by any fresh
+    measurement in gpurun_out/calib/ — so an on-GPU-box merge sees the
+    full calibration history even though gpurun_out/ never travels."""
+    out = {}
+    for d in (os.path.join(REPO, "calib_raw"), CALIB):
+        p = os.path.join(d, name)
+        if os.path.exists(p):
+            with open(p) as f:
+                out.update(json.load(f))
+    return out
 
 
 def _overlay_insitu(table, name):
