@@ -1,0 +1,102 @@
+"""Contract tests: the in-situ timing registry must emit byte-identical
+shape-key strings to the ones the simulator looks up, or the calibration
+overlay silently misses. Guards kernels/insitu.py against drift in
+core/module.py's key generators."""
+
+import pytest
+
+from simumax_amd.core.config import ModelConfig, StrategyConfig, SystemConfig
+from simumax_amd.kernels import insitu
+from simumax_amd import (PerfLLM, get_simu_model_config,
+                         get_simu_system_config)
+
+
+@pytest.fixture(scope="module")
+def perf():
+    mc = ModelConfig.init_from_config_file(get_simu_model_config("llama3-8b"))
+    mc.layer_num = 2
+    st = StrategyConfig(
+        seq_len=4096, micro_batch_size=1, micro_batch_num=1,
+        world_size=1, tp_size=1, pp_size=1,
+        enable_sequence_parallel=False, zero_state=0,
+        use_fp32_accum_grad=True, enable_recompute=False,
+        cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
+        mem_factor=1.0)
+    p = PerfLLM()
+    p.configure(st, mc, SystemConfig.init_from_config_file(
+        get_simu_system_config("mi355x")))
+    p.run_estimate()
+    return p
+
+
+def _leaf(perf, cls_name, name_part=""):
+    for leaf in perf.chunks[0].leaf_modules():
+        if type(leaf).__name__ == cls_name and name_part in leaf.full_name:
+            return leaf
+    raise AssertionError(f"{cls_name} {name_part} not found")
+
+
+def test_gemm_fwd_key_matches(perf):
+    qkv = _leaf(perf, "LinearCol", "qkv_proj")
+    sim_key = qkv.get_input_shapes_desc("fwd")
+    # the trainer calls FusedLinear with x [B,S,K] and weight [N,K]
+    ins_key = insitu.gemm_key(1, 4096, 4096, 6144, "TN", False, "bf16")
+    assert ins_key == sim_key
+
+
+def test_gemm_dgrad_key_matches(perf):
+    qkv = _leaf(perf, "LinearCol", "qkv_proj")
+    sim_key = qkv.get_input_shapes_desc("bwd_grad_act")
+    ins_key = insitu.gemm_key(1, 4096, 6144, 4096, "NN", False, "bf16")
+    assert ins_key == sim_key
+
+
+def test_gemm_wgrad_key_matches(perf):
+    qkv = _leaf(perf, "LinearCol", "qkv_proj")
+    sim_key = qkv.get_input_shapes_desc("bwd_grad_w")
+    ins_key = insitu.gemm_key(1, 6144, 4096, 4096, "NT", True, "fp32")
+    assert ins_key == sim_key
+
+
+def test_sdp_key_matches(perf):
+    core = _leaf(perf, "CoreAttention")
+    sim_key = core.get_input_shapes_desc("fwd")
+    # the trainer calls flash_attention(q [B,S,Hq,D], k/v [B,S,Hkv,D])
+    ins_key = insitu.sdp_key(1, 4096, 32, 8, 128, 128)
+    assert ins_key == sim_key
+
+
+def test_grouped_key_matches():
+    from simumax_amd.train.moe import _grouped_key
+
+    mc = ModelConfig.init_from_config_file(
+        get_simu_model_config("mixtral-8x7b-l8"))
+    st = StrategyConfig(
+        seq_len=4096, micro_batch_size=1, micro_batch_num=1,
+        world_size=1, tp_size=1, pp_size=1,
+        enable_sequence_parallel=False, zero_state=0,
+        use_fp32_accum_grad=True, enable_recompute=False,
+        cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
+        mem_factor=1.0)
+    p = PerfLLM()
+    p.configure(st, mc, SystemConfig.init_from_config_file(
+        get_simu_system_config("mi355x")))
+    p.run_estimate()
+    gl = _leaf(p, "GroupLinearCol")
+    sim_key = gl.get_input_shapes_desc("fwd")
+    # trainer: x [E, cap, H] @ w1 [E, H, 2I]; cap = N*topk/E
+    ins_key = _grouped_key(8, 1024, 4096, 2 * 14336, "fwd")
+    assert ins_key == sim_key
+
+
+def test_efficiency_definitions_roundtrip():
+    """insitu eff -> merge -> SystemConfig pricing must reproduce the
+    measured time for a GEMM shape."""
+    key = insitu.gemm_key(1, 4096, 4096, 6144, "TN", False, "bf16")
+    flops = insitu._gemm_flops(key)
+    t_ms = 0.25
+    eff = flops / (t_ms / 1e3) / (insitu.PEAK_BF16_TFLOPS * 1e12)
+    sysc = SystemConfig.init_from_config_file(get_simu_system_config("mi355x"))
+    sysc.accelerator.op["matmul"].accurate_efficient_factor[key] = eff
+    t = sysc.compute_op_accuracy_time("matmul", flops, shape_desc=key)
+    assert abs(t - t_ms) / t_ms < 1e-9
