@@ -100,6 +100,43 @@ class SimuSystem:
                         executed += 1
                         progressed = True
                         continue
+                    # async p2p (strategy.pp_comm_async): the receiver
+                    # posted its irecv one unit ahead; the wait job pays
+                    # only the remaining transfer time
+                    if job.kind == "p2p_post_recv":
+                        t = lane_t[r]["comp"]
+                        done.setdefault(job.gid + "#rposted", t)
+                        self.log.append(LogEvent(r, job.name, "p2p", "comm",
+                                                 t, t, job.mb, job.call_stack,
+                                                 t, None))
+                        ptr[r] += 1
+                        executed += 1
+                        progressed = True
+                        continue
+                    if job.kind == "p2p_wait":
+                        key = job.gid + "#posted"
+                        if key not in done:
+                            break  # sender has not posted yet
+                        rpost = done.get(job.gid + "#rposted",
+                                         lane_t[r]["comp"])
+                        # the transfer runs on the network from the moment
+                        # both sides have posted; the wait only blocks the
+                        # receiver until the data is ready
+                        xfer_start = max(done[key], rpost)
+                        ready = xfer_start + job.dur
+                        arrive = lane_t[r]["comp"]
+                        start = xfer_start
+                        end = max(ready, arrive)
+                        lane_t[r]["comm"] = max(lane_t[r]["comm"], end)
+                        lane_t[r]["comp"] = max(lane_t[r]["comp"], end)
+                        self.log.append(LogEvent(r, job.name, "p2p", "comm",
+                                                 start, end, job.mb,
+                                                 job.call_stack, arrive,
+                                                 job.mem))
+                        ptr[r] += 1
+                        executed += 1
+                        progressed = True
+                        continue
                     # p2p: eager send (sender posts and proceeds; receiver
                     # pays the transfer after the send is posted) — matches
                     # the analytic recurrence F_dep = f_end[prev] + p2p and

@@ -99,6 +99,45 @@ def _leaf_jobs_bwd(chunk, stage, mb):
     return jobs
 
 
+def _assemble_units(units, r, async_p2p):
+    """Flatten per-unit job lists into one rank stream.
+
+    Sync mode: recv jobs block at their original position (kind "p2p").
+    Async mode (strategy.pp_comm_async, Megatron irecv/isend semantics):
+    each unit's recv is POSTED one unit ahead (free) and a WAIT sits at
+    the original position, so the transfer overlaps the preceding
+    compute — reference parity: pipeline_schedule.py:200-267 (batched
+    async bundles).
+    """
+    def is_recv(j):
+        return j.kind == "p2p" and j.peers and j.peers[1] == r
+
+    if not async_p2p:
+        return [j for u in units for j in u]
+
+    def post_of(j):
+        return Job(name=j.name.replace("recv", "post_recv"),
+                   kind="p2p_post_recv", dur=0.0, lane="comm", mb=j.mb,
+                   gid=j.gid, peers=j.peers, call_stack=j.call_stack)
+
+    def wait_of(j):
+        return Job(name=j.name.replace("recv", "wait_recv"),
+                   kind="p2p_wait", dur=j.dur, lane="comm", mb=j.mb,
+                   gid=j.gid, peers=j.peers, mem=j.mem,
+                   call_stack=j.call_stack)
+
+    out = []
+    for i, u in enumerate(units):
+        if i == 0 and u and is_recv(u[0]):
+            out.append(post_of(u[0]))
+        nxt = units[i + 1] if i + 1 < len(units) else None
+        if nxt and nxt and is_recv(nxt[0]):
+            out.append(post_of(nxt[0]))
+        for j in u:
+            out.append(wait_of(j) if is_recv(j) else j)
+    return out
+
+
 class PpSchedule:
     """Builds the per-(simulated-)rank 1F1B job lists."""
 
@@ -146,35 +185,39 @@ class PpSchedule:
                     stream.append(("F", nf)); nf += 1
                 stream.append(("B", nb)); nb += 1
             for r in stage_ranks:
+                units = []
                 for kind, m in stream:
+                    u = []
                     if kind == "F":
                         if stage > 0:
-                            jobs[r].append(Job(
+                            u.append(Job(
                                 name=f"recv_fwd.mb{m}", kind="p2p",
                                 dur=p2p_time, lane="comm", mb=m,
                                 gid=f"p2p-f-mb{m}-{stage-1}-{stage}",
                                 peers=(per_stage_rank[stage - 1], r)))
-                        jobs[r].extend(_leaf_jobs_fwd(chunk, stage, m))
+                        u.extend(_leaf_jobs_fwd(chunk, stage, m))
                         if stage < pp - 1:
-                            jobs[r].append(Job(
+                            u.append(Job(
                                 name=f"send_fwd.mb{m}", kind="p2p",
                                 dur=p2p_time, lane="comm", mb=m,
                                 gid=f"p2p-f-mb{m}-{stage}-{stage+1}",
                                 peers=(r, per_stage_rank[stage + 1])))
                     else:
                         if stage < pp - 1:
-                            jobs[r].append(Job(
+                            u.append(Job(
                                 name=f"recv_bwd.mb{m}", kind="p2p",
                                 dur=p2p_time, lane="comm", mb=m,
                                 gid=f"p2p-b-mb{m}-{stage+1}-{stage}",
                                 peers=(per_stage_rank[stage + 1], r)))
-                        jobs[r].extend(_leaf_jobs_bwd(chunk, stage, m))
+                        u.extend(_leaf_jobs_bwd(chunk, stage, m))
                         if stage > 0:
-                            jobs[r].append(Job(
+                            u.append(Job(
                                 name=f"send_bwd.mb{m}", kind="p2p",
                                 dur=p2p_time, lane="comm", mb=m,
                                 gid=f"p2p-b-mb{m}-{stage}-{stage-1}",
                                 peers=(r, per_stage_rank[stage - 1])))
+                    units.append(u)
+                jobs[r].extend(_assemble_units(units, r, s.pp_comm_async))
         self._append_optimizer(jobs, ranks, per_stage_rank)
         return jobs
 
@@ -206,40 +249,44 @@ class PpSchedule:
                 if nf < total:
                     stream.append(("F", nf)); nf += 1
                 stream.append(("B", nb)); nb += 1
+            units = []
             for kind, k in stream:
                 fwd = kind == "F"
                 c = chunk_id_of(k, pp, vp, fwd)
                 m = mb_id_of(k, pp, vp)
                 v = c * pp + stage
                 chunk = self.perf.vchunks[stage][c]
+                u = []
                 if fwd:
                     if v > 0:
                         src = per_stage_rank[stage - 1 if stage > 0 else pp - 1]
-                        jobs[r].append(Job(
+                        u.append(Job(
                             name=f"recv_fwd.v{v}.mb{m}", kind="p2p",
                             dur=p2p_time, lane="comm", mb=m,
                             gid=f"p2p-f-mb{m}-v{v-1}-v{v}", peers=(src, r)))
-                    jobs[r].extend(_leaf_jobs_fwd(chunk, stage, m * vp + c))
+                    u.extend(_leaf_jobs_fwd(chunk, stage, m * vp + c))
                     if v < nv - 1:
                         dst = per_stage_rank[stage + 1 if stage < pp - 1 else 0]
-                        jobs[r].append(Job(
+                        u.append(Job(
                             name=f"send_fwd.v{v}.mb{m}", kind="p2p",
                             dur=p2p_time, lane="comm", mb=m,
                             gid=f"p2p-f-mb{m}-v{v}-v{v+1}", peers=(r, dst)))
                 else:
                     if v < nv - 1:
                         src = per_stage_rank[stage + 1 if stage < pp - 1 else 0]
-                        jobs[r].append(Job(
+                        u.append(Job(
                             name=f"recv_bwd.v{v}.mb{m}", kind="p2p",
                             dur=p2p_time, lane="comm", mb=m,
                             gid=f"p2p-b-mb{m}-v{v+1}-v{v}", peers=(src, r)))
-                    jobs[r].extend(_leaf_jobs_bwd(chunk, stage, m * vp + c))
+                    u.extend(_leaf_jobs_bwd(chunk, stage, m * vp + c))
                     if v > 0:
                         dst = per_stage_rank[stage - 1 if stage > 0 else pp - 1]
-                        jobs[r].append(Job(
+                        u.append(Job(
                             name=f"send_bwd.v{v}.mb{m}", kind="p2p",
                             dur=p2p_time, lane="comm", mb=m,
                             gid=f"p2p-b-mb{m}-v{v}-v{v-1}", peers=(r, dst)))
+                units.append(u)
+            jobs[r].extend(_assemble_units(units, r, s.pp_comm_async))
         self._append_optimizer(jobs, ranks, per_stage_rank)
         return jobs
 
