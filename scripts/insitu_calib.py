@@ -35,6 +35,7 @@ CASES = [
     ("mixtral-8x7b-l8", 4096, 1, 2),
     ("llama3-8b", 16384, 1, 1),
     ("qwen3-32b-l12", 4096, 1, 2),
+    ("deepseekv2-l4", 4096, 1, 2),
 ]
 
 OUTDIR = "gpurun_out/calib"

@@ -29,6 +29,7 @@ CASES = [
     ("mixtral_l8_moe", "mixtral-8x7b-l8", 4096, 1, 2, 3),
     ("8b_seq16384_mbc1", "llama3-8b", 16384, 1, 1, 3),
     ("qwen32b_l12_seq4096", "qwen3-32b-l12", 4096, 1, 2, 3),
+    ("deepseekv2_l4_mla_moe", "deepseekv2-l4", 4096, 1, 2, 3),
 ]
 
 OUT = "gpurun_out/validation.jsonl"

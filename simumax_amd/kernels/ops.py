@@ -337,7 +337,8 @@ class _FlashAttnFn(torch.autograd.Function):
             if insitu.ENABLED:
                 stop = insitu.start("sdp_fwd", insitu.sdp_key(
                     q.shape[0], q.shape[1], q.shape[2], k.shape[2],
-                    q.shape[3], v.shape[3]))
+                    q.shape[3], v.shape[3],
+                    contiguous=q.shape[3] == v.shape[3]))
                 o, lse = ext().fa_fwd(q, k, v, causal)
                 stop()
             else:
@@ -357,7 +358,8 @@ class _FlashAttnFn(torch.autograd.Function):
             if insitu.ENABLED:
                 stop = insitu.start("sdp_bwd", insitu.sdp_key(
                     q.shape[0], q.shape[1], q.shape[2], k.shape[2],
-                    q.shape[3], v.shape[3]))
+                    q.shape[3], v.shape[3],
+                    contiguous=q.shape[3] == v.shape[3]))
                 dq, dk, dv = ext().fa_bwd(do.contiguous(), q, k, v, o, lse,
                                           ctx.causal)
                 stop()

@@ -494,6 +494,14 @@ class MLACoreAttention(CoreAttention):
         super().__init__(head_num, head_num, qk_head_dim, v_head_dim,
                          strategy, system, name, qkv_contiguous=False)
 
+    def _leaf_compute_info(self, info):
+        super()._leaf_compute_info(info)
+        # trainer parity: q/k are materialized by concatenating the nope
+        # and RoPE sub-dims (fwd) and split back in backward
+        q, k, v, o, _ = self._sdp_bytes()
+        info.fwd_extra_mem = 2 * (q + k) + 2 * v  # cats + v contiguous
+        info.bwd_grad_act_extra_mem = 2 * (q + k)
+
 
 class Swiglu(MetaModule):
     """Fused SwiGLU: y = silu(x1) * x2 over the fc1 output's two halves.

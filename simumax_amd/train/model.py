@@ -19,7 +19,9 @@ from ..core.config import ModelConfig
 from ..kernels import ops as K
 
 
-class LlamaDecoderLayer(nn.Module):
+class GQAAttention(nn.Module):
+    """Fused-qkv GQA attention (Llama/Qwen/Mixtral family)."""
+
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None):
         super().__init__()
         h = cfg.hidden_size
@@ -27,26 +29,12 @@ class LlamaDecoderLayer(nn.Module):
         self.kv_heads = cfg.kv_head_num
         self.head_size = cfg.head_size
         qkv_out = (cfg.head_num + 2 * cfg.kv_head_num) * cfg.head_size
-        self.input_norm = K.RMSNorm(h, dtype=dtype, device=device)
         self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype, device=device)
         self.out_proj = K.FusedLinear(cfg.head_num * cfg.head_size, h,
                                       dtype=dtype, device=device)
-        self.pre_mlp_norm = K.RMSNorm(h, dtype=dtype, device=device)
-        assert cfg.use_swiglu
-        self.use_moe = cfg.model_type == "moe"
-        if self.use_moe:
-            from .moe import MoEMLP
 
-            self.moe_mlp = MoEMLP(cfg, dtype=dtype, device=device)
-        else:
-            self.fc1 = K.FusedLinear(h, 2 * cfg.intermediate_size, dtype=dtype, device=device)
-            self.fc2 = K.FusedLinear(cfg.intermediate_size, h, dtype=dtype, device=device)
-
-    def forward(self, x, rope_cs, pos):
-        # x: [B, S, H]
-        B, S, H = x.shape
-        res = x
-        y = self.input_norm(x)
+    def forward(self, y, rope_cs, pos):
+        B, S, _ = y.shape
         qkv = self.qkv_proj(y)
         d = self.head_size
         q, k, v = qkv.split(
@@ -57,7 +45,83 @@ class LlamaDecoderLayer(nn.Module):
         k = k.view(B, S, self.kv_heads, d)
         v = v.reshape(B, S, self.kv_heads, d)
         ctx = K.flash_attention(q, k, v, causal=True)
-        x = res + self.out_proj(ctx.reshape(B, S, self.heads * d))
+        return self.out_proj(ctx.reshape(B, S, self.heads * d))
+
+
+class MLAAttention(nn.Module):
+    """DeepSeek multi-head latent attention: low-rank q/kv projections,
+    RoPE only on the positional sub-dims, asymmetric-head flash SDP
+    (Dqk = qk_head_dim + qk_pos_emb_head_dim, Dv = v_head_dim). Mirrors
+    the simulator's MLAAttention op graph (ops/dense.py MLAAttention)."""
+
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None):
+        super().__init__()
+        h = cfg.hidden_size
+        self.heads = cfg.head_num
+        self.dn = cfg.qk_head_dim            # nope part (128)
+        self.dp = cfg.qk_pos_emb_head_dim    # rope part (64)
+        self.dv = cfg.v_head_dim
+        self.kv_lora = cfg.kv_lora_rank
+        qk_total = self.dn + self.dp
+        assert cfg.q_lora_rank, "MLA trainer assumes q_lora_rank (DeepSeek-V2+)"
+        self.q_down = K.FusedLinear(h, cfg.q_lora_rank, dtype=dtype,
+                                    device=device)
+        self.q_norm = K.RMSNorm(cfg.q_lora_rank, dtype=dtype, device=device)
+        self.q_up = K.FusedLinear(cfg.q_lora_rank, self.heads * qk_total,
+                                  dtype=dtype, device=device)
+        self.kv_down = K.FusedLinear(h, self.kv_lora + self.dp, dtype=dtype,
+                                     device=device)
+        self.kv_norm = K.RMSNorm(self.kv_lora, dtype=dtype, device=device)
+        self.kv_up = K.FusedLinear(self.kv_lora,
+                                   self.heads * (self.dn + self.dv),
+                                   dtype=dtype, device=device)
+        self.out_proj = K.FusedLinear(self.heads * self.dv, h, dtype=dtype,
+                                      device=device)
+
+    def forward(self, y, rope_cs, pos):
+        B, S, _ = y.shape
+        H, dn, dp, dv = self.heads, self.dn, self.dp, self.dv
+        q = self.q_up(self.q_norm(self.q_down(y))).view(B * S, H, dn + dp)
+        q_pe = K.apply_rope(q[..., dn:].contiguous(), rope_cs, pos)
+        kv = self.kv_down(y)
+        k_pe = K.apply_rope(
+            kv[..., self.kv_lora:].reshape(B * S, 1, dp).contiguous(),
+            rope_cs, pos)
+        kvu = self.kv_up(self.kv_norm(kv[..., :self.kv_lora]))             .view(B * S, H, dn + dv)
+        qf = torch.cat([q[..., :dn], q_pe], dim=-1).view(B, S, H, dn + dp)
+        kf = torch.cat([kvu[..., :dn], k_pe.expand(B * S, H, dp)], dim=-1)             .view(B, S, H, dn + dp)
+        v = kvu[..., dn:].reshape(B, S, H, dv).contiguous()
+        ctx = K.flash_attention(qf, kf, v, causal=True)
+        return self.out_proj(ctx.reshape(B, S, H * dv))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
+                 layer_idx=0):
+        super().__init__()
+        h = cfg.hidden_size
+        self.input_norm = K.RMSNorm(h, dtype=dtype, device=device)
+        if getattr(cfg, "attention_type", "gqa") == "mla":
+            self.attention = MLAAttention(cfg, dtype=dtype, device=device)
+        else:
+            self.attention = GQAAttention(cfg, dtype=dtype, device=device)
+        self.pre_mlp_norm = K.RMSNorm(h, dtype=dtype, device=device)
+        assert cfg.use_swiglu
+        self.use_moe = (cfg.model_type == "moe"
+                        and layer_idx >= (cfg.dense_layers or 0))
+        if self.use_moe:
+            from .moe import MoEMLP
+
+            self.moe_mlp = MoEMLP(cfg, dtype=dtype, device=device)
+        else:
+            self.fc1 = K.FusedLinear(h, 2 * cfg.intermediate_size, dtype=dtype, device=device)
+            self.fc2 = K.FusedLinear(cfg.intermediate_size, h, dtype=dtype, device=device)
+
+    def forward(self, x, rope_cs, pos):
+        # x: [B, S, H]
+        res = x
+        y = self.input_norm(x)
+        x = res + self.attention(y, rope_cs, pos)
         res = x
         y = self.pre_mlp_norm(x)
         if self.use_moe:
@@ -76,11 +140,15 @@ class LlamaForTraining(nn.Module):
         self.embedding = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
                                       dtype=dtype, device=device)
         self.layers = nn.ModuleList(
-            [LlamaDecoderLayer(cfg, dtype, device) for _ in range(cfg.layer_num)])
+            [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i)
+             for i in range(cfg.layer_num)])
         self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
         self.lm_head = K.FusedLinear(cfg.hidden_size, cfg.vocab_size,
                                      dtype=dtype, device=device)
-        cs = K.build_rope_cache(seq_len, cfg.head_size, base=rope_base,
+        rope_dim = (cfg.qk_pos_emb_head_dim
+                    if getattr(cfg, "attention_type", "gqa") == "mla"
+                    else cfg.head_size)
+        cs = K.build_rope_cache(seq_len, rope_dim, base=rope_base,
                                 device=device or "cpu")
         self.register_buffer("rope_cs", cs, persistent=False)
 

@@ -125,6 +125,16 @@ class MoEMLP(nn.Module):
         nn.init.normal_(self.w2, std=0.02)
         self.w1._fused_wgrad = True
         self.w2._fused_wgrad = True
+        shared_i = getattr(cfg, "moe_shared_expert_intermediate_size", 0)
+        self.shared = None
+        if shared_i:
+            from ..kernels.ops import FusedLinear
+
+            self.shared_fc1 = FusedLinear(h, 2 * shared_i, dtype=dtype,
+                                          device=device)
+            self.shared_fc2 = FusedLinear(shared_i, h, dtype=dtype,
+                                          device=device)
+            self.shared = True
 
     def forward(self, x):
         from ..kernels import ops as K
@@ -170,4 +180,8 @@ class MoEMLP(nn.Module):
         out = torch.zeros_like(xf)
         out.index_add_(0, src_tok,
                        y_flat.index_select(0, slot_index) * w_kept[:, None])
-        return out.view(B, S, H)
+        out = out.view(B, S, H)
+        if self.shared:
+            out = out + self.shared_fc2(
+                K.swiglu(self.shared_fc1(x)))
+        return out
