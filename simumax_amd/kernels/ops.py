@@ -56,6 +56,7 @@ def _use_hip(*tensors):
 class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
+        x = x.contiguous()  # saved for backward; the kernel requires it
         if _use_hip(x):
             if insitu.ENABLED:
                 stop = insitu.start("bw_rmsnorm_fwd", str(2 * x.numel() * x.element_size()))
