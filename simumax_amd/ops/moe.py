@@ -46,7 +46,7 @@ class Router(LinearBase, ParamMixin):
         info.activation_mem_cache = t.mem_bytes() + logits + b * s * self.topk * (FP32 + 4)
 
     fwd_extra_op = "moe_routing"
-    bwd_act_extra_op = "moe_routing"
+    bwd_act_extra_op = "moe_routing_bwd"
 
     def _leaf_compute_info(self, info):
         k = self.get_gemm_bmnk("fwd")
