@@ -216,6 +216,7 @@ def test_moe_mlp_gpu_numerics_and_training():
 def test_mla_attention_module_gpu_vs_cpu():
     """MLA attention module (low-rank q/kv, positional-subdim RoPE,
     asymmetric 48/32-dim flash SDP) on GPU bf16 vs the CPU reference."""
+    from simumax_amd.core.config import ModelConfig
     from simumax_amd.kernels.ops import build_rope_cache
     from simumax_amd.train.model import MLAAttention
 
@@ -247,6 +248,7 @@ def test_mla_attention_module_gpu_vs_cpu():
 @pytest.mark.gpu
 def test_mla_moe_trainer_gpu():
     """Tiny DeepSeek-shaped model (MLA + shared-expert MoE) trains on GPU."""
+    from simumax_amd.core.config import ModelConfig
     from simumax_amd.train.trainer import (TrainConfig, build_trainer,
                                            make_synthetic_batch, train_step)
 
