@@ -220,15 +220,17 @@ def test_mla_attention_module_gpu_vs_cpu():
     from simumax_amd.kernels.ops import build_rope_cache
     from simumax_amd.train.model import MLAAttention
 
+    # the FA kernel supports Dqk in {128, 192}: use the real MLA dims
+    # (128 nope + 64 rope, Dv 128) with few heads to keep the test small
     cfg = ModelConfig(
         model_type="dense", attention_type="mla", hidden_size=256,
-        head_num=8, kv_head_num=8, head_size=32, intermediate_size=512,
+        head_num=4, kv_head_num=4, head_size=128, intermediate_size=512,
         layer_num=1, vocab_size=1000, use_swiglu=True,
-        v_head_dim=32, qk_head_dim=32, qk_pos_emb_head_dim=16,
+        v_head_dim=128, qk_head_dim=128, qk_pos_emb_head_dim=64,
         q_lora_rank=96, kv_lora_rank=64)
     torch.manual_seed(7)
     att = MLAAttention(cfg, device=DEV)
-    cs = build_rope_cache(64, 16, device=DEV)
+    cs = build_rope_cache(64, 64, device=DEV)
     pos = torch.arange(64, device=DEV, dtype=torch.int32).repeat(2)
     x = torch.randn(2, 64, 256, device=DEV, dtype=torch.bfloat16,
                     requires_grad=True)
@@ -254,10 +256,10 @@ def test_mla_moe_trainer_gpu():
 
     cfg = ModelConfig(
         model_type="moe", attention_type="mla", hidden_size=256,
-        head_num=8, kv_head_num=8, head_size=32, intermediate_size=512,
+        head_num=4, kv_head_num=4, head_size=128, intermediate_size=512,
         moe_ffn_hidden_size=128, moe_shared_expert_intermediate_size=96,
         layer_num=2, dense_layers=1, expert_num=4, topk=2,
-        v_head_dim=32, qk_head_dim=32, qk_pos_emb_head_dim=16,
+        v_head_dim=128, qk_head_dim=128, qk_pos_emb_head_dim=64,
         q_lora_rank=96, kv_lora_rank=64, vocab_size=1024, use_swiglu=True)
     tc = TrainConfig(seq_len=128, micro_batch_size=1, micro_batch_num=2,
                      lr=3e-4)
