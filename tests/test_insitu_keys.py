@@ -100,3 +100,61 @@ def test_efficiency_definitions_roundtrip():
     sysc.accelerator.op["matmul"].accurate_efficient_factor[key] = eff
     t = sysc.compute_op_accuracy_time("matmul", flops, shape_desc=key)
     assert abs(t - t_ms) / t_ms < 1e-9
+
+
+def test_bandwidth_summarize_units():
+    """bw_<op> records aggregate to a stream efficiency vs the 8 TB/s
+    peak (minus per-launch latency)."""
+    import simumax_amd.kernels.insitu as I
+
+    class FakeEvt:
+        def __init__(self, t):
+            self.t = t
+
+        def elapsed_time(self, other):
+            return other.t - self.t
+
+    I._RECORDS.clear()
+    # 4 GiB moved in 1 ms -> 4000 GiB/s -> eff 0.5 of 8000 GiB/s
+    I._RECORDS[("bw_rmsnorm_fwd", str(4 * 1024**3))] = [
+        (FakeEvt(0.0), FakeEvt(1.0 + 0.004))]
+    import torch
+
+    sync = torch.cuda.synchronize
+    torch.cuda.synchronize = lambda: None
+    try:
+        out = I.summarize()
+    finally:
+        torch.cuda.synchronize = sync
+        I._RECORDS.clear()
+    eff = out["bandwidth"]["rmsnorm_fwd_eff"]
+    assert abs(eff - 0.5) < 1e-3, eff
+
+
+def test_mla_sdp_key_matches():
+    """The flash wrapper's qkv_contiguous heuristic (Dqk != Dv -> False)
+    must match MLACoreAttention's key."""
+    from simumax_amd.core.config import ModelConfig
+    from simumax_amd.kernels import insitu
+    from simumax_amd import (PerfLLM, StrategyConfig, SystemConfig,
+                             get_simu_model_config, get_simu_system_config)
+
+    mc = ModelConfig.init_from_config_file(
+        get_simu_model_config("deepseekv2-l4"))
+    st = StrategyConfig(
+        seq_len=4096, micro_batch_size=1, micro_batch_num=1,
+        world_size=1, tp_size=1, pp_size=1, ep_size=1,
+        enable_sequence_parallel=False, zero_state=0,
+        use_fp32_accum_grad=True, enable_recompute=False,
+        cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
+        mem_factor=1.0)
+    p = PerfLLM()
+    p.configure(st, mc, SystemConfig.init_from_config_file(
+        get_simu_system_config("mi355x")))
+    p.run_estimate()
+    core = next(l for l in p.chunks[0].leaf_modules()
+                if type(l).__name__ == "MLACoreAttention")
+    sim_key = core.get_input_shapes_desc("fwd")
+    ins_key = insitu.sdp_key(1, 4096, 128, 128, 192, 128,
+                             contiguous=192 == 128)
+    assert ins_key == sim_key

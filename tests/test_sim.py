@@ -155,3 +155,27 @@ def test_async_p2p_matches_analytic():
         t_async, analytic = run(True, vp)
         assert t_async <= t_sync + 1e-6
         assert abs(t_async - analytic) / analytic < 0.02
+
+
+def test_assemble_units_async_ordering():
+    """Async assembly: each unit's recv is posted one unit ahead and the
+    wait sits where the sync recv was."""
+    from simumax_amd.sim.schedule import _assemble_units
+
+    def recv(i):
+        return Job(name=f"recv{i}", kind="p2p", dur=1.0, lane="comm",
+                   gid=f"g{i}", peers=(0, 1))
+
+    def comp(i):
+        return Job(name=f"comp{i}", kind="fwd", dur=5.0)
+
+    units = [[recv(0), comp(0)], [recv(1), comp(1)]]
+    sync = _assemble_units(units, 1, False)
+    assert [j.name for j in sync] == ["recv0", "comp0", "recv1", "comp1"]
+    a = _assemble_units(units, 1, True)
+    names = [j.name for j in a]
+    assert names == ["post_recv0", "post_recv1", "wait_recv0", "comp0",
+                     "wait_recv1", "comp1"]
+    kinds = [j.kind for j in a]
+    assert kinds == ["p2p_post_recv", "p2p_post_recv", "p2p_wait", "fwd",
+                     "p2p_wait", "fwd"]
