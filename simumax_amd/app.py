@@ -93,6 +93,14 @@ def cmd_configs(args):
     print("system:  ", ", ".join(show_simu_system_configs()))
 
 
+def cmd_serve(args):
+    import uvicorn
+
+    from .webapp import app as webapp
+
+    uvicorn.run(webapp, host=args.host, port=args.port)
+
+
 def main(argv=None):
     ap = argparse.ArgumentParser(prog="simumax_amd", description=__doc__,
                                  formatter_class=argparse.RawDescriptionHelpFormatter)
@@ -121,6 +129,10 @@ def main(argv=None):
     sp.set_defaults(fn=cmd_search)
     sp = sub.add_parser("configs")
     sp.set_defaults(fn=cmd_configs)
+    sp = sub.add_parser("serve", help="launch the web front-end")
+    sp.add_argument("--host", default="127.0.0.1")
+    sp.add_argument("--port", type=int, default=8642)
+    sp.set_defaults(fn=cmd_serve)
 
     args = ap.parse_args(argv)
     args.fn(args)
