@@ -19,7 +19,8 @@ OUT = os.path.join(HERE, "simumax_hip.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
 KERNEL_SOURCES = ["rmsnorm.hip", "rope.hip", "swiglu.hip", "cross_entropy.hip",
-                  "attention.hip", "mfma_probe.hip", "gemm_bench.hip"]
+                  "attention.hip", "mfma_probe.hip", "gemm_bench.hip",
+                  "grouped_gemm.hip"]
 
 
 def _newest_mtime(paths):

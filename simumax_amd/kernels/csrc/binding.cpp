@@ -194,6 +194,55 @@ std::vector<torch::Tensor> fa_bwd(torch::Tensor dout, torch::Tensor q,
 
 void wgrad_accum(torch::Tensor dout, torch::Tensor x, torch::Tensor main_grad);
 
+extern "C" {
+void gg_fwd(const void *, const void *, void *, int, int, int, int,
+            hipStream_t);
+void gg_dgrad(const void *, const void *, void *, int, int, int, int,
+              hipStream_t);
+void gg_wgrad(const void *, const void *, void *, int, int, int, int,
+              hipStream_t);
+}
+
+// grouped expert GEMMs (see grouped_gemm.hip); weights are [E, N, K]
+torch::Tensor grouped_fwd(torch::Tensor x, torch::Tensor w) {
+    CHECK_IN(x);
+    CHECK_IN(w);
+    TORCH_CHECK(x.dim() == 3 && w.dim() == 3, "x [E,M,K], w [E,N,K]");
+    const int E = x.size(0), M = x.size(1), K = x.size(2), N = w.size(1);
+    TORCH_CHECK(w.size(0) == E && (int)w.size(2) == K);
+    TORCH_CHECK(N % 128 == 0 && K % 32 == 0, "grouped fwd needs N%128, K%32");
+    auto c = torch::empty({E, M, N}, x.options());
+    gg_fwd(x.data_ptr(), w.data_ptr(), c.data_ptr(), E, M, N, K,
+           cur_stream());
+    return c;
+}
+
+torch::Tensor grouped_dgrad(torch::Tensor dout, torch::Tensor w) {
+    CHECK_IN(dout);
+    CHECK_IN(w);
+    const int E = dout.size(0), M = dout.size(1), N = dout.size(2);
+    const int K = w.size(2);
+    TORCH_CHECK((int)w.size(1) == N);
+    TORCH_CHECK(K % 128 == 0 && N % 32 == 0, "grouped dgrad needs K%128, N%32");
+    auto dx = torch::empty({E, M, K}, dout.options());
+    gg_dgrad(dout.data_ptr(), w.data_ptr(), dx.data_ptr(), E, M, N, K,
+             cur_stream());
+    return dx;
+}
+
+void grouped_wgrad(torch::Tensor dout, torch::Tensor x, torch::Tensor g) {
+    CHECK_IN(dout);
+    CHECK_IN(x);
+    CHECK_IN(g);
+    const int E = dout.size(0), M = dout.size(1), N = dout.size(2);
+    const int K = x.size(2);
+    TORCH_CHECK(g.scalar_type() == torch::kFloat32, "main_grad must be fp32");
+    TORCH_CHECK((int)g.size(1) == N && (int)g.size(2) == K);
+    TORCH_CHECK(N % 128 == 0 && K % 128 == 0, "grouped wgrad needs N%128, K%128");
+    gg_wgrad(dout.data_ptr(), x.data_ptr(), g.data_ptr(), E, M, N, K,
+             cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("wgrad_accum", &wgrad_accum,
           "main_grad(fp32) += dout^T @ x (bf16 in, hipBLAS GemmEx)");
@@ -207,4 +256,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward (gfx950)");
     m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (gfx950)");
     m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward (gfx950)");
+    m.def("grouped_fwd", &grouped_fwd,
+          "grouped expert GEMM fwd: [E,M,K]x[E,N,K]^T -> [E,M,N] (MFMA)");
+    m.def("grouped_dgrad", &grouped_dgrad,
+          "grouped expert GEMM dgrad: [E,M,N]x[E,N,K] -> [E,M,K] (MFMA)");
+    m.def("grouped_wgrad", &grouped_wgrad,
+          "grouped expert wgrad: fp32 G[E,N,K] += dout^T @ x (MFMA)");
 }
