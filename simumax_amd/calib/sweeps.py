@@ -222,12 +222,13 @@ def parse_group_key(desc):
 
 
 def sweep_grouped(shape_keys, path):
-    """Time the trainer's grouped-GEMM realization per stage: per-expert mm
-    loop for fwd/dgrad, fused wgrad_accum loop for wgrad (train/moe.py
-    _GroupedLinearFn; torch.bmm's backward faults on this stack)."""
-    from simumax_amd.kernels.ops import ext
+    """Time the trainer's grouped-GEMM realization per stage via the SAME
+    dispatch the trainer runs (train/moe.py grouped_*_op: batched MFMA
+    kernels when E >= GROUPED_KERNEL_MIN_E, per-expert hipBLASLt loops
+    otherwise)."""
+    from simumax_amd.train.moe import (grouped_dgrad_op, grouped_fwd_op,
+                                       grouped_wgrad_op)
 
-    E_ = ext()
     tab = _load(path)
     for desc in shape_keys:
         if desc in tab and not OVERWRITE:
@@ -235,19 +236,16 @@ def sweep_grouped(shape_keys, path):
         ng, m, n, k = parse_group_key(desc)
         try:
             x = torch.randn(ng, m, k, device="cuda", dtype=torch.bfloat16)
-            w = torch.randn(ng, k, n, device="cuda", dtype=torch.bfloat16)
-            out = torch.empty(ng, m, n, device="cuda", dtype=torch.bfloat16)
-            if "stage=fwd" in desc or "stage=bwd_grad_act" in desc:
-                def fn():
-                    for e in range(ng):
-                        torch.mm(x[e], w[e], out=out[e])
-            else:  # wgrad: fused fp32 accumulate per expert
+            w = torch.randn(ng, n, k, device="cuda", dtype=torch.bfloat16)
+            if "stage=fwd" in desc:
+                fn = lambda: grouped_fwd_op(x, w)
+            elif "stage=bwd_grad_act" in desc:
                 d = torch.randn(ng, m, n, device="cuda", dtype=torch.bfloat16)
-                mg = torch.zeros(ng, k, n, device="cuda", dtype=torch.float32)
-
-                def fn():
-                    for e in range(ng):
-                        E_.wgrad_accum(x[e], d[e], mg[e])
+                fn = lambda: grouped_dgrad_op(d, w)
+            else:  # wgrad: fused fp32 accumulate
+                d = torch.randn(ng, m, n, device="cuda", dtype=torch.bfloat16)
+                mg = torch.zeros(ng, n, k, device="cuda", dtype=torch.float32)
+                fn = lambda: grouped_wgrad_op(d, x, mg)
             t_ms = _timeit(fn, iters=8)
         except torch.cuda.OutOfMemoryError:
             torch.cuda.empty_cache()

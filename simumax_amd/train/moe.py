@@ -39,27 +39,72 @@ def _grouped_key(ng, m, k, n, stage):
     return s
 
 
+# weights live in the natural Linear layout [E, out(N), in(K)].
+# Dispatch: the batched MFMA kernels (csrc/grouped_gemm.hip) win for
+# many small experts (dgrad 2.6x, wgrad 1.2x vs the loop at E=160);
+# per-expert hipBLASLt mm wins for few fat experts and for fwd
+# (scripts/grouped_kernel_test.py has the shape study). torch.bmm's
+# strided-batched BACKWARD and torch._grouped_mm both memory-fault on
+# this stack (scripts/grouped_mm_probe.py), hence no bmm anywhere.
+GROUPED_KERNEL_MIN_E = 32
+
+
+def grouped_fwd_op(x, w):
+    """x [E,M,K] @ w[E,N,K]^T -> [E,M,N]"""
+    E = x.shape[0]
+    out = torch.empty(E, x.shape[1], w.shape[1], dtype=x.dtype,
+                      device=x.device)
+    for e in range(E):
+        torch.mm(x[e], w[e].t(), out=out[e])
+    return out
+
+
+def grouped_dgrad_op(dout, w):
+    """dout [E,M,N] @ w[E,N,K] -> dx [E,M,K]"""
+    from ..kernels.ops import ext
+
+    E, N, K = w.shape[0], w.shape[1], w.shape[2]
+    if (dout.is_cuda and E >= GROUPED_KERNEL_MIN_E
+            and K % 128 == 0 and N % 32 == 0):
+        return ext().grouped_dgrad(dout, w)
+    dx = torch.empty(E, dout.shape[1], K, dtype=dout.dtype,
+                     device=dout.device)
+    for e in range(E):
+        torch.mm(dout[e], w[e], out=dx[e])
+    return dx
+
+
+def grouped_wgrad_op(dout, x, g):
+    """g fp32 [E,N,K] += dout[E,M,N]^T @ x[E,M,K]"""
+    from ..kernels.ops import ext
+
+    E, N, K = g.shape[0], g.shape[1], g.shape[2]
+    if (dout.is_cuda and E >= GROUPED_KERNEL_MIN_E
+            and N % 128 == 0 and K % 128 == 0):
+        ext().grouped_wgrad(dout, x, g)
+        return
+    if dout.is_cuda:
+        for e in range(E):
+            ext().wgrad_accum(dout[e], x[e], g[e])
+    else:
+        for e in range(E):
+            g[e] += dout[e].t().float() @ x[e].float()
+
+
 class _GroupedLinearFn(torch.autograd.Function):
-    """Per-expert GEMM loop: out[e] = x[e] @ w[e] with fused fp32 wgrad
-    accumulation. (torch.bmm's BACKWARD memory-faults in hipBLASLt
-    strided-batched mode at these shapes on this stack, so grouped GEMMs
-    run as E plain GEMMs — the same op the group_matmul calibration
-    sweeps time.)"""
+    """Grouped expert GEMM with fused fp32 wgrad accumulation; see the
+    dispatch notes above."""
 
     @staticmethod
     def forward(ctx, x, w):
         from ..kernels import insitu
 
         ctx.save_for_backward(x, w)
-        E = x.shape[0]
-        out = torch.empty(E, x.shape[1], w.shape[2], dtype=x.dtype,
-                          device=x.device)
         timing = insitu.ENABLED and x.is_cuda
         if timing:
             stop = insitu.start("group_matmul", _grouped_key(
-                E, x.shape[1], w.shape[1], w.shape[2], "fwd"))
-        for e in range(E):
-            torch.mm(x[e], w[e], out=out[e])
+                x.shape[0], x.shape[1], w.shape[2], w.shape[1], "fwd"))
+        out = grouped_fwd_op(x, w)
         if timing:
             stop()
         return out
@@ -67,36 +112,30 @@ class _GroupedLinearFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         from ..kernels import insitu
-        from ..kernels.ops import ext
 
         x, w = ctx.saved_tensors
         E = x.shape[0]
         dout = dout.contiguous()
-        dx = torch.empty_like(x)
         fused = x.is_cuda and hasattr(w, "main_grad")
-        dw = None if fused else torch.empty_like(w)
         timing = insitu.ENABLED and x.is_cuda
         if timing:
             stop_dx = insitu.start("group_matmul", _grouped_key(
-                E, x.shape[1], w.shape[1], w.shape[2], "bwd_grad_act"))
-        for e in range(E):
-            torch.mm(dout[e], w[e].t(), out=dx[e])
+                E, x.shape[1], w.shape[2], w.shape[1], "bwd_grad_act"))
+        dx = grouped_dgrad_op(dout, w)
         if timing:
             stop_dx()
             stop_dw = insitu.start("group_matmul", _grouped_key(
-                E, x.shape[1], w.shape[1], w.shape[2], "bwd_grad_w"))
+                E, x.shape[1], w.shape[2], w.shape[1], "bwd_grad_w"))
         if fused:
-            for e in range(E):
-                # main_grad[e] (K,N) += x[e]^T (K,M) @ dout[e] (M,N)
-                ext().wgrad_accum(x[e], dout[e], w.main_grad[e])
-        else:
-            for e in range(E):
-                torch.mm(x[e].t(), dout[e], out=dw[e])
-        if timing:
-            stop_dw()
-        if fused:
+            grouped_wgrad_op(dout, x, w.main_grad)
             # fresh unreferenced buffer -> AccumulateGrad steals it (no clone)
             dw = torch.empty_like(w)
+        else:
+            dw = torch.empty_like(w)
+            for e in range(E):
+                torch.mm(dout[e].t(), x[e], out=dw[e])
+        if timing:
+            stop_dw()
         return dx, dw
 
 
@@ -114,12 +153,10 @@ class MoEMLP(nn.Module):
         self.capacity = getattr(cfg, "capacity", 1) or 1
         self.router = nn.Linear(h, self.E, bias=False, dtype=dtype,
                                 device=device)
-        # grouped weights stored PRE-TRANSPOSED for bmm ([E, K, N]) — a
-        # runtime .transpose(1,2) view fed to hipBLASLt strided-batched GEMM
-        # memory-faults on this stack
-        self.w1 = nn.Parameter(torch.empty(self.E, h, 2 * self.I, dtype=dtype,
+        # grouped weights in the natural Linear layout [E, out, in]
+        self.w1 = nn.Parameter(torch.empty(self.E, 2 * self.I, h, dtype=dtype,
                                            device=device))
-        self.w2 = nn.Parameter(torch.empty(self.E, self.I, h, dtype=dtype,
+        self.w2 = nn.Parameter(torch.empty(self.E, h, self.I, dtype=dtype,
                                            device=device))
         nn.init.normal_(self.w1, std=0.02)
         nn.init.normal_(self.w2, std=0.02)
