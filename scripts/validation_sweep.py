@@ -101,9 +101,9 @@ def run_case(name, model, seq, mbs, mbc, steps, warmup=1):
 
 def main():
     os.makedirs("gpurun_out", exist_ok=True)
-    only = sys.argv[1] if len(sys.argv) > 1 else None
+    only = sys.argv[1:] or None
     for case in CASES:
-        if only and only not in case[0]:
+        if only and not any(o in case[0] for o in only):
             continue
         try:
             run_case(*case)
