@@ -273,3 +273,15 @@ def test_fast_estimator_matches_exact():
     # pp sweep shares ONE profile
     assert fe.cache.misses == 1
     assert fe.cache.hits == 2
+
+
+def test_dualpp_trace_export(tmp_path):
+    import json
+
+    from simumax_amd.perf.dualpp import export_dualpp_trace
+
+    p = str(tmp_path / "dual.json")
+    export_dualpp_trace(4, 8, 10.0, 20.0, p)
+    with open(p) as f:
+        d = json.load(f)
+    assert len(d["traceEvents"]) > 4 * 8
