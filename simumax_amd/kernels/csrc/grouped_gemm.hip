@@ -63,8 +63,8 @@ DEV bf16x8v gg_frag(const bf16raw *p) {
 __global__ __launch_bounds__(GG_BLOCK, 2)
 void gg_fwd_kernel(const bf16raw *__restrict__ A, const bf16raw *__restrict__ W,
                    bf16raw *__restrict__ C, int E, int M, int N, int K) {
-    __shared__ __attribute__((aligned(16))) bf16raw a_lds[2][GG_TM * GG_PAD];
-    __shared__ __attribute__((aligned(16))) bf16raw w_lds[2][GG_TN * GG_PAD];
+    __shared__ __attribute__((aligned(16))) bf16raw a_lds[GG_TM * GG_PAD];
+    __shared__ __attribute__((aligned(16))) bf16raw w_lds[GG_TN * GG_PAD];
     const int tiles_m = (M + GG_TM - 1) / GG_TM;
     const int m_tile = blockIdx.x % tiles_m;
     const int n_tile = blockIdx.x / tiles_m;
@@ -88,51 +88,31 @@ void gg_fwd_kernel(const bf16raw *__restrict__ A, const bf16raw *__restrict__ W,
 #pragma unroll
         for (int cs = 0; cs < 8; ++cs) acc[rs][cs] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    // T14 staging: global loads for step t+1 issue before step t's MFMAs
-    bf16x8 st_a[2], st_w[2];
-    auto stage_load = [&](int k0) {
-        if (k0 >= K) return;
-#pragma unroll
-        for (int h = 0; h < 2; ++h) {
-            const int r = srow + h * 64;
-            if (r < mrem) st_a[h] = load8(Ae + (long)r * K + k0 + scol);
-            else st_a[h].raw = uint4{0, 0, 0, 0};
-            st_w[h] = load8(We + (long)r * K + k0 + scol);
-        }
-    };
-    auto stage_write = [&](int k0, int buf) {
-        if (k0 >= K) return;
-#pragma unroll
-        for (int h = 0; h < 2; ++h) {
-            const int r = srow + h * 64;
-            store8(a_lds[buf] + r * GG_PAD + scol, st_a[h]);
-            store8(w_lds[buf] + r * GG_PAD + scol, st_w[h]);
-        }
-    };
-    stage_load(0);
-    stage_write(0, 0);
-    stage_load(GG_RED);
-    __syncthreads();
-
     for (int k0 = 0; k0 < K; k0 += GG_RED) {
-        const int buf = (k0 / GG_RED) & 1;
-        bf16x8v a0 = gg_frag(a_lds[buf] + (wave * 32 + (lane & 15)) * GG_PAD
+        __syncthreads();
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+            const int r = srow + h * 64;
+            bf16x8 av;
+            if (r < mrem) av = load8(Ae + (long)r * K + k0 + scol);
+            else av.raw = uint4{0, 0, 0, 0};
+            store8(a_lds + r * GG_PAD + scol, av);
+            store8(w_lds + r * GG_PAD + scol, load8(We + (long)r * K + k0 + scol));
+        }
+        __syncthreads();
+        bf16x8v a0 = gg_frag(a_lds + (wave * 32 + (lane & 15)) * GG_PAD
                              + (lane >> 4) * 8);
-        bf16x8v a1 = gg_frag(a_lds[buf] + (wave * 32 + 16 + (lane & 15)) * GG_PAD
+        bf16x8v a1 = gg_frag(a_lds + (wave * 32 + 16 + (lane & 15)) * GG_PAD
                              + (lane >> 4) * 8);
 #pragma unroll
         for (int cs = 0; cs < 8; ++cs) {
-            bf16x8v b = gg_frag(w_lds[buf] + (cs * 16 + (lane & 15)) * GG_PAD
+            bf16x8v b = gg_frag(w_lds + (cs * 16 + (lane & 15)) * GG_PAD
                                 + (lane >> 4) * 8);
             acc[0][cs] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 a0, b, acc[0][cs], 0, 0, 0);
             acc[1][cs] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 a1, b, acc[1][cs], 0, 0, 0);
         }
-        __syncthreads();
-        stage_write(k0 + GG_RED, buf ^ 1);
-        stage_load(k0 + 2 * GG_RED);
-        __syncthreads();
     }
 #pragma unroll
     for (int rs = 0; rs < 2; ++rs)
