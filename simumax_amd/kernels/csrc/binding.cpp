@@ -210,7 +210,7 @@ torch::Tensor grouped_fwd(torch::Tensor x, torch::Tensor w) {
     TORCH_CHECK(x.dim() == 3 && w.dim() == 3, "x [E,M,K], w [E,N,K]");
     const int E = x.size(0), M = x.size(1), K = x.size(2), N = w.size(1);
     TORCH_CHECK(w.size(0) == E && (int)w.size(2) == K);
-    TORCH_CHECK(N % 128 == 0 && K % 32 == 0, "grouped fwd needs N%128, K%32");
+    TORCH_CHECK(N % 128 == 0 && K % 128 == 0, "grouped fwd needs N%128, K%128");
     auto c = torch::empty({E, M, N}, x.options());
     gg_fwd(x.data_ptr(), w.data_ptr(), c.data_ptr(), E, M, N, K,
            cur_stream());
