@@ -20,8 +20,13 @@ GOLDEN_DIR = os.path.join(os.path.dirname(__file__), "goldens")
 CASES = [
     ("llama3-8b", "tp1_pp2_dp4_mbs1"),
     ("llama3-8b", "tp8_pp1_dp1_mbs1"),
+    ("llama3-8b", "tp1_pp4_vp2_sync_mbs1_mbc8"),
     ("llama3-70b-l12", "tp2_pp2_dp2_mbs1_selective"),
+    ("llama3-405b", "tp8_pp1_dp1_mbs1"),
+    ("qwen3-32b-l12", "tp2_pp1_dp4_mbs1"),
+    ("mixtral-8x7b-l8", "ep8_pp1_dp8_mbs1"),
     ("deepseekv2-l4", "ep8_pp1_dp8_mbs1"),
+    ("deepseekv2-l4", "ep4_pp2_dp4_mbs1"),
 ]
 
 
