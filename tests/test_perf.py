@@ -285,3 +285,30 @@ def test_dualpp_trace_export(tmp_path):
     with open(p) as f:
         d = json.load(f)
     assert len(d["traceEvents"]) > 4 * 8
+
+
+def test_strategy_search_moe_with_ep():
+    """Grid search over a MoE model including EP degrees (fast-prefilter
+    + exact top-k) returns feasible ranked rows."""
+    from simumax_amd.core.config import (ModelConfig, StrategyConfig,
+                                         SystemConfig)
+    from simumax_amd import (get_simu_model_config, get_simu_strategy_config,
+                             get_simu_system_config)
+    from simumax_amd.tuning.strategy_searcher import (SearchSpace,
+                                                      StrategySearcher)
+
+    mc = ModelConfig.init_from_config_file(
+        get_simu_model_config("mixtral-8x7b-l8"))
+    base = StrategyConfig.init_from_config_file(
+        get_simu_strategy_config("ep8_pp1_dp8_mbs1"))
+    searcher = StrategySearcher(mc, SystemConfig.init_from_config_file(
+        get_simu_system_config("mi355x")), base)
+    res = searcher.search(world_size=8, global_batch_size=32,
+                          space=SearchSpace(tp=(1, 2), pp=(1, 2),
+                                            ep=(1, 2, 4, 8),
+                                            recompute=(None,), max_mbs=2),
+                          fast_prefilter=True, exact_top_k=4)
+    assert len(res.rows) >= 2
+    best = res.best
+    assert best["mfu"] > 0.1
+    assert best["peak_mem"] < 288 * 1024**3
