@@ -97,7 +97,7 @@ class MLAAttention(nn.Module):
 
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
-                 layer_idx=0):
+                 layer_idx=0, ep_group=None, ep_size=1):
         super().__init__()
         h = cfg.hidden_size
         self.input_norm = K.RMSNorm(h, dtype=dtype, device=device)
@@ -112,7 +112,8 @@ class LlamaDecoderLayer(nn.Module):
         if self.use_moe:
             from .moe import MoEMLP
 
-            self.moe_mlp = MoEMLP(cfg, dtype=dtype, device=device)
+            self.moe_mlp = MoEMLP(cfg, dtype=dtype, device=device,
+                                  ep_group=ep_group, ep_size=ep_size)
         else:
             self.fc1 = K.FusedLinear(h, 2 * cfg.intermediate_size, dtype=dtype, device=device)
             self.fc2 = K.FusedLinear(cfg.intermediate_size, h, dtype=dtype, device=device)
@@ -133,14 +134,15 @@ class LlamaDecoderLayer(nn.Module):
 
 class LlamaForTraining(nn.Module):
     def __init__(self, cfg: ModelConfig, seq_len: int, dtype=torch.bfloat16,
-                 rope_base=500000.0, device=None):
+                 rope_base=500000.0, device=None, ep_group=None, ep_size=1):
         super().__init__()
         self.cfg = cfg
         self.seq_len = seq_len
         self.embedding = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
                                       dtype=dtype, device=device)
         self.layers = nn.ModuleList(
-            [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i)
+            [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i,
+                               ep_group=ep_group, ep_size=ep_size)
              for i in range(cfg.layer_num)])
         self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
         self.lm_head = K.FusedLinear(cfg.hidden_size, cfg.vocab_size,

@@ -9,8 +9,14 @@ capacity dispatch), matching the simulator's MoE cost/memory accounting:
   group_matmul calibration sweep times (calib/sweeps.sweep_grouped)
 * weighted combine scatter (UnPermutation)
 
-Single-rank experts (EP1); the EP all-to-all path is simulator-side this
-round (multi-GPU EP training is a round-2 item).
+Expert parallelism (ep_size > 1): experts are sharded over the EP
+process group; capacity padding makes every rank's per-expert
+contribution a FIXED size, so dispatch/combine are single equal-split
+`all_to_all_single` calls (autograd via _AllToAll below — a2a is its own
+adjoint for symmetric equal splits). Expert weights carry `_is_expert`
+and their grads reduce over the edp group (DataParallelGradReducer).
+Mirrors the simulator's Permutation/UnPermutation comm model
+(ops/moe.py).
 """
 
 from __future__ import annotations
@@ -18,7 +24,30 @@ from __future__ import annotations
 import math
 
 import torch
+import torch.distributed as dist
 import torch.nn as nn
+
+
+class _AllToAll(torch.autograd.Function):
+    """Equal-split all_to_all_single with autograd (backward = a2a of the
+    incoming grads — exact adjoint for symmetric splits)."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        out = torch.empty_like(x)
+        dist.all_to_all_single(out, x.contiguous(), group=group)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        g = torch.empty_like(grad)
+        dist.all_to_all_single(g, grad.contiguous(), group=ctx.group)
+        return g, None
+
+
+def all_to_all(x, group):
+    return _AllToAll.apply(x, group)
 
 
 def _grouped_key(ng, m, k, n, stage):
@@ -144,24 +173,31 @@ def grouped_linear(x, w):
 
 
 class MoEMLP(nn.Module):
-    def __init__(self, cfg, dtype=torch.bfloat16, device=None):
+    def __init__(self, cfg, dtype=torch.bfloat16, device=None, ep_group=None,
+                 ep_size=1):
         super().__init__()
         h = cfg.hidden_size
         self.E = cfg.expert_num
         self.topk = cfg.topk
         self.I = cfg.moe_ffn_hidden_size
         self.capacity = getattr(cfg, "capacity", 1) or 1
+        self.ep_group = ep_group
+        self.ep = ep_size
+        assert self.E % self.ep == 0
+        self.le = self.E // self.ep          # local experts
         self.router = nn.Linear(h, self.E, bias=False, dtype=dtype,
                                 device=device)
-        # grouped weights in the natural Linear layout [E, out, in]
-        self.w1 = nn.Parameter(torch.empty(self.E, 2 * self.I, h, dtype=dtype,
+        # LOCAL expert weights in the natural Linear layout [le, out, in]
+        self.w1 = nn.Parameter(torch.empty(self.le, 2 * self.I, h, dtype=dtype,
                                            device=device))
-        self.w2 = nn.Parameter(torch.empty(self.E, h, self.I, dtype=dtype,
+        self.w2 = nn.Parameter(torch.empty(self.le, h, self.I, dtype=dtype,
                                            device=device))
         nn.init.normal_(self.w1, std=0.02)
         nn.init.normal_(self.w2, std=0.02)
         self.w1._fused_wgrad = True
         self.w2._fused_wgrad = True
+        self.w1._is_expert = True
+        self.w2._is_expert = True
         shared_i = getattr(cfg, "moe_shared_expert_intermediate_size", 0)
         self.shared = None
         if shared_i:
@@ -202,18 +238,34 @@ class MoEMLP(nn.Module):
         slot_index = dst_exp * cap + dst_slot                # [M]
         w_kept = weight.reshape(-1)[order][keep].to(x.dtype)
 
-        # dispatch: [E*cap, H] padded buffer (Permutation)
+        # dispatch: [E*cap, H] padded buffer, ordered by GLOBAL expert
+        # (Permutation permute1)
         xp = torch.zeros(self.E * cap, H, dtype=x.dtype, device=x.device)
         xp.index_copy_(0, slot_index, xf.index_select(0, src_tok))
-        xp = xp.view(self.E, cap, H)
 
-        # grouped GEMMs (per-expert mm loop, the calibrated path) + swiglu
-        h1 = grouped_linear(xp, self.w1)                     # [E, cap, 2I]
-        a = K.swiglu(h1.reshape(-1, 2 * self.I)).reshape(self.E, cap, self.I)
-        y = grouped_linear(a, self.w2)                       # [E, cap, H]
+        if self.ep > 1:
+            # EP dispatch a2a: the block for dest rank d = experts
+            # [d*le, (d+1)*le) is contiguous; equal splits by capacity.
+            recv = all_to_all(xp, self.ep_group)             # [ep*le*cap, H]
+            # [src][le][cap] -> expert-major [le, ep*cap, H] (permute2)
+            xg = (recv.view(self.ep, self.le, cap, H)
+                  .transpose(0, 1).reshape(self.le, self.ep * cap, H)
+                  .contiguous())
+        else:
+            xg = xp.view(self.E, cap, H)
 
-        # combine: weighted scatter back (UnPermutation)
-        y_flat = y.reshape(self.E * cap, H)
+        # grouped GEMMs (dispatch in grouped_*_op) + swiglu
+        h1 = grouped_linear(xg, self.w1)                     # [le, M, 2I]
+        a = K.swiglu(h1.reshape(-1, 2 * self.I)).reshape(h1.shape[0], -1, self.I)
+        y = grouped_linear(a, self.w2)                       # [le, M, H]
+
+        if self.ep > 1:
+            # reverse permute2 + combine a2a back to source ranks
+            yb = (y.view(self.le, self.ep, cap, H).transpose(0, 1)
+                  .reshape(self.ep * self.le * cap, H).contiguous())
+            y_flat = all_to_all(yb, self.ep_group)           # [E*cap, H]
+        else:
+            y_flat = y.reshape(self.E * cap, H)
         out = torch.zeros_like(xf)
         out.index_add_(0, src_tok,
                        y_flat.index_select(0, slot_index) * w_kept[:, None])

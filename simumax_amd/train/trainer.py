@@ -34,6 +34,37 @@ class TrainConfig:
     grad_clip: float = 1.0
     overlap_grad_reduce: bool = True
     bucket_bytes: int = 100 * 1024**2
+    ep_size: int = 1
+
+
+_EP_GROUPS = {}
+
+
+def get_ep_groups(ep_size):
+    """EP group = ep_size consecutive ranks; edp group = same offset
+    strided by ep (order matches core/utils.get_rank_group: ep fastest).
+    dist.new_group must be called by every rank in the same order, so
+    groups are built for all slices and cached."""
+    if ep_size <= 1 or not dist.is_initialized():
+        return None, None
+    key = (ep_size, dist.get_world_size())
+    if key not in _EP_GROUPS:
+        world = dist.get_world_size()
+        assert world % ep_size == 0
+        ep_groups, edp_groups = {}, {}
+        for start in range(0, world, ep_size):
+            g = dist.new_group(list(range(start, start + ep_size)))
+            for r in range(start, start + ep_size):
+                ep_groups[r] = g
+        for off in range(ep_size):
+            ranks = list(range(off, world, ep_size))
+            g = dist.new_group(ranks)
+            for r in ranks:
+                edp_groups[r] = g
+        _EP_GROUPS[key] = (ep_groups, edp_groups)
+    ep_groups, edp_groups = _EP_GROUPS[key]
+    r = dist.get_rank()
+    return ep_groups[r], edp_groups[r]
 
 
 class MixedPrecisionAdam:
@@ -46,7 +77,13 @@ class MixedPrecisionAdam:
     matching the calibrated optimizer bandwidth."""
 
     def __init__(self, params, cfg: TrainConfig):
-        self.params = [p for p in params if p.requires_grad]
+        params = [p for p in params if p.requires_grad]
+        # dense params first, expert params last: the flat grad buffer then
+        # splits into one contiguous slice per reduction group
+        self.params = ([p for p in params if not getattr(p, "_is_expert", False)]
+                       + [p for p in params if getattr(p, "_is_expert", False)])
+        self.dense_numel = sum(p.numel() for p in self.params
+                               if not getattr(p, "_is_expert", False))
         self.cfg = cfg
         total = sum(p.numel() for p in self.params)
         dev = self.params[0].device
@@ -105,17 +142,24 @@ class DataParallelGradReducer:
     optimizer's FLAT grad buffer, overlapped with backward (Megatron
     DistributedDataParallel grad-buffer semantics)."""
 
-    def __init__(self, opt, overlap: bool, bucket_bytes: int):
+    def __init__(self, opt, overlap: bool, bucket_bytes: int,
+                 edp_group=None, edp_size=1):
         self.params = opt.params
         self.flat_grad = opt.flat_grad
-        self.overlap = overlap and dist.is_initialized() and dist.get_world_size() > 1
-        self.enabled = dist.is_initialized() and dist.get_world_size() > 1
+        self.dense_numel = opt.dense_numel
+        # dense grads reduce over the world (= dp when tp=pp=1); expert
+        # grads are replicated only across edp and reduce over that group
+        self.edp_group = edp_group
+        self.edp_size = edp_size
+        world = dist.get_world_size() if dist.is_initialized() else 1
+        self.overlap = overlap and dist.is_initialized() and world > 1
+        self.enabled = dist.is_initialized() and world > 1
         self.bucket_bytes = bucket_bytes
         self.handles = []
         # Megatron no_sync semantics: only the LAST microbatch's backward
         # triggers the bucketed all_reduce
         self.reduce_this_pass = True
-        # param offsets in the flat buffer (registration order)
+        # param offsets in the flat buffer (optimizer order: dense first)
         offs, off = {}, 0
         for p in self.params:
             offs[id(p)] = (off, off + p.numel())
@@ -125,24 +169,31 @@ class DataParallelGradReducer:
         # the hook the weight-sized placeholder wgrads pile up in p.grad
         # until backward ends (~25 GiB on llama3-70b-l12).
         # reverse order (grads become ready back-to-front); reversed
-        # consecutive params are a contiguous flat slice
-        buckets, cur, cur_bytes = [], [], 0
-        for p in reversed(self.params):
-            cur.append(p)
-            cur_bytes += p.numel() * 4
-            if cur_bytes >= bucket_bytes:
+        # consecutive params are a contiguous flat slice. Buckets are built
+        # PER PARTITION so none straddles the dense/expert boundary.
+        buckets = []
+        for part in (
+            [p for p in self.params if getattr(p, "_is_expert", False)],
+            [p for p in self.params if not getattr(p, "_is_expert", False)],
+        ):
+            cur, cur_bytes = [], 0
+            for p in reversed(part):
+                cur.append(p)
+                cur_bytes += p.numel() * 4
+                if cur_bytes >= bucket_bytes:
+                    buckets.append(cur)
+                    cur, cur_bytes = [], 0
+            if cur:
                 buckets.append(cur)
-                cur, cur_bytes = [], 0
-        if cur:
-            buckets.append(cur)
         self._hook_handles = []
         for bucket in buckets:
             lo = min(offs[id(p)][0] for p in bucket)
             hi = max(offs[id(p)][1] for p in bucket)
+            is_exp = getattr(bucket[0], "_is_expert", False)
             remaining = {id(p) for p in bucket}
             for p in bucket:
                 self._hook_handles.append(p.register_post_accumulate_grad_hook(
-                    self._make_hook((lo, hi), remaining)))
+                    self._make_hook((lo, hi), remaining, is_exp)))
         self._buckets = buckets
 
     def remove_hooks(self):
@@ -154,7 +205,7 @@ class DataParallelGradReducer:
             h.remove()
         self._hook_handles.clear()
 
-    def _make_hook(self, span, remaining):
+    def _make_hook(self, span, remaining, is_expert=False):
         def hook(p):
             if not getattr(p, "_fused_wgrad", False) or not p.is_cuda:
                 p.main_grad.add_(p.grad.float())
@@ -163,8 +214,14 @@ class DataParallelGradReducer:
             if not remaining:
                 if self.overlap and self.reduce_this_pass:
                     sl = self.flat_grad[span[0]:span[1]]
-                    sl.div_(dist.get_world_size())
-                    self.handles.append(dist.all_reduce(sl, async_op=True))
+                    if is_expert:
+                        if self.edp_size > 1:
+                            sl.div_(self.edp_size)
+                            self.handles.append(dist.all_reduce(
+                                sl, group=self.edp_group, async_op=True))
+                    else:
+                        sl.div_(dist.get_world_size())
+                        self.handles.append(dist.all_reduce(sl, async_op=True))
                 # rearm for the next backward pass
                 remaining.update(self._bucket_ids(span))
         return hook
@@ -185,8 +242,13 @@ class DataParallelGradReducer:
                 h.wait()
             self.handles.clear()
         elif self.enabled:
-            self.flat_grad.div_(dist.get_world_size())
-            dist.all_reduce(self.flat_grad)
+            dense = self.flat_grad[:self.dense_numel]
+            dense.div_(dist.get_world_size())
+            dist.all_reduce(dense)
+            if self.dense_numel < self.flat_grad.numel() and self.edp_size > 1:
+                exp = self.flat_grad[self.dense_numel:]
+                exp.div_(self.edp_size)
+                dist.all_reduce(exp, group=self.edp_group)
 
 
 def accumulate_main_grads(params):
@@ -203,10 +265,16 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
     # Megatron-style vocab padding (keeps CE vocab a GPU-friendly multiple
     # and makes GEMM shape keys match the calibration tables)
     model_cfg.maybe_pad_vocab_size(tp_size)
-    model = LlamaForTraining(model_cfg, cfg.seq_len, device=device)
+    ep_group, edp_group = get_ep_groups(cfg.ep_size)
+    edp_size = 1
+    if cfg.ep_size > 1 and dist.is_initialized():
+        edp_size = dist.get_world_size() // cfg.ep_size
+    model = LlamaForTraining(model_cfg, cfg.seq_len, device=device,
+                             ep_group=ep_group, ep_size=cfg.ep_size)
     opt = MixedPrecisionAdam(model.parameters(), cfg)
     reducer = DataParallelGradReducer(opt, cfg.overlap_grad_reduce,
-                                      cfg.bucket_bytes)
+                                      cfg.bucket_bytes,
+                                      edp_group=edp_group, edp_size=edp_size)
     return model, opt, reducer
 
 
