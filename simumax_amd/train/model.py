@@ -20,22 +20,39 @@ from ..kernels import ops as K
 
 
 class GQAAttention(nn.Module):
-    """Fused-qkv GQA attention (Llama/Qwen/Mixtral family)."""
+    """Fused-qkv GQA attention (Llama/Qwen/Mixtral family). With tp > 1
+    the heads are split over the TP group: column-parallel qkv (local
+    head sections), row-parallel out projection (Megatron layout)."""
 
-    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
+                 tp_group=None, tp_size=1):
         super().__init__()
         h = cfg.hidden_size
-        self.heads = cfg.head_num
-        self.kv_heads = cfg.kv_head_num
+        assert cfg.head_num % tp_size == 0 and cfg.kv_head_num % tp_size == 0
+        self.heads = cfg.head_num // tp_size
+        self.kv_heads = cfg.kv_head_num // tp_size
         self.head_size = cfg.head_size
-        qkv_out = (cfg.head_num + 2 * cfg.kv_head_num) * cfg.head_size
-        self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype, device=device)
-        self.out_proj = K.FusedLinear(cfg.head_num * cfg.head_size, h,
-                                      dtype=dtype, device=device)
+        self.tp_group = tp_group
+        qkv_out = (self.heads + 2 * self.kv_heads) * cfg.head_size
+        if tp_size > 1:
+            from .tp import copy_to_tp, reduce_from_tp  # noqa: F401
+
+            self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype,
+                                          device=device)
+            self.qkv_proj.weight._is_tp_shard = True
+            self.out_proj = K.FusedLinear(self.heads * cfg.head_size, h,
+                                          dtype=dtype, device=device)
+            self.out_proj.weight._is_tp_shard = True
+        else:
+            self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype, device=device)
+            self.out_proj = K.FusedLinear(self.heads * cfg.head_size, h,
+                                          dtype=dtype, device=device)
 
     def forward(self, y, rope_cs, pos):
+        from .tp import copy_to_tp, reduce_from_tp
+
         B, S, _ = y.shape
-        qkv = self.qkv_proj(y)
+        qkv = self.qkv_proj(copy_to_tp(y, self.tp_group))
         d = self.head_size
         q, k, v = qkv.split(
             [self.heads * d, self.kv_heads * d, self.kv_heads * d], dim=-1)
@@ -45,7 +62,8 @@ class GQAAttention(nn.Module):
         k = k.view(B, S, self.kv_heads, d)
         v = v.reshape(B, S, self.kv_heads, d)
         ctx = K.flash_attention(q, k, v, causal=True)
-        return self.out_proj(ctx.reshape(B, S, self.heads * d))
+        return reduce_from_tp(
+            self.out_proj(ctx.reshape(B, S, self.heads * d)), self.tp_group)
 
 
 class MLAAttention(nn.Module):
@@ -97,14 +115,18 @@ class MLAAttention(nn.Module):
 
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
-                 layer_idx=0, ep_group=None, ep_size=1):
+                 layer_idx=0, ep_group=None, ep_size=1, tp_group=None,
+                 tp_size=1):
         super().__init__()
         h = cfg.hidden_size
+        self.tp_group = tp_group
         self.input_norm = K.RMSNorm(h, dtype=dtype, device=device)
         if getattr(cfg, "attention_type", "gqa") == "mla":
+            assert tp_size == 1, "MLA requires tp_size == 1 (simulator parity)"
             self.attention = MLAAttention(cfg, dtype=dtype, device=device)
         else:
-            self.attention = GQAAttention(cfg, dtype=dtype, device=device)
+            self.attention = GQAAttention(cfg, dtype=dtype, device=device,
+                                          tp_group=tp_group, tp_size=tp_size)
         self.pre_mlp_norm = K.RMSNorm(h, dtype=dtype, device=device)
         assert cfg.use_swiglu
         self.use_moe = (cfg.model_type == "moe"
@@ -112,13 +134,23 @@ class LlamaDecoderLayer(nn.Module):
         if self.use_moe:
             from .moe import MoEMLP
 
+            assert tp_size == 1, "trainer supports tp XOR ep for now"
             self.moe_mlp = MoEMLP(cfg, dtype=dtype, device=device,
                                   ep_group=ep_group, ep_size=ep_size)
         else:
-            self.fc1 = K.FusedLinear(h, 2 * cfg.intermediate_size, dtype=dtype, device=device)
-            self.fc2 = K.FusedLinear(cfg.intermediate_size, h, dtype=dtype, device=device)
+            # Megatron MLP split: gate and up each sharded I/tp; fc2 row-
+            # parallel over I/tp with one fwd all_reduce
+            assert cfg.intermediate_size % tp_size == 0
+            i_local = cfg.intermediate_size // tp_size
+            self.fc1 = K.FusedLinear(h, 2 * i_local, dtype=dtype, device=device)
+            self.fc2 = K.FusedLinear(i_local, h, dtype=dtype, device=device)
+            if tp_size > 1:
+                self.fc1.weight._is_tp_shard = True
+                self.fc2.weight._is_tp_shard = True
 
     def forward(self, x, rope_cs, pos):
+        from .tp import copy_to_tp, reduce_from_tp
+
         # x: [B, S, H]
         res = x
         y = self.input_norm(x)
@@ -128,25 +160,35 @@ class LlamaDecoderLayer(nn.Module):
         if self.use_moe:
             y = self.moe_mlp(y)
         else:
-            y = self.fc2(K.swiglu(self.fc1(y)))
+            y = self.fc2(K.swiglu(self.fc1(copy_to_tp(y, self.tp_group))))
+            y = reduce_from_tp(y, self.tp_group)
         return res + y
 
 
 class LlamaForTraining(nn.Module):
     def __init__(self, cfg: ModelConfig, seq_len: int, dtype=torch.bfloat16,
-                 rope_base=500000.0, device=None, ep_group=None, ep_size=1):
+                 rope_base=500000.0, device=None, ep_group=None, ep_size=1,
+                 tp_group=None, tp_size=1, tp_rank=0):
         super().__init__()
         self.cfg = cfg
         self.seq_len = seq_len
+        self.tp_group = tp_group
+        self.tp_size = tp_size
+        assert cfg.vocab_size % tp_size == 0
+        self.vocab_local = cfg.vocab_size // tp_size
+        self.vocab_start = tp_rank * self.vocab_local
         self.embedding = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
                                       dtype=dtype, device=device)
         self.layers = nn.ModuleList(
             [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i,
-                               ep_group=ep_group, ep_size=ep_size)
+                               ep_group=ep_group, ep_size=ep_size,
+                               tp_group=tp_group, tp_size=tp_size)
              for i in range(cfg.layer_num)])
         self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
-        self.lm_head = K.FusedLinear(cfg.hidden_size, cfg.vocab_size,
+        self.lm_head = K.FusedLinear(cfg.hidden_size, self.vocab_local,
                                      dtype=dtype, device=device)
+        if tp_size > 1:
+            self.lm_head.weight._is_tp_shard = True
         rope_dim = (cfg.qk_pos_emb_head_dim
                     if getattr(cfg, "attention_type", "gqa") == "mla"
                     else cfg.head_size)
@@ -163,9 +205,17 @@ class LlamaForTraining(nn.Module):
         for layer in self.layers:
             x = layer(x, self.rope_cs, pos)
         x = self.final_norm(x)
-        logits = self.lm_head(x)
-        loss = K.fused_cross_entropy(
-            logits.reshape(B * S, -1), labels.reshape(-1))
+        if self.tp_size > 1:
+            from .tp import copy_to_tp, vocab_parallel_ce
+
+            logits = self.lm_head(copy_to_tp(x, self.tp_group))
+            loss = vocab_parallel_ce(logits.reshape(B * S, -1),
+                                     labels.reshape(-1), self.tp_group,
+                                     self.vocab_start)
+        else:
+            logits = self.lm_head(x)
+            loss = K.fused_cross_entropy(
+                logits.reshape(B * S, -1), labels.reshape(-1))
         return loss.mean()
 
     def num_params(self):

@@ -35,6 +35,7 @@ class TrainConfig:
     overlap_grad_reduce: bool = True
     bucket_bytes: int = 100 * 1024**2
     ep_size: int = 1
+    tp_size: int = 1
 
 
 _EP_GROUPS = {}
@@ -143,17 +144,20 @@ class DataParallelGradReducer:
     DistributedDataParallel grad-buffer semantics)."""
 
     def __init__(self, opt, overlap: bool, bucket_bytes: int,
-                 edp_group=None, edp_size=1):
+                 edp_group=None, edp_size=1, dp_group=None, dp_size=None):
         self.params = opt.params
         self.flat_grad = opt.flat_grad
         self.dense_numel = opt.dense_numel
-        # dense grads reduce over the world (= dp when tp=pp=1); expert
-        # grads are replicated only across edp and reduce over that group
+        # dense grads reduce over the dp group (= world when tp=ep=1);
+        # expert grads are replicated only across edp and reduce there
         self.edp_group = edp_group
         self.edp_size = edp_size
+        self.dp_group = dp_group
         world = dist.get_world_size() if dist.is_initialized() else 1
-        self.overlap = overlap and dist.is_initialized() and world > 1
-        self.enabled = dist.is_initialized() and world > 1
+        self.dp_size = dp_size if dp_size is not None else world
+        self.overlap = (overlap and dist.is_initialized()
+                        and self.dp_size > 1)
+        self.enabled = dist.is_initialized() and self.dp_size > 1
         self.bucket_bytes = bucket_bytes
         self.handles = []
         # Megatron no_sync semantics: only the LAST microbatch's backward
@@ -220,8 +224,9 @@ class DataParallelGradReducer:
                             self.handles.append(dist.all_reduce(
                                 sl, group=self.edp_group, async_op=True))
                     else:
-                        sl.div_(dist.get_world_size())
-                        self.handles.append(dist.all_reduce(sl, async_op=True))
+                        sl.div_(self.dp_size)
+                        self.handles.append(dist.all_reduce(
+                            sl, group=self.dp_group, async_op=True))
                 # rearm for the next backward pass
                 remaining.update(self._bucket_ids(span))
         return hook
@@ -243,8 +248,8 @@ class DataParallelGradReducer:
             self.handles.clear()
         elif self.enabled:
             dense = self.flat_grad[:self.dense_numel]
-            dense.div_(dist.get_world_size())
-            dist.all_reduce(dense)
+            dense.div_(self.dp_size)
+            dist.all_reduce(dense, group=self.dp_group)
             if self.dense_numel < self.flat_grad.numel() and self.edp_size > 1:
                 exp = self.flat_grad[self.dense_numel:]
                 exp.div_(self.edp_size)
@@ -262,6 +267,8 @@ def accumulate_main_grads(params):
 def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                   tp_size: int = 1):
     torch.manual_seed(1234)
+    assert not (cfg.ep_size > 1 and cfg.tp_size > 1), "trainer: tp XOR ep"
+    tp_size = max(tp_size, cfg.tp_size)
     # Megatron-style vocab padding (keeps CE vocab a GPU-friendly multiple
     # and makes GEMM shape keys match the calibration tables)
     model_cfg.maybe_pad_vocab_size(tp_size)
@@ -269,12 +276,21 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
     edp_size = 1
     if cfg.ep_size > 1 and dist.is_initialized():
         edp_size = dist.get_world_size() // cfg.ep_size
+    from .tp import get_tp_groups
+
+    tp_group, dp_group, tp_rank = get_tp_groups(tp_size)
+    dp_size = None
+    if tp_group is not None:
+        dp_size = dist.get_world_size() // tp_size
     model = LlamaForTraining(model_cfg, cfg.seq_len, device=device,
-                             ep_group=ep_group, ep_size=cfg.ep_size)
+                             ep_group=ep_group, ep_size=cfg.ep_size,
+                             tp_group=tp_group, tp_size=tp_size,
+                             tp_rank=tp_rank)
     opt = MixedPrecisionAdam(model.parameters(), cfg)
     reducer = DataParallelGradReducer(opt, cfg.overlap_grad_reduce,
                                       cfg.bucket_bytes,
-                                      edp_group=edp_group, edp_size=edp_size)
+                                      edp_group=edp_group, edp_size=edp_size,
+                                      dp_group=dp_group, dp_size=dp_size)
     return model, opt, reducer
 
 
