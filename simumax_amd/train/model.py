@@ -25,7 +25,7 @@ class GQAAttention(nn.Module):
     head sections), row-parallel out projection (Megatron layout)."""
 
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
-                 tp_group=None, tp_size=1):
+                 tp_group=None, tp_size=1, sp=False):
         super().__init__()
         h = cfg.hidden_size
         assert cfg.head_num % tp_size == 0 and cfg.kv_head_num % tp_size == 0
@@ -33,6 +33,7 @@ class GQAAttention(nn.Module):
         self.kv_heads = cfg.kv_head_num // tp_size
         self.head_size = cfg.head_size
         self.tp_group = tp_group
+        self.sp = sp
         qkv_out = (self.heads + 2 * self.kv_heads) * cfg.head_size
         if tp_size > 1:
             from .tp import copy_to_tp, reduce_from_tp  # noqa: F401
@@ -49,10 +50,14 @@ class GQAAttention(nn.Module):
                                           dtype=dtype, device=device)
 
     def forward(self, y, rope_cs, pos):
-        from .tp import copy_to_tp, reduce_from_tp
+        from .tp import copy_to_tp, gather_seq, reduce_from_tp, scatter_seq
 
+        if self.sp:
+            y = gather_seq(y, self.tp_group)   # [B, S/tp, H] -> [B, S, H]
+            qkv = self.qkv_proj(y)
+        else:
+            qkv = self.qkv_proj(copy_to_tp(y, self.tp_group))
         B, S, _ = y.shape
-        qkv = self.qkv_proj(copy_to_tp(y, self.tp_group))
         d = self.head_size
         q, k, v = qkv.split(
             [self.heads * d, self.kv_heads * d, self.kv_heads * d], dim=-1)
@@ -62,8 +67,10 @@ class GQAAttention(nn.Module):
         k = k.view(B, S, self.kv_heads, d)
         v = v.reshape(B, S, self.kv_heads, d)
         ctx = K.flash_attention(q, k, v, causal=True)
-        return reduce_from_tp(
-            self.out_proj(ctx.reshape(B, S, self.heads * d)), self.tp_group)
+        out = self.out_proj(ctx.reshape(B, S, self.heads * d))
+        if self.sp:
+            return scatter_seq(out, self.tp_group)
+        return reduce_from_tp(out, self.tp_group)
 
 
 class MLAAttention(nn.Module):
@@ -116,18 +123,27 @@ class MLAAttention(nn.Module):
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
                  layer_idx=0, ep_group=None, ep_size=1, tp_group=None,
-                 tp_size=1):
+                 tp_size=1, sp=False):
         super().__init__()
         h = cfg.hidden_size
         self.tp_group = tp_group
+        self.sp = sp
         self.input_norm = K.RMSNorm(h, dtype=dtype, device=device)
         if getattr(cfg, "attention_type", "gqa") == "mla":
             assert tp_size == 1, "MLA requires tp_size == 1 (simulator parity)"
             self.attention = MLAAttention(cfg, dtype=dtype, device=device)
         else:
             self.attention = GQAAttention(cfg, dtype=dtype, device=device,
-                                          tp_group=tp_group, tp_size=tp_size)
+                                          tp_group=tp_group, tp_size=tp_size,
+                                          sp=sp)
+        if sp:
+            # SP norms see only the local seq shard: their weight grads
+            # are partial sums and need a tp all_reduce (reducer handles
+            # params flagged this way)
+            self.input_norm.weight._needs_tp_grad_reduce = True
         self.pre_mlp_norm = K.RMSNorm(h, dtype=dtype, device=device)
+        if sp:
+            self.pre_mlp_norm.weight._needs_tp_grad_reduce = True
         assert cfg.use_swiglu
         self.use_moe = (cfg.model_type == "moe"
                         and layer_idx >= (cfg.dense_layers or 0))
@@ -149,9 +165,9 @@ class LlamaDecoderLayer(nn.Module):
                 self.fc2.weight._is_tp_shard = True
 
     def forward(self, x, rope_cs, pos):
-        from .tp import copy_to_tp, reduce_from_tp
+        from .tp import copy_to_tp, gather_seq, reduce_from_tp, scatter_seq
 
-        # x: [B, S, H]
+        # x: [B, S, H] (seq-sharded under SP)
         res = x
         y = self.input_norm(x)
         x = res + self.attention(y, rope_cs, pos)
@@ -159,6 +175,9 @@ class LlamaDecoderLayer(nn.Module):
         y = self.pre_mlp_norm(x)
         if self.use_moe:
             y = self.moe_mlp(y)
+        elif self.sp:
+            y = self.fc2(K.swiglu(self.fc1(gather_seq(y, self.tp_group))))
+            y = scatter_seq(y, self.tp_group)
         else:
             y = self.fc2(K.swiglu(self.fc1(copy_to_tp(y, self.tp_group))))
             y = reduce_from_tp(y, self.tp_group)
@@ -168,12 +187,13 @@ class LlamaDecoderLayer(nn.Module):
 class LlamaForTraining(nn.Module):
     def __init__(self, cfg: ModelConfig, seq_len: int, dtype=torch.bfloat16,
                  rope_base=500000.0, device=None, ep_group=None, ep_size=1,
-                 tp_group=None, tp_size=1, tp_rank=0):
+                 tp_group=None, tp_size=1, tp_rank=0, sp=False):
         super().__init__()
         self.cfg = cfg
         self.seq_len = seq_len
         self.tp_group = tp_group
         self.tp_size = tp_size
+        self.sp = sp and tp_size > 1
         assert cfg.vocab_size % tp_size == 0
         self.vocab_local = cfg.vocab_size // tp_size
         self.vocab_start = tp_rank * self.vocab_local
@@ -182,9 +202,12 @@ class LlamaForTraining(nn.Module):
         self.layers = nn.ModuleList(
             [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i,
                                ep_group=ep_group, ep_size=ep_size,
-                               tp_group=tp_group, tp_size=tp_size)
+                               tp_group=tp_group, tp_size=tp_size,
+                               sp=sp and tp_size > 1)
              for i in range(cfg.layer_num)])
         self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
+        if self.sp:
+            self.final_norm.weight._needs_tp_grad_reduce = True
         self.lm_head = K.FusedLinear(cfg.hidden_size, self.vocab_local,
                                      dtype=dtype, device=device)
         if tp_size > 1:
@@ -202,13 +225,20 @@ class LlamaForTraining(nn.Module):
         pos = (torch.arange(S, device=tokens.device, dtype=torch.int32)
                .repeat(B))
         x = self.embedding(tokens)
+        if self.sp:
+            from .tp import slice_seq
+
+            x = slice_seq(x, self.tp_group)
         for layer in self.layers:
             x = layer(x, self.rope_cs, pos)
         x = self.final_norm(x)
         if self.tp_size > 1:
-            from .tp import copy_to_tp, vocab_parallel_ce
+            from .tp import copy_to_tp, gather_seq, vocab_parallel_ce
 
-            logits = self.lm_head(copy_to_tp(x, self.tp_group))
+            if self.sp:
+                logits = self.lm_head(gather_seq(x, self.tp_group))
+            else:
+                logits = self.lm_head(copy_to_tp(x, self.tp_group))
             loss = vocab_parallel_ce(logits.reshape(B * S, -1),
                                      labels.reshape(-1), self.tp_group,
                                      self.vocab_start)

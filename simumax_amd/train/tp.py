@@ -158,3 +158,89 @@ class _VocabParallelCE(torch.autograd.Function):
 
 def vocab_parallel_ce(logits, labels, tp_group, vocab_start):
     return _VocabParallelCE.apply(logits, labels, tp_group, vocab_start)
+
+
+# ---------------------------------------------------------------------
+# Sequence parallelism (Megatron SP): activations between the TP regions
+# are sequence-sharded; the f/g all_reduces become all_gather /
+# reduce-scatter pairs. gloo has no reduce_scatter, so the CPU test path
+# emulates it with all_reduce + narrow (RCCL uses the native collective).
+# ---------------------------------------------------------------------
+def _reduce_scatter_seq(x, group):
+    """x [B, S, H] -> [B, S/tp, H] (sum of shards)."""
+    tp = dist.get_world_size(group)
+    r = dist.get_rank(group)
+    shard = x.shape[1] // tp
+    if dist.get_backend(group) == "nccl":
+        xt = x.contiguous().view(x.shape[0], tp, shard, -1)             .transpose(0, 1).contiguous()
+        out = torch.empty_like(xt[0])
+        dist.reduce_scatter_tensor(out, xt, group=group)
+        return out
+    x = x.contiguous()
+    dist.all_reduce(x, group=group)
+    return x[:, r * shard:(r + 1) * shard].contiguous()
+
+
+def _all_gather_seq(x, group):
+    """x [B, S/tp, H] -> [B, S, H]."""
+    tp = dist.get_world_size(group)
+    parts = [torch.empty_like(x) for _ in range(tp)]
+    dist.all_gather(parts, x.contiguous(), group=group)
+    return torch.cat(parts, dim=1)
+
+
+class _GatherSeq(torch.autograd.Function):
+    """fwd all_gather over seq; bwd reduce_scatter (the SP g-bar op
+    before column-parallel GEMMs)."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return _all_gather_seq(x, group)
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _reduce_scatter_seq(grad, ctx.group), None
+
+
+class _ScatterSeq(torch.autograd.Function):
+    """fwd reduce_scatter over seq; bwd all_gather (after row-parallel
+    GEMMs)."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return _reduce_scatter_seq(x, group)
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _all_gather_seq(grad, ctx.group), None
+
+
+def gather_seq(x, group):
+    return _GatherSeq.apply(x, group) if group is not None else x
+
+
+def scatter_seq(x, group):
+    return _ScatterSeq.apply(x, group) if group is not None else x
+
+
+class _SliceSeq(torch.autograd.Function):
+    """Plain seq slice (after the replicated embedding); backward
+    all_gathers so every rank reconstructs the full-seq gradient."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        tp = dist.get_world_size(group)
+        r = dist.get_rank(group)
+        shard = x.shape[1] // tp
+        return x[:, r * shard:(r + 1) * shard].contiguous()
+
+    @staticmethod
+    def backward(ctx, grad):
+        return _all_gather_seq(grad, ctx.group), None
+
+
+def slice_seq(x, group):
+    return _SliceSeq.apply(x, group) if group is not None else x

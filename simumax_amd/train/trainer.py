@@ -36,6 +36,7 @@ class TrainConfig:
     bucket_bytes: int = 100 * 1024**2
     ep_size: int = 1
     tp_size: int = 1
+    sequence_parallel: bool = False
 
 
 _EP_GROUPS = {}
@@ -144,7 +145,8 @@ class DataParallelGradReducer:
     DistributedDataParallel grad-buffer semantics)."""
 
     def __init__(self, opt, overlap: bool, bucket_bytes: int,
-                 edp_group=None, edp_size=1, dp_group=None, dp_size=None):
+                 edp_group=None, edp_size=1, dp_group=None, dp_size=None,
+                 tp_group=None):
         self.params = opt.params
         self.flat_grad = opt.flat_grad
         self.dense_numel = opt.dense_numel
@@ -153,6 +155,11 @@ class DataParallelGradReducer:
         self.edp_group = edp_group
         self.edp_size = edp_size
         self.dp_group = dp_group
+        self.tp_group = tp_group
+        # sequence-parallel norms produce PARTIAL weight grads (each tp
+        # rank saw only its seq shard): summed over tp before DP averaging
+        self._tp_partial = [p for p in self.params
+                            if getattr(p, "_needs_tp_grad_reduce", False)]
         world = dist.get_world_size() if dist.is_initialized() else 1
         self.dp_size = dp_size if dp_size is not None else world
         self.overlap = (overlap and dist.is_initialized()
@@ -242,6 +249,9 @@ class DataParallelGradReducer:
         return ids
 
     def finalize(self):
+        if self._tp_partial and self.tp_group is not None:
+            for p in self._tp_partial:
+                dist.all_reduce(p.main_grad, group=self.tp_group)
         if self.overlap:
             for h in self.handles:
                 h.wait()
@@ -285,12 +295,13 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
     model = LlamaForTraining(model_cfg, cfg.seq_len, device=device,
                              ep_group=ep_group, ep_size=cfg.ep_size,
                              tp_group=tp_group, tp_size=tp_size,
-                             tp_rank=tp_rank)
+                             tp_rank=tp_rank, sp=cfg.sequence_parallel)
     opt = MixedPrecisionAdam(model.parameters(), cfg)
     reducer = DataParallelGradReducer(opt, cfg.overlap_grad_reduce,
                                       cfg.bucket_bytes,
                                       edp_group=edp_group, edp_size=edp_size,
-                                      dp_group=dp_group, dp_size=dp_size)
+                                      dp_group=dp_group, dp_size=dp_size,
+                                      tp_group=tp_group)
     return model, opt, reducer
 
 

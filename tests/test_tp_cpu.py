@@ -52,7 +52,7 @@ def _shard(ref, model, rank, tp, cfg):
                 p.copy_(r)
 
 
-def _worker(rank, world, port, q):
+def _worker(rank, world, port, q, sp=False):
     import torch.distributed as dist
 
     sys.path.insert(0, REPO)
@@ -68,7 +68,8 @@ def _worker(rank, world, port, q):
 
         cfg = _tiny_cfg()
         tc = TrainConfig(seq_len=32, micro_batch_size=2, micro_batch_num=1,
-                         overlap_grad_reduce=False, tp_size=2)
+                         overlap_grad_reduce=False, tp_size=2,
+                         sequence_parallel=sp)
         model, opt, red = build_trainer(cfg, tc, "cpu")
 
         torch.manual_seed(1234)
@@ -125,12 +126,11 @@ def _worker(rank, world, port, q):
         dist.destroy_process_group()
 
 
-@pytest.mark.timeout(300)
-def test_tp2_matches_tp1_gradients():
+def _run(port, sp):
     mp.set_start_method("spawn", force=True)
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_worker, args=(r, 2, 29519, q))
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, q, sp))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -143,4 +143,16 @@ def test_tp2_matches_tp1_gradients():
         assert p.exitcode == 0
     for rank, errs in results.items():
         bad = {n: e for n, e in errs.items() if e > 4e-2}
-        assert not bad, f"rank {rank} mismatches: {bad}"
+        assert not bad, f"rank {rank} (sp={sp}) mismatches: {bad}"
+
+
+@pytest.mark.timeout(300)
+def test_tp2_matches_tp1_gradients():
+    _run(29519, sp=False)
+
+
+@pytest.mark.timeout(300)
+def test_tp2_sp_matches_tp1_gradients():
+    """Sequence parallelism: gather/scatter seq shards + partial-grad tp
+    reduction for the norms must still reproduce TP1 gradients."""
+    _run(29521, sp=True)
