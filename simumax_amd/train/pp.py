@@ -1,0 +1,213 @@
+"""Pipeline parallelism (1F1B) for the reference trainer.
+
+Mirrors the simulator's PP model (perf/perf_llm.py schedule_1f1b,
+sim/schedule.py): the layer stack is split into pp contiguous stages
+(stage 0 holds the embedding, the last stage final-norm + lm_head + CE);
+microbatches flow through the classic warmup / steady-1F1B / cooldown
+schedule with isend/recv at stage boundaries (non-blocking sends — see
+_isend — so the steady state cannot rendezvous-deadlock). Cross-stage
+autograd is stitched manually: each stage keeps (input, output) per
+in-flight microbatch, backward receives the output grad from the next
+stage and sends its input grad to the previous one.
+
+Rank layout matches core/utils.get_rank_group (pp outermost):
+stage = rank // dp, peers at rank ± dp. Combine with DP freely; TP/EP
+composition is a round-2 item.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..core.config import ModelConfig
+from ..kernels import ops as K
+from .model import LlamaDecoderLayer
+
+
+def stage_layer_range(layer_num, pp, stage):
+    assert layer_num % pp == 0, "pp trainer needs layer_num % pp == 0"
+    per = layer_num // pp
+    return stage * per, (stage + 1) * per
+
+
+class PipelineStageModel(nn.Module):
+    """One PP stage of LlamaForTraining (dense/GQA path)."""
+
+    def __init__(self, cfg: ModelConfig, seq_len: int, stage: int, pp: int,
+                 dtype=torch.bfloat16, rope_base=500000.0, device=None):
+        super().__init__()
+        self.cfg = cfg
+        self.stage = stage
+        self.pp = pp
+        lo, hi = stage_layer_range(cfg.layer_num, pp, stage)
+        if stage == 0:
+            self.embedding = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
+                                          dtype=dtype, device=device)
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i)
+             for i in range(lo, hi)])
+        if stage == pp - 1:
+            self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype,
+                                        device=device)
+            self.lm_head = K.FusedLinear(cfg.hidden_size, cfg.vocab_size,
+                                         dtype=dtype, device=device)
+        cs = K.build_rope_cache(seq_len, cfg.head_size, base=rope_base,
+                                device=device or "cpu")
+        self.register_buffer("rope_cs", cs, persistent=False)
+
+    def forward(self, x, labels=None):
+        # x: tokens [B, S] on stage 0, hidden [B, S, H] elsewhere
+        if self.stage == 0:
+            B, S = x.shape
+            pos = (torch.arange(S, device=x.device, dtype=torch.int32)
+                   .repeat(B))
+            x = self.embedding(x)
+        else:
+            B, S, _ = x.shape
+            pos = (torch.arange(S, device=x.device, dtype=torch.int32)
+                   .repeat(B))
+        for layer in self.layers:
+            x = layer(x, self.rope_cs, pos)
+        if self.stage == self.pp - 1:
+            x = self.final_norm(x)
+            logits = self.lm_head(x)
+            loss = K.fused_cross_entropy(
+                logits.reshape(B * S, -1), labels.reshape(-1))
+            return loss.mean()
+        return x
+
+
+_PP_GROUPS = {}
+
+
+def get_pp_layout(pp_size):
+    """Returns (stage, prev_rank, next_rank, dp_group, dp_size) for this
+    rank with pp as the OUTERMOST grid dimension."""
+    if pp_size <= 1 or not dist.is_initialized():
+        return 0, None, None, None, None
+    world = dist.get_world_size()
+    assert world % pp_size == 0
+    dp = world // pp_size
+    r = dist.get_rank()
+    stage = r // dp
+    prev_rank = r - dp if stage > 0 else None
+    next_rank = r + dp if stage < pp_size - 1 else None
+    key = (pp_size, world)
+    if key not in _PP_GROUPS:
+        groups = {}
+        for s in range(pp_size):
+            ranks = list(range(s * dp, (s + 1) * dp))
+            g = dist.new_group(ranks)
+            for rr in ranks:
+                groups[rr] = g
+        _PP_GROUPS[key] = groups
+    return stage, prev_rank, next_rank, _PP_GROUPS[key][r], dp
+
+
+def _isend(t, dst, pending):
+    """Non-blocking send (blocking rendezvous sends deadlock the 1F1B
+    steady state: a stage can sit in send-fwd while its peer sits in
+    send-grad). The tensor must stay alive until the wait."""
+    t = t.contiguous()
+    pending.append((dist.isend(t, dst), t))
+
+
+def _recv(shape, dtype, src, device):
+    t = torch.empty(shape, dtype=dtype, device=device)
+    dist.recv(t, src)
+    return t
+
+
+def pp_train_step(model: PipelineStageModel, opt, reducer, toks, labels,
+                  mbc, prev_rank, next_rank, hidden_shape, dtype):
+    """One optimizer step of 1F1B over mbc microbatches.
+
+    toks/labels: [mbc, B, S] (every rank gets the full set; stage 0 reads
+    toks, the last stage labels). Returns the mean loss on the last stage
+    (0.0 elsewhere).
+    """
+    stage, pp = model.stage, model.pp
+    dev = toks.device
+    opt.zero_grad()
+
+    warmup = min(pp - stage - 1, mbc)
+    in_flight = []           # (input_leaf_or_None, output)
+    pending = []             # outstanding isends (handle, tensor)
+    losses = []
+    nf = nb = 0
+
+    def fwd_one():
+        nonlocal nf
+        m = nf
+        if stage == 0:
+            inp = None
+            out = model(toks[m])
+        else:
+            h = _recv(hidden_shape, dtype, prev_rank, dev)
+            inp = h.requires_grad_(True)
+            out = model(inp, labels[m] if stage == pp - 1 else None)
+        if stage == pp - 1:
+            losses.append(out)
+        else:
+            _isend(out.detach(), next_rank, pending)
+        in_flight.append((inp, out))
+        nf += 1
+
+    def bwd_one(last):
+        nonlocal nb
+        reducer.reduce_this_pass = last
+        inp, out = in_flight.pop(0)
+        if stage == pp - 1:
+            out.backward()
+        else:
+            g = _recv(out.shape, dtype, next_rank, dev)
+            out.backward(gradient=g)
+        if stage > 0:
+            _isend(inp.grad, prev_rank, pending)
+        nb += 1
+
+    for _ in range(warmup):
+        fwd_one()
+    while nb < mbc:
+        if nf < mbc:
+            fwd_one()
+        bwd_one(last=(nb == mbc - 1))
+    for h, _t in pending:
+        h.wait()
+    reducer.finalize()
+    opt.step()
+    if losses:
+        return float(torch.stack([l.detach() for l in losses]).mean())
+    return 0.0
+
+
+def build_pp_trainer(model_cfg: ModelConfig, cfg, device="cpu"):
+    """PP analog of train.trainer.build_trainer (pp = cfg.pp_size)."""
+    from .trainer import DataParallelGradReducer, MixedPrecisionAdam
+
+    torch.manual_seed(1234)
+    model_cfg.maybe_pad_vocab_size(1)
+    stage, prev_rank, next_rank, dp_group, dp_size = get_pp_layout(cfg.pp_size)
+    model = PipelineStageModel(model_cfg, cfg.seq_len, stage, cfg.pp_size,
+                               device=device)
+    opt = MixedPrecisionAdam(model.parameters(), cfg)
+    if cfg.pp_size > 1 and dist.is_initialized():
+        # pp "model" group: one rank per stage with the same dp index
+        world = dist.get_world_size()
+        dp = world // cfg.pp_size
+        key = ("pp_mp", cfg.pp_size, world)
+        if key not in _PP_GROUPS:
+            groups = {}
+            for d in range(dp):
+                ranks = list(range(d, world, dp))
+                g = dist.new_group(ranks)
+                for rr in ranks:
+                    groups[rr] = g
+            _PP_GROUPS[key] = groups
+        opt.set_model_parallel_norm(_PP_GROUPS[key][dist.get_rank()])
+    reducer = DataParallelGradReducer(opt, cfg.overlap_grad_reduce,
+                                      cfg.bucket_bytes,
+                                      dp_group=dp_group, dp_size=dp_size)
+    return model, opt, reducer, (stage, prev_rank, next_rank)

@@ -36,6 +36,7 @@ class TrainConfig:
     bucket_bytes: int = 100 * 1024**2
     ep_size: int = 1
     tp_size: int = 1
+    pp_size: int = 1
     sequence_parallel: bool = False
 
 
@@ -103,6 +104,36 @@ class MixedPrecisionAdam:
         self.m = torch.zeros_like(self.master)
         self.v = torch.zeros_like(self.master)
         self.t = 0
+        # model-parallel grad-norm clipping (Megatron semantics: the clip
+        # uses the GLOBAL grad norm): sum shard-unique ||g||^2 over the
+        # model-parallel group, count replicated params once
+        self._norm_group = None
+        self._replicated_slices = []  # flat slices counted once
+
+    def set_model_parallel_norm(self, group, replicated_flag="__none__"):
+        """group: ranks holding DISTINCT model shards (pp stages, tp or
+        ep groups). replicated_flag: param attribute marking params that
+        are REPLICATED across that group (counted once, not summed)."""
+        self._norm_group = group
+        self._replicated_slices = []
+        off = 0
+        for p in self.params:
+            n = p.numel()
+            if getattr(p, replicated_flag, False):
+                self._replicated_slices.append((off, off + n))
+            off += n
+
+    def _global_grad_norm(self):
+        g = self.flat_grad
+        total_sq = g.pow(2).sum()
+        if self._norm_group is None:
+            return total_sq.sqrt()
+        rep_sq = g.new_zeros(())
+        for lo, hi in self._replicated_slices:
+            rep_sq += g[lo:hi].pow(2).sum()
+        uni_sq = total_sq - rep_sq
+        dist.all_reduce(uni_sq, group=self._norm_group)
+        return (uni_sq + rep_sq).sqrt()
 
     def zero_grad(self):
         self.flat_grad.zero_()
@@ -115,8 +146,9 @@ class MixedPrecisionAdam:
         self.t += 1
         b1, b2 = self.cfg.adam_betas
         g = self.flat_grad
-        # global grad-norm clip (Megatron clip_grad)
-        norm = g.norm(2)
+        # global grad-norm clip (Megatron clip_grad; global across the
+        # model-parallel group when configured)
+        norm = self._global_grad_norm()
         scale = self.cfg.grad_clip / (norm + 1e-6)
         if scale < 1.0:
             g.mul_(scale)
@@ -297,6 +329,15 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                              tp_group=tp_group, tp_size=tp_size,
                              tp_rank=tp_rank, sp=cfg.sequence_parallel)
     opt = MixedPrecisionAdam(model.parameters(), cfg)
+    if tp_group is not None:
+        # tp shards are unique; norms/embedding are replicated across tp
+        for p in opt.params:
+            p._replicated_tp = not getattr(p, "_is_tp_shard", False)
+        opt.set_model_parallel_norm(tp_group, "_replicated_tp")
+    elif ep_group is not None:
+        for p in opt.params:
+            p._replicated_ep = not getattr(p, "_is_expert", False)
+        opt.set_model_parallel_norm(ep_group, "_replicated_ep")
     reducer = DataParallelGradReducer(opt, cfg.overlap_grad_reduce,
                                       cfg.bucket_bytes,
                                       edp_group=edp_group, edp_size=edp_size,
