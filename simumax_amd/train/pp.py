@@ -33,26 +33,36 @@ def stage_layer_range(layer_num, pp, stage):
 
 
 class PipelineStageModel(nn.Module):
-    """One PP stage of LlamaForTraining (dense/GQA path)."""
+    """One PP stage of LlamaForTraining (dense/GQA path), optionally with
+    tensor parallelism inside the stage (tp x pp composition)."""
 
     def __init__(self, cfg: ModelConfig, seq_len: int, stage: int, pp: int,
-                 dtype=torch.bfloat16, rope_base=500000.0, device=None):
+                 dtype=torch.bfloat16, rope_base=500000.0, device=None,
+                 tp_group=None, tp_size=1, tp_rank=0):
         super().__init__()
         self.cfg = cfg
         self.stage = stage
         self.pp = pp
+        self.tp_group = tp_group
+        self.tp_size = tp_size
         lo, hi = stage_layer_range(cfg.layer_num, pp, stage)
         if stage == 0:
             self.embedding = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
                                           dtype=dtype, device=device)
         self.layers = nn.ModuleList(
-            [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i)
+            [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i,
+                               tp_group=tp_group, tp_size=tp_size)
              for i in range(lo, hi)])
         if stage == pp - 1:
+            assert cfg.vocab_size % tp_size == 0
+            self.vocab_local = cfg.vocab_size // tp_size
+            self.vocab_start = tp_rank * self.vocab_local
             self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype,
                                         device=device)
-            self.lm_head = K.FusedLinear(cfg.hidden_size, cfg.vocab_size,
+            self.lm_head = K.FusedLinear(cfg.hidden_size, self.vocab_local,
                                          dtype=dtype, device=device)
+            if tp_size > 1:
+                self.lm_head.weight._is_tp_shard = True
         cs = K.build_rope_cache(seq_len, cfg.head_size, base=rope_base,
                                 device=device or "cpu")
         self.register_buffer("rope_cs", cs, persistent=False)
@@ -72,9 +82,17 @@ class PipelineStageModel(nn.Module):
             x = layer(x, self.rope_cs, pos)
         if self.stage == self.pp - 1:
             x = self.final_norm(x)
-            logits = self.lm_head(x)
-            loss = K.fused_cross_entropy(
-                logits.reshape(B * S, -1), labels.reshape(-1))
+            if self.tp_size > 1:
+                from .tp import copy_to_tp, vocab_parallel_ce
+
+                logits = self.lm_head(copy_to_tp(x, self.tp_group))
+                loss = vocab_parallel_ce(logits.reshape(B * S, -1),
+                                         labels.reshape(-1), self.tp_group,
+                                         self.vocab_start)
+            else:
+                logits = self.lm_head(x)
+                loss = K.fused_cross_entropy(
+                    logits.reshape(B * S, -1), labels.reshape(-1))
             return loss.mean()
         return x
 
@@ -184,30 +202,27 @@ def pp_train_step(model: PipelineStageModel, opt, reducer, toks, labels,
 
 
 def build_pp_trainer(model_cfg: ModelConfig, cfg, device="cpu"):
-    """PP analog of train.trainer.build_trainer (pp = cfg.pp_size)."""
+    """PP (optionally x TP) analog of train.trainer.build_trainer
+    (pp = cfg.pp_size, tp = cfg.tp_size). Returns
+    (model, opt, reducer, parallel_state)."""
+    from .parallel_state import init_parallel_state
     from .trainer import DataParallelGradReducer, MixedPrecisionAdam
 
     torch.manual_seed(1234)
-    model_cfg.maybe_pad_vocab_size(1)
-    stage, prev_rank, next_rank, dp_group, dp_size = get_pp_layout(cfg.pp_size)
-    model = PipelineStageModel(model_cfg, cfg.seq_len, stage, cfg.pp_size,
-                               device=device)
+    model_cfg.maybe_pad_vocab_size(cfg.tp_size)
+    ps = init_parallel_state(tp_size=cfg.tp_size, pp_size=cfg.pp_size)
+    model = PipelineStageModel(model_cfg, cfg.seq_len, ps.stage, cfg.pp_size,
+                               device=device, tp_group=ps.tp_group,
+                               tp_size=cfg.tp_size, tp_rank=ps.tp_rank)
     opt = MixedPrecisionAdam(model.parameters(), cfg)
-    if cfg.pp_size > 1 and dist.is_initialized():
-        # pp "model" group: one rank per stage with the same dp index
-        world = dist.get_world_size()
-        dp = world // cfg.pp_size
-        key = ("pp_mp", cfg.pp_size, world)
-        if key not in _PP_GROUPS:
-            groups = {}
-            for d in range(dp):
-                ranks = list(range(d, world, dp))
-                g = dist.new_group(ranks)
-                for rr in ranks:
-                    groups[rr] = g
-            _PP_GROUPS[key] = groups
-        opt.set_model_parallel_norm(_PP_GROUPS[key][dist.get_rank()])
+    if ps.tp_group is not None or ps.pp_norm_group is not None:
+        for p in opt.params:
+            p._replicated_tp = not getattr(p, "_is_tp_shard", False)
+        opt.set_model_parallel_norm(ps.tp_group, "_replicated_tp",
+                                    pp_group=ps.pp_norm_group)
     reducer = DataParallelGradReducer(opt, cfg.overlap_grad_reduce,
                                       cfg.bucket_bytes,
-                                      dp_group=dp_group, dp_size=dp_size)
-    return model, opt, reducer, (stage, prev_rank, next_rank)
+                                      dp_group=ps.dp_group,
+                                      dp_size=ps.dp_size,
+                                      tp_group=ps.tp_group)
+    return model, opt, reducer, ps

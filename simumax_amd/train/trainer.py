@@ -108,13 +108,17 @@ class MixedPrecisionAdam:
         # uses the GLOBAL grad norm): sum shard-unique ||g||^2 over the
         # model-parallel group, count replicated params once
         self._norm_group = None
+        self._pp_norm_group = None
         self._replicated_slices = []  # flat slices counted once
 
-    def set_model_parallel_norm(self, group, replicated_flag="__none__"):
-        """group: ranks holding DISTINCT model shards (pp stages, tp or
-        ep groups). replicated_flag: param attribute marking params that
-        are REPLICATED across that group (counted once, not summed)."""
+    def set_model_parallel_norm(self, group, replicated_flag="__none__",
+                                pp_group=None):
+        """group: ranks holding DISTINCT shards of THIS stage's params
+        (tp or ep groups); replicated_flag marks params replicated across
+        it (counted once). pp_group: one rank per pipeline stage — the
+        per-stage totals are then summed across stages."""
         self._norm_group = group
+        self._pp_norm_group = pp_group
         self._replicated_slices = []
         off = 0
         for p in self.params:
@@ -126,14 +130,21 @@ class MixedPrecisionAdam:
     def _global_grad_norm(self):
         g = self.flat_grad
         total_sq = g.pow(2).sum()
-        if self._norm_group is None:
+        if self._norm_group is None and getattr(self, "_pp_norm_group",
+                                                None) is None:
             return total_sq.sqrt()
-        rep_sq = g.new_zeros(())
-        for lo, hi in self._replicated_slices:
-            rep_sq += g[lo:hi].pow(2).sum()
-        uni_sq = total_sq - rep_sq
-        dist.all_reduce(uni_sq, group=self._norm_group)
-        return (uni_sq + rep_sq).sqrt()
+        if self._norm_group is not None:
+            rep_sq = g.new_zeros(())
+            for lo, hi in self._replicated_slices:
+                rep_sq += g[lo:hi].pow(2).sum()
+            uni_sq = total_sq - rep_sq
+            dist.all_reduce(uni_sq, group=self._norm_group)
+            stage_sq = uni_sq + rep_sq
+        else:
+            stage_sq = total_sq
+        if getattr(self, "_pp_norm_group", None) is not None:
+            dist.all_reduce(stage_sq, group=self._pp_norm_group)
+        return stage_sq.sqrt()
 
     def zero_grad(self):
         self.flat_grad.zero_()

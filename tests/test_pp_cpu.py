@@ -44,8 +44,8 @@ def _worker(rank, world, port, q):
         tc = TrainConfig(seq_len=32, micro_batch_size=2, micro_batch_num=mbc,
                          overlap_grad_reduce=False, pp_size=2, lr=0.0,
                          grad_clip=1e9)
-        model, opt, red, (stage, prev_rank, next_rank) = build_pp_trainer(
-            cfg, tc, "cpu")
+        model, opt, red, ps = build_pp_trainer(cfg, tc, "cpu")
+        stage, prev_rank, next_rank = ps.stage, ps.pp_prev, ps.pp_next
 
         torch.manual_seed(1234)
         ref = LlamaForTraining(cfg, tc.seq_len, device="cpu")
@@ -117,4 +117,121 @@ def test_pp2_matches_single_process_gradients():
         assert p.exitcode == 0
     for rank, errs in results.items():
         bad = {n: e for n, e in errs.items() if e > 1e-3}
+        assert not bad, f"rank {rank} mismatches: {bad}"
+
+
+def _tp_shard_name_map(cfg, tp, rank_tp, lo):
+    """Return fn(name, ref_tensor) -> sharded view for tp x pp."""
+    d = cfg.head_size
+    hq, hkv = cfg.head_num, cfg.kv_head_num
+    hq_l, hkv_l = hq // tp, hkv // tp
+    i_l = cfg.intermediate_size // tp
+    v_l = cfg.vocab_size // tp
+    I = cfg.intermediate_size
+
+    def shard(name, r):
+        if "qkv_proj" in name:
+            return torch.cat([
+                r[:hq * d][rank_tp * hq_l * d:(rank_tp + 1) * hq_l * d],
+                r[hq * d:(hq + hkv) * d][rank_tp * hkv_l * d:(rank_tp + 1) * hkv_l * d],
+                r[(hq + hkv) * d:][rank_tp * hkv_l * d:(rank_tp + 1) * hkv_l * d]])
+        if "out_proj" in name:
+            return r[:, rank_tp * hq_l * d:(rank_tp + 1) * hq_l * d]
+        if "fc1" in name:
+            return torch.cat([r[:I][rank_tp * i_l:(rank_tp + 1) * i_l],
+                              r[I:][rank_tp * i_l:(rank_tp + 1) * i_l]])
+        if "fc2" in name:
+            return r[:, rank_tp * i_l:(rank_tp + 1) * i_l]
+        if "lm_head" in name:
+            return r[rank_tp * v_l:(rank_tp + 1) * v_l]
+        return r
+
+    def ref_name(name):
+        if name.startswith("layers."):
+            parts = name.split(".")
+            parts[1] = str(int(parts[1]) + lo)
+            return ".".join(parts)
+        return name
+
+    return shard, ref_name
+
+
+def _worker_tp_pp(rank, world, port, q):
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from simumax_amd.train.model import LlamaForTraining
+        from simumax_amd.train.pp import (build_pp_trainer, pp_train_step,
+                                          stage_layer_range)
+        from simumax_amd.train.trainer import (MixedPrecisionAdam,
+                                               TrainConfig,
+                                               accumulate_main_grads,
+                                               make_synthetic_batch)
+
+        cfg = _tiny_cfg()
+        mbc = 2
+        tc = TrainConfig(seq_len=32, micro_batch_size=2, micro_batch_num=mbc,
+                         overlap_grad_reduce=False, pp_size=2, tp_size=2,
+                         lr=0.0, grad_clip=1e9)
+        model, opt, red, ps = build_pp_trainer(cfg, tc, "cpu")
+        lo, hi = stage_layer_range(cfg.layer_num, 2, ps.stage)
+        shard, ref_name = _tp_shard_name_map(cfg, 2, ps.tp_rank, lo)
+
+        torch.manual_seed(1234)
+        ref = LlamaForTraining(cfg, tc.seq_len, device="cpu")
+        ref_opt = MixedPrecisionAdam(ref.parameters(), tc)
+        rd = dict(ref.named_parameters())
+        with torch.no_grad():
+            for name, p in model.named_parameters():
+                p.copy_(shard(name, rd[ref_name(name)]))
+
+        toks, labels = make_synthetic_batch(cfg.vocab_size, mbc, 2, 32,
+                                            "cpu", seed=9)
+        loss = pp_train_step(model, opt, red, toks, labels, mbc,
+                             ps.pp_prev, ps.pp_next,
+                             (2, 32, cfg.hidden_size), torch.bfloat16)
+
+        ref_opt.zero_grad()
+        for m in range(mbc):
+            ref(toks[m], labels[m]).backward()
+            accumulate_main_grads(ref_opt.params)
+
+        rg = {n: p.main_grad for n, p in ref.named_parameters()}
+        errs = {}
+        for name, p in model.named_parameters():
+            want = shard(name, rg[ref_name(name)])
+            denom = want.abs().max().clamp(min=1e-4)
+            errs[name] = float((p.main_grad - want).abs().max() / denom)
+        gn = float(opt._global_grad_norm())
+        ref_gn = float(ref_opt.flat_grad.norm())
+        errs["global_grad_norm"] = abs(gn - ref_gn) / max(ref_gn, 1e-6)
+        q.put((rank, errs))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(400)
+def test_tp2_pp2_matches_single_process_gradients():
+    """Composed tp2 x pp2 on 4 ranks: sharded + staged grads and the
+    two-level global grad norm must match the single-process run."""
+    mp.set_start_method("spawn", force=True)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_tp_pp, args=(r, 4, 29527, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, errs = q.get(timeout=300)
+        results[rank] = errs
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    for rank, errs in results.items():
+        bad = {n: e for n, e in errs.items() if e > 4e-2}
         assert not bad, f"rank {rank} mismatches: {bad}"
