@@ -235,3 +235,95 @@ def test_tp2_pp2_matches_single_process_gradients():
     for rank, errs in results.items():
         bad = {n: e for n, e in errs.items() if e > 4e-2}
         assert not bad, f"rank {rank} mismatches: {bad}"
+
+
+def _worker_pp_dp(rank, world, port, q):
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from simumax_amd.train.model import LlamaForTraining
+        from simumax_amd.train.pp import (build_pp_trainer, pp_train_step,
+                                          stage_layer_range)
+        from simumax_amd.train.trainer import (MixedPrecisionAdam,
+                                               TrainConfig,
+                                               accumulate_main_grads,
+                                               make_synthetic_batch)
+
+        cfg = _tiny_cfg()
+        mbc = 2
+        tc = TrainConfig(seq_len=32, micro_batch_size=2, micro_batch_num=mbc,
+                         overlap_grad_reduce=False, pp_size=2, lr=0.0,
+                         grad_clip=1e9)
+        model, opt, red, ps = build_pp_trainer(cfg, tc, "cpu")
+        assert ps.dp_size == 2
+        lo, hi = stage_layer_range(cfg.layer_num, 2, ps.stage)
+
+        torch.manual_seed(1234)
+        ref = LlamaForTraining(cfg, tc.seq_len, device="cpu")
+        ref_opt = MixedPrecisionAdam(ref.parameters(), tc)
+        rd = dict(ref.named_parameters())
+
+        def ref_name(name):
+            if name.startswith("layers."):
+                parts = name.split(".")
+                parts[1] = str(int(parts[1]) + lo)
+                return ".".join(parts)
+            return name
+
+        with torch.no_grad():
+            for name, p in model.named_parameters():
+                p.copy_(rd[ref_name(name)])
+
+        # 2 dp columns x mbc microbatches of distinct data
+        toks, labels = make_synthetic_batch(cfg.vocab_size, 2 * mbc, 2, 32,
+                                            "cpu", seed=13)
+        my_toks = toks[ps.dp_rank * mbc:(ps.dp_rank + 1) * mbc]
+        my_labels = labels[ps.dp_rank * mbc:(ps.dp_rank + 1) * mbc]
+        pp_train_step(model, opt, red, my_toks, my_labels, mbc,
+                      ps.pp_prev, ps.pp_next, (2, 32, cfg.hidden_size),
+                      torch.bfloat16)
+
+        # reference: ALL 2*mbc microbatches, then compare against the
+        # dp-averaged (1/2) distributed grads
+        ref_opt.zero_grad()
+        for m in range(2 * mbc):
+            ref(toks[m], labels[m]).backward()
+            accumulate_main_grads(ref_opt.params)
+
+        rg = {n: p.main_grad for n, p in ref.named_parameters()}
+        errs = {}
+        for name, p in model.named_parameters():
+            want = rg[ref_name(name)]
+            got = p.main_grad * 2  # dp averaging factor
+            denom = want.abs().max().clamp(min=1e-4)
+            errs[name] = float((got - want).abs().max() / denom)
+        q.put((rank, errs))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(400)
+def test_pp2_dp2_data_sharding():
+    """pp2 x dp2 on 4 ranks: per-column data shards, dense grads
+    dp-averaged within each stage."""
+    mp.set_start_method("spawn", force=True)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_pp_dp, args=(r, 4, 29529, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(4):
+        rank, errs = q.get(timeout=300)
+        results[rank] = errs
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    for rank, errs in results.items():
+        bad = {n: e for n, e in errs.items() if e > 4e-2}
+        assert not bad, f"rank {rank} mismatches: {bad}"
