@@ -97,33 +97,6 @@ class PipelineStageModel(nn.Module):
         return x
 
 
-_PP_GROUPS = {}
-
-
-def get_pp_layout(pp_size):
-    """Returns (stage, prev_rank, next_rank, dp_group, dp_size) for this
-    rank with pp as the OUTERMOST grid dimension."""
-    if pp_size <= 1 or not dist.is_initialized():
-        return 0, None, None, None, None
-    world = dist.get_world_size()
-    assert world % pp_size == 0
-    dp = world // pp_size
-    r = dist.get_rank()
-    stage = r // dp
-    prev_rank = r - dp if stage > 0 else None
-    next_rank = r + dp if stage < pp_size - 1 else None
-    key = (pp_size, world)
-    if key not in _PP_GROUPS:
-        groups = {}
-        for s in range(pp_size):
-            ranks = list(range(s * dp, (s + 1) * dp))
-            g = dist.new_group(ranks)
-            for rr in ranks:
-                groups[rr] = g
-        _PP_GROUPS[key] = groups
-    return stage, prev_rank, next_rank, _PP_GROUPS[key][r], dp
-
-
 def _isend(t, dst, pending):
     """Non-blocking send (blocking rendezvous sends deadlock the 1F1B
     steady state: a stage can sit in send-fwd while its peer sits in
