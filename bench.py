@@ -36,6 +36,10 @@ def parse_args():
     ap.add_argument("--mbc", type=int, default=MICRO_BATCH_NUM)
     ap.add_argument("--layers", type=int, default=0,
                     help="override layer count (0 = full model)")
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor-parallel degree (world must be tp*pp*dp)")
+    ap.add_argument("--pp", type=int, default=1,
+                    help="pipeline-parallel degree (1F1B)")
     return ap.parse_args()
 
 
@@ -49,7 +53,7 @@ def predict(model_cfg, world, args):
         micro_batch_size=args.mbs,
         micro_batch_num=args.mbc,
         world_size=world,
-        tp_size=1, pp_size=1, ep_size=1,
+        tp_size=args.tp, pp_size=args.pp, ep_size=1,
         enable_sequence_parallel=False,
         zero_state=0,                # trainer replicates optimizer state
         use_fp32_accum_grad=True,
@@ -96,26 +100,45 @@ def main():
                                            make_synthetic_batch, train_step)
 
     tc = TrainConfig(seq_len=args.seq_len, micro_batch_size=args.mbs,
-                     micro_batch_num=args.mbc)
+                     micro_batch_num=args.mbc, tp_size=args.tp,
+                     pp_size=args.pp)
     device = f"cuda:{local_rank}"
     t0 = time.time()
-    model, opt, reducer = build_trainer(model_cfg, tc, device)
+    ps = None
+    if args.pp > 1:
+        from simumax_amd.train.pp import build_pp_trainer, pp_train_step
+
+        model, opt, reducer, ps = build_pp_trainer(model_cfg, tc, device)
+
+        def step_fn(m, o, r, t, l, mbc):
+            return pp_train_step(
+                m, o, r, t, l, mbc, ps.pp_prev, ps.pp_next,
+                (args.mbs, args.seq_len, model_cfg.hidden_size),
+                __import__("torch").bfloat16)
+    else:
+        model, opt, reducer = build_trainer(model_cfg, tc, device)
+        step_fn = train_step
     if rank == 0:
-        print(f"[bench] built {args.model} ({model.num_params()/1e9:.2f}B params) "
+        n_params = sum(p.numel() for p in model.parameters())
+        print(f"[bench] built {args.model} ({n_params/1e9:.2f}B params"
+              f"{' on this rank' if args.tp * args.pp > 1 else ''}) "
               f"in {time.time()-t0:.1f}s", file=sys.stderr)
+    # each DATA-parallel column gets distinct data (tp/pp peers share it)
+    dp_rank = ps.dp_rank if ps is not None else (
+        rank // args.tp if args.tp > 1 else rank)
     toks, labels = make_synthetic_batch(model_cfg.vocab_size, args.mbc,
                                         args.mbs, args.seq_len, device,
-                                        seed=1000 + rank)
+                                        seed=1000 + dp_rank)
 
     for _ in range(args.warmup):
-        train_step(model, opt, reducer, toks, labels, args.mbc)
+        step_fn(model, opt, reducer, toks, labels, args.mbc)
     torch.cuda.reset_peak_memory_stats()
     if distributed:
         dist.barrier()
     torch.cuda.synchronize()
     t_start = time.time()
     for _ in range(args.steps):
-        train_step(model, opt, reducer, toks, labels, args.mbc)
+        step_fn(model, opt, reducer, toks, labels, args.mbc)
     if distributed:
         dist.barrier()
     torch.cuda.synchronize()
@@ -131,7 +154,8 @@ def main():
     peak_bytes = pk.item()
 
     if rank == 0:
-        tokens_per_iter = args.mbs * args.mbc * world * args.seq_len
+        dp = world // (args.tp * args.pp)
+        tokens_per_iter = args.mbs * args.mbc * dp * args.seq_len
         flops_token = model_cfg.flops_per_token(args.seq_len)
         peak_tflops = 2500.0
         measured_mfu = (flops_token * tokens_per_iter / (ms_per_step / 1e3)
@@ -162,9 +186,11 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": args.model,
-                "global_batch": args.mbs * args.mbc * world,
+                "global_batch": args.mbs * args.mbc * dp,
                 "seq_len": args.seq_len,
-                "parallelism": f"dp{world}",
+                "parallelism": (f"tp{args.tp}." if args.tp > 1 else "")
+                               + (f"pp{args.pp}." if args.pp > 1 else "")
+                               + f"dp{dp}",
                 "layers": model_cfg.layer_num,
                 "timing_error_pct": round(time_err, 3),
                 "mem_error_pct": round(mem_err, 3),
