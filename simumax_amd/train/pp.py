@@ -199,3 +199,171 @@ def build_pp_trainer(model_cfg: ModelConfig, cfg, device="cpu"):
                                       dp_size=ps.dp_size,
                                       tp_group=ps.tp_group)
     return model, opt, reducer, ps
+
+
+# ---------------------------------------------------------------------
+# Interleaved VPP (virtual pipeline) training — Megatron sync-VPP
+# semantics mirrored from perf/vpp.py's schedule helpers (the same
+# chunk_id_of / mb_id_of table the analytic scheduler and the event
+# simulator replay).
+# ---------------------------------------------------------------------
+class VppChunkModel(nn.Module):
+    """One VIRTUAL stage v = chunk*pp + stage: a slice of the layer
+    stack, embedding on v==0, head on v==nv-1."""
+
+    def __init__(self, cfg: ModelConfig, seq_len: int, v: int, nv: int,
+                 dtype=torch.bfloat16, rope_base=500000.0, device=None):
+        super().__init__()
+        assert cfg.layer_num % nv == 0, "vpp needs layer_num % (pp*vp) == 0"
+        per = cfg.layer_num // nv
+        self.v = v
+        self.nv = nv
+        if v == 0:
+            self.embedding = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
+                                          dtype=dtype, device=device)
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i)
+             for i in range(v * per, (v + 1) * per)])
+        if v == nv - 1:
+            self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype,
+                                        device=device)
+            self.lm_head = K.FusedLinear(cfg.hidden_size, cfg.vocab_size,
+                                         dtype=dtype, device=device)
+        cs = K.build_rope_cache(seq_len, cfg.head_size, base=rope_base,
+                                device=device or "cpu")
+        self.register_buffer("rope_cs", cs, persistent=False)
+
+    def forward(self, x, labels=None):
+        if self.v == 0:
+            B, S = x.shape
+            x = self.embedding(x)
+        else:
+            B, S, _ = x.shape
+        pos = (torch.arange(S, device=x.device, dtype=torch.int32).repeat(B))
+        for layer in self.layers:
+            x = layer(x, self.rope_cs, pos)
+        if self.v == self.nv - 1:
+            x = self.final_norm(x)
+            logits = self.lm_head(x)
+            loss = K.fused_cross_entropy(
+                logits.reshape(B * S, -1), labels.reshape(-1))
+            return loss.mean()
+        return x
+
+
+class VppStageModel(nn.Module):
+    """All vp chunks living on one pipeline stage."""
+
+    def __init__(self, cfg: ModelConfig, seq_len: int, stage: int, pp: int,
+                 vp: int, dtype=torch.bfloat16, device=None):
+        super().__init__()
+        self.stage = stage
+        self.pp = pp
+        self.vp = vp
+        self.nv = pp * vp
+        self.chunks = nn.ModuleList(
+            [VppChunkModel(cfg, seq_len, c * pp + stage, self.nv,
+                           dtype=dtype, device=device) for c in range(vp)])
+
+
+def vpp_train_step(model: VppStageModel, opt, reducer, toks, labels, mbc,
+                   ps, hidden_shape, dtype):
+    """One optimizer step of the Megatron interleaved schedule
+    (perf/vpp.py stream: warmup (pp-stage-1)*2 + (vp-1)*pp, then 1F1B
+    over k-indices decoded by chunk_id_of/mb_id_of). Virtual-stage
+    boundaries v->v+1 cross to the next pipeline stage for the same
+    chunk, wrapping from the last stage to stage 0 of the next chunk."""
+    from ..perf.vpp import chunk_id_of, mb_id_of
+
+    stage, pp, vp, nv = model.stage, model.pp, model.vp, model.nv
+    assert mbc % pp == 0, "interleaved schedule requires mbc % pp == 0"
+    dev = toks.device
+    stage_span = (dist.get_world_size() // pp if dist.is_initialized() else 1)
+    r = dist.get_rank() if dist.is_initialized() else 0
+
+    def rank_of_stage(s):
+        return s * stage_span + (r % stage_span)
+
+    opt.zero_grad()
+    total = mbc * vp
+    warm = min((pp - stage - 1) * 2 + (vp - 1) * pp, total)
+    inflight = {}            # (chunk, mb) -> (input_leaf, output)
+    pending = []
+    losses = []
+    nf = nb = 0
+
+    def fwd_one():
+        nonlocal nf
+        k = nf
+        c = chunk_id_of(k, pp, vp, True)
+        m = mb_id_of(k, pp, vp)
+        v = c * pp + stage
+        if v == 0:
+            inp = None
+            out = model.chunks[c](toks[m])
+        else:
+            src = rank_of_stage(stage - 1 if stage > 0 else pp - 1)
+            h = _recv(hidden_shape, dtype, src, dev)
+            inp = h.requires_grad_(True)
+            out = model.chunks[c](inp, labels[m] if v == nv - 1 else None)
+        if v == nv - 1:
+            losses.append(out)
+        else:
+            dst = rank_of_stage(stage + 1 if stage < pp - 1 else 0)
+            _isend(out.detach(), dst, pending)
+        inflight[(c, m)] = (inp, out)
+        nf += 1
+
+    def bwd_one(last):
+        nonlocal nb
+        k = nb
+        c = chunk_id_of(k, pp, vp, False)
+        m = mb_id_of(k, pp, vp)
+        v = c * pp + stage
+        reducer.reduce_this_pass = last
+        inp, out = inflight.pop((c, m))
+        if v == nv - 1:
+            out.backward()
+        else:
+            src = rank_of_stage(stage + 1 if stage < pp - 1 else 0)
+            g = _recv(out.shape, dtype, src, dev)
+            out.backward(gradient=g)
+        if v > 0:
+            dst = rank_of_stage(stage - 1 if stage > 0 else pp - 1)
+            _isend(inp.grad, dst, pending)
+        nb += 1
+
+    for _ in range(warm):
+        fwd_one()
+    while nb < total:
+        if nf < total:
+            fwd_one()
+        bwd_one(last=(nb == total - 1))
+    for h, _t in pending:
+        h.wait()
+    reducer.finalize()
+    opt.step()
+    if losses:
+        return float(torch.stack([l.detach() for l in losses]).mean())
+    return 0.0
+
+
+def build_vpp_trainer(model_cfg: ModelConfig, cfg, vp: int, device="cpu"):
+    """Interleaved-VPP analog of build_pp_trainer (tp inside stages is a
+    round-2 composition)."""
+    from .parallel_state import init_parallel_state
+    from .trainer import DataParallelGradReducer, MixedPrecisionAdam
+
+    torch.manual_seed(1234)
+    model_cfg.maybe_pad_vocab_size(1)
+    ps = init_parallel_state(tp_size=1, pp_size=cfg.pp_size)
+    model = VppStageModel(model_cfg, cfg.seq_len, ps.stage, cfg.pp_size, vp,
+                          device=device)
+    opt = MixedPrecisionAdam(model.parameters(), cfg)
+    if ps.pp_norm_group is not None:
+        opt.set_model_parallel_norm(None, pp_group=ps.pp_norm_group)
+    reducer = DataParallelGradReducer(opt, cfg.overlap_grad_reduce,
+                                      cfg.bucket_bytes,
+                                      dp_group=ps.dp_group,
+                                      dp_size=ps.dp_size)
+    return model, opt, reducer, ps

@@ -327,3 +327,97 @@ def test_pp2_dp2_data_sharding():
     for rank, errs in results.items():
         bad = {n: e for n, e in errs.items() if e > 4e-2}
         assert not bad, f"rank {rank} mismatches: {bad}"
+
+
+def _worker_vpp(rank, world, port, q, mbc=2):
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from simumax_amd.train.model import LlamaForTraining
+        from simumax_amd.train.pp import build_vpp_trainer, vpp_train_step
+        from simumax_amd.train.trainer import (MixedPrecisionAdam,
+                                               TrainConfig,
+                                               accumulate_main_grads,
+                                               make_synthetic_batch)
+
+        cfg = _tiny_cfg()          # 4 layers -> nv=4 with pp2 x vp2
+        pp, vp = 2, 2
+        tc = TrainConfig(seq_len=32, micro_batch_size=2, micro_batch_num=mbc,
+                         overlap_grad_reduce=False, pp_size=pp, lr=0.0,
+                         grad_clip=1e9)
+        model, opt, red, ps = build_vpp_trainer(cfg, tc, vp, "cpu")
+
+        torch.manual_seed(1234)
+        ref = LlamaForTraining(cfg, tc.seq_len, device="cpu")
+        ref_opt = MixedPrecisionAdam(ref.parameters(), tc)
+        rd = dict(ref.named_parameters())
+
+        def ref_name(name):
+            # chunks.<c>.layers.<i>.rest -> layers.<v>.rest (per=1)
+            parts = name.split(".")
+            c = int(parts[1])
+            v = c * pp + ps.stage
+            if parts[2] == "layers":
+                rest = ".".join(parts[4:])
+                return f"layers.{v}.{rest}"
+            return ".".join(parts[2:])
+
+        with torch.no_grad():
+            for name, p in model.named_parameters():
+                p.copy_(rd[ref_name(name)])
+
+        toks, labels = make_synthetic_batch(cfg.vocab_size, mbc, 2, 32,
+                                            "cpu", seed=17)
+        loss = vpp_train_step(model, opt, red, toks, labels, mbc, ps,
+                              (2, 32, cfg.hidden_size), torch.bfloat16)
+
+        ref_opt.zero_grad()
+        ref_losses = []
+        for m in range(mbc):
+            l = ref(toks[m], labels[m])
+            l.backward()
+            accumulate_main_grads(ref_opt.params)
+            ref_losses.append(float(l.detach()))
+
+        rg = {n: p.main_grad for n, p in ref.named_parameters()}
+        errs = {}
+        for name, p in model.named_parameters():
+            want = rg[ref_name(name)]
+            denom = want.abs().max().clamp(min=1e-4)
+            errs[name] = float((p.main_grad - want).abs().max() / denom)
+        if ps.stage == pp - 1:
+            mref = sum(ref_losses) / mbc
+            errs["loss"] = abs(loss - mref) / max(abs(mref), 1e-6)
+        q.put((rank, errs))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(400)
+@pytest.mark.parametrize("mbc,port", [(2, 29531), (4, 29533)])
+def test_vpp2_matches_single_process_gradients(mbc, port):
+    """Interleaved VPP (pp2 x vp2, 4 virtual stages on 2 ranks): the
+    Megatron schedule-table replay must reproduce the single-process
+    gradients exactly — mbc=4 exercises the steady-state channel
+    orderings beyond the warmup-dominated mbc=2 case."""
+    mp.set_start_method("spawn", force=True)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_vpp, args=(r, 2, port, q, mbc))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, errs = q.get(timeout=300)
+        results[rank] = errs
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    for rank, errs in results.items():
+        bad = {n: e for n, e in errs.items() if e > 1e-3}
+        assert not bad, f"rank {rank} mismatches: {bad}"
