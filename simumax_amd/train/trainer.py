@@ -38,6 +38,7 @@ class TrainConfig:
     tp_size: int = 1
     pp_size: int = 1
     sequence_parallel: bool = False
+    zero_state: int = 0   # 1 = ZeRO-1 distributed optimizer (sharded state)
 
 
 _EP_GROUPS = {}
@@ -79,7 +80,7 @@ class MixedPrecisionAdam:
     step is ~8 full-size kernels instead of ~3 small kernels per param —
     matching the calibrated optimizer bandwidth."""
 
-    def __init__(self, params, cfg: TrainConfig):
+    def __init__(self, params, cfg: TrainConfig, zero_group=None):
         params = [p for p in params if p.requires_grad]
         # dense params first, expert params last: the flat grad buffer then
         # splits into one contiguous slice per reduction group
@@ -90,9 +91,24 @@ class MixedPrecisionAdam:
         self.cfg = cfg
         total = sum(p.numel() for p in self.params)
         dev = self.params[0].device
-        self.flat_param = torch.empty(total, dtype=self.params[0].dtype,
+        # ZeRO-1 (Megatron distributed optimizer): fp32 master/m/v exist
+        # only for this rank's flat shard; params are all_gathered after
+        # the step. Buffers are padded so the shards are equal-sized.
+        self.zero = (cfg.zero_state == 1 and dist.is_initialized()
+                     and dist.get_world_size(zero_group) > 1)
+        self.zero_group = zero_group if self.zero else None
+        if self.zero:
+            zr = dist.get_rank(self.zero_group)
+            zw = dist.get_world_size(self.zero_group)
+            pad = (total + zw - 1) // zw * zw
+            self.shard = slice(zr * pad // zw, (zr + 1) * pad // zw)
+            self.zero_world = zw
+        else:
+            pad = total
+            self.shard = slice(0, total)
+        self.flat_param = torch.empty(pad, dtype=self.params[0].dtype,
                                       device=dev)
-        self.flat_grad = torch.zeros(total, dtype=torch.float32, device=dev)
+        self.flat_grad = torch.zeros(pad, dtype=torch.float32, device=dev)
         off = 0
         for p in self.params:
             n = p.numel()
@@ -100,7 +116,9 @@ class MixedPrecisionAdam:
             p.data = self.flat_param[off:off + n].view_as(p)
             p.main_grad = self.flat_grad[off:off + n].view(p.shape)
             off += n
-        self.master = self.flat_param.float()
+        if self.zero:
+            self.flat_param[total:].zero_()
+        self.master = self.flat_param[self.shard].float()
         self.m = torch.zeros_like(self.master)
         self.v = torch.zeros_like(self.master)
         self.t = 0
@@ -156,7 +174,7 @@ class MixedPrecisionAdam:
         allocate 2x the fp32 state = +64 GiB on an 8B model)."""
         self.t += 1
         b1, b2 = self.cfg.adam_betas
-        g = self.flat_grad
+        g = self.flat_grad[self.shard] if self.zero else self.flat_grad
         # global grad-norm clip (Megatron clip_grad; global across the
         # model-parallel group when configured)
         norm = self._global_grad_norm()
@@ -179,7 +197,21 @@ class MixedPrecisionAdam:
             sl = slice(off, min(off + CHUNK, self.v.numel()))
             denom = self.v[sl].sqrt().add_(eps2)
             self.master[sl].addcdiv_(self.m[sl], denom, value=step_size)
-        self.flat_param.copy_(self.master)
+        if self.zero:
+            self.flat_param[self.shard].copy_(self.master)
+            if dist.get_backend(self.zero_group) == "nccl":
+                dist.all_gather_into_tensor(
+                    self.flat_param, self.flat_param[self.shard].contiguous(),
+                    group=self.zero_group)
+            else:
+                shards = [torch.empty_like(self.flat_param[self.shard])
+                          for _ in range(self.zero_world)]
+                dist.all_gather(shards,
+                                self.flat_param[self.shard].contiguous(),
+                                group=self.zero_group)
+                self.flat_param.copy_(torch.cat(shards))
+        else:
+            self.flat_param.copy_(self.master)
 
 
 class DataParallelGradReducer:
@@ -321,6 +353,9 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                   tp_size: int = 1):
     torch.manual_seed(1234)
     assert not (cfg.ep_size > 1 and cfg.tp_size > 1), "trainer: tp XOR ep"
+    if cfg.zero_state == 1:
+        assert cfg.tp_size == 1 and cfg.ep_size == 1 and cfg.pp_size == 1, \
+            "ZeRO-1 composes with pure DP in the trainer for now"
     tp_size = max(tp_size, cfg.tp_size)
     # Megatron-style vocab padding (keeps CE vocab a GPU-friendly multiple
     # and makes GEMM shape keys match the calibration tables)

@@ -69,3 +69,61 @@ def test_dp2_gloo_train_step_param_sync():
     # losses finite
     for losses, _ in results.values():
         assert all(l == l for l in losses)
+
+
+def _worker_zero(rank, world, port, q):
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from simumax_amd.core.config import ModelConfig
+        from simumax_amd.train.trainer import (TrainConfig, build_trainer,
+                                               make_synthetic_batch,
+                                               train_step)
+
+        cfg = ModelConfig(hidden_size=128, head_num=2, kv_head_num=1,
+                          head_size=64, intermediate_size=256, layer_num=2,
+                          vocab_size=512, use_swiglu=True)
+        results = {}
+        for zero in (0, 1):
+            tc = TrainConfig(seq_len=32, micro_batch_size=2,
+                             micro_batch_num=2, overlap_grad_reduce=False,
+                             zero_state=zero, grad_clip=1e9)
+            model, opt, red = build_trainer(cfg, tc, "cpu")
+            if zero == 1:
+                # state really is sharded
+                assert opt.master.numel() * 2 >= opt.flat_param.numel()
+                assert opt.master.numel() < opt.flat_param.numel()
+            toks, labels = make_synthetic_batch(cfg.vocab_size, 2, 2, 32,
+                                                "cpu", seed=rank * 7 + 1)
+            for _ in range(2):
+                train_step(model, opt, red, toks, labels, 2)
+            results[zero] = opt.flat_param.detach().float().clone()
+            red.remove_hooks()
+        total = min(r.numel() for r in results.values())
+        diff = (results[0][:total] - results[1][:total]).abs().max()
+        q.put((rank, float(diff)))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_zero1_matches_zero0_params():
+    """ZeRO-1 (sharded optimizer state + param all_gather) must produce
+    the same parameters as the replicated optimizer after training."""
+    mp.set_start_method("spawn", force=True)
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_zero, args=(r, 2, 29535, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for _ in range(2):
+        rank, diff = q.get(timeout=240)
+        assert diff < 1e-6, f"rank {rank}: param divergence {diff}"
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
