@@ -35,19 +35,12 @@ class GQAAttention(nn.Module):
         self.tp_group = tp_group
         self.sp = sp
         qkv_out = (self.heads + 2 * self.kv_heads) * cfg.head_size
+        self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype, device=device)
+        self.out_proj = K.FusedLinear(self.heads * cfg.head_size, h,
+                                      dtype=dtype, device=device)
         if tp_size > 1:
-            from .tp import copy_to_tp, reduce_from_tp  # noqa: F401
-
-            self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype,
-                                          device=device)
             self.qkv_proj.weight._is_tp_shard = True
-            self.out_proj = K.FusedLinear(self.heads * cfg.head_size, h,
-                                          dtype=dtype, device=device)
             self.out_proj.weight._is_tp_shard = True
-        else:
-            self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype, device=device)
-            self.out_proj = K.FusedLinear(self.heads * cfg.head_size, h,
-                                          dtype=dtype, device=device)
 
     def forward(self, y, rope_cs, pos):
         from .tp import copy_to_tp, gather_seq, reduce_from_tp, scatter_seq
