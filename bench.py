@@ -168,7 +168,9 @@ def main():
         pred_peak = mem["max_peak_mem"]
         time_err = (pred_ms - ms_per_step) / ms_per_step * 100.0
         mem_err = (pred_peak - peak_bytes) / peak_bytes * 100.0
-        value = abs(time_err)
+        # the metric is MFU & peak-mem error: the headline is the worse of
+        # the two halves, not timing alone
+        value = max(abs(time_err), abs(mem_err))
 
         out = {
             "metric": "predicted-vs-measured MFU & peak-mem error (%), "

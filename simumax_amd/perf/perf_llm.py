@@ -42,14 +42,17 @@ GiB = 1024**3
 # straggler model (reference parity: perf_llm.py:255-291)
 # --------------------------------------------------------------------------
 def get_effective_straggler_sample_count(strategy: StrategyConfig, num_per_node: int) -> int:
-    nodes = max(1, strategy.world_size // num_per_node)
+    nodes = max(1, math.ceil(strategy.world_size / num_per_node))
     return min(nodes, strategy.dp_size, max(1, strategy.edp_size))
 
 
 def estimate_straggler_increase_ratio(n: int) -> float:
+    """1 + log2(n)/(log2(n)+1) * 0.09 * sqrt(log2(n)) — log-damped
+    machine-level slowdown (reference perf_llm.py:255-291)."""
     if n <= 1:
         return 1.0
-    return 1.0 + n / (n + 1) * 0.09 * math.sqrt(math.log2(n))
+    ns = math.log2(n)
+    return 1.0 + ns / (ns + 1) * 0.09 * math.sqrt(ns)
 
 
 # --------------------------------------------------------------------------
@@ -182,15 +185,23 @@ class PerfBase:
             cur = getattr(s, attr)
             if cur != "auto" and not re_analysis:
                 continue
-            if sp <= 1:
-                tier = "high_intra_node"
+            if attr == "pp_net":
+                # adjacent-stage p2p: consecutive stages are co-located in a
+                # node when the stage stride (world/pp ranks) is smaller than
+                # the node, even if the full pp group spans nodes
+                # (ref analysis_high_link_net: world_size//pp_size < num_per_node)
+                stride = s.world_size // pp if pp > 1 else 1
+                tier = "high_intra_node" if stride < n else (
+                    "high_intra_node" if sp <= n else "inter_node")
             elif sp <= n:
                 tier = "high_intra_node"
             else:
                 tier = "inter_node"
             if tier == "high_intra_node" and has_low and attr in ("dp_net", "edp_net"):
-                # dense-DP traffic can ride the lower tier when configured
-                tier = "high_intra_node"
+                # dense-DP/edp traffic rides the lower intra-node tier when
+                # the system config defines one (ref: intra_with_pcie tiers);
+                # single-fabric xGMI configs simply omit low_intra_node
+                tier = "low_intra_node"
             setattr(s, attr, tier)
         return {k: getattr(s, k) for k in choice}
 
@@ -410,9 +421,11 @@ class PerfLLM(PerfBase):
                     t += self.system.compute_net_op_time(
                         "reduce_scatter", per, group, net=net,
                         comm_stage=stage_name, strategy=s)
+                    # gathered params are bf16 (2 B); the bucket bytes were
+                    # built with grad_e B/elem, so convert with the SAME
+                    # grad_e regardless of use_fp32_accum_grad
                     t += self.system.compute_net_op_time(
-                        "all_gather",
-                        per / (4 if s.use_fp32_accum_grad and not s.grad_reduce_in_bf16 else 2) * 2,
+                        "all_gather", per / grad_e * 2,
                         group, net=net, comm_stage=stage_name, strategy=s)
                 else:
                     t += self.system.compute_net_op_time(

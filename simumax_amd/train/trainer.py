@@ -146,15 +146,21 @@ class MixedPrecisionAdam:
             off += n
 
     def _global_grad_norm(self):
+        # torch.dot computes the squared norm with NO temporary — g.pow(2)
+        # would materialize a full fp32 copy of the flat grad (~30 GiB on an
+        # 8B model) and move the step's peak-allocated point into the
+        # optimizer, which is exactly the fresh-box -6.7% memory anomaly in
+        # BENCH_r01.json.
         g = self.flat_grad
-        total_sq = g.pow(2).sum()
+        total_sq = torch.dot(g, g)
         if self._norm_group is None and getattr(self, "_pp_norm_group",
                                                 None) is None:
             return total_sq.sqrt()
         if self._norm_group is not None:
             rep_sq = g.new_zeros(())
             for lo, hi in self._replicated_slices:
-                rep_sq += g[lo:hi].pow(2).sum()
+                sl = g[lo:hi]
+                rep_sq += torch.dot(sl, sl)
             uni_sq = total_sq - rep_sq
             dist.all_reduce(uni_sq, group=self._norm_group)
             stage_sq = uni_sq + rep_sq
@@ -294,7 +300,9 @@ class DataParallelGradReducer:
     def _make_hook(self, span, remaining, is_expert=False):
         def hook(p):
             if not getattr(p, "_fused_wgrad", False) or not p.is_cuda:
-                p.main_grad.add_(p.grad.float())
+                # add_ casts bf16->fp32 inside the kernel; .float() here
+                # would allocate a param-sized fp32 temporary per grad
+                p.main_grad.add_(p.grad)
             p.grad = None
             remaining.discard(id(p))
             if not remaining:
@@ -345,7 +353,7 @@ def accumulate_main_grads(params):
     for p in params:
         if p.grad is not None:
             if not getattr(p, "_fused_wgrad", False) or not p.is_cuda:
-                p.main_grad.add_(p.grad.float())
+                p.main_grad.add_(p.grad)
             p.grad = None
 
 

@@ -90,7 +90,8 @@ def test_activation_replay_consistency():
 def test_straggler_model():
     assert estimate_straggler_increase_ratio(1) == 1.0
     r4 = estimate_straggler_increase_ratio(4)
-    assert r4 == pytest.approx(1 + 4 / 5 * 0.09 * (2 ** 0.5))
+    # log2(4)=2: 1 + 2/3 * 0.09 * sqrt(2)
+    assert r4 == pytest.approx(1 + 2 / 3 * 0.09 * (2 ** 0.5))
 
 
 def test_moe_deepseek():
