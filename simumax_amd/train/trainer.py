@@ -145,6 +145,17 @@ class MixedPrecisionAdam:
                 self._replicated_slices.append((off, off + n))
             off += n
 
+    @staticmethod
+    def _sq_sum(t):
+        """Temp-free squared sum; chunked because rocBLAS dot is
+        int32-indexed (an 8B-param flat grad exceeds 2^31 elements)."""
+        total = t.new_zeros(())
+        CH = 1 << 30
+        for off in range(0, t.numel(), CH):
+            sl = t[off:off + CH]
+            total += torch.dot(sl, sl)
+        return total
+
     def _global_grad_norm(self):
         # torch.dot computes the squared norm with NO temporary — g.pow(2)
         # would materialize a full fp32 copy of the flat grad (~30 GiB on an
@@ -152,15 +163,14 @@ class MixedPrecisionAdam:
         # optimizer, which is exactly the fresh-box -6.7% memory anomaly in
         # BENCH_r01.json.
         g = self.flat_grad
-        total_sq = torch.dot(g, g)
+        total_sq = self._sq_sum(g)
         if self._norm_group is None and getattr(self, "_pp_norm_group",
                                                 None) is None:
             return total_sq.sqrt()
         if self._norm_group is not None:
             rep_sq = g.new_zeros(())
             for lo, hi in self._replicated_slices:
-                sl = g[lo:hi]
-                rep_sq += torch.dot(sl, sl)
+                rep_sq += self._sq_sum(g[lo:hi])
             uni_sq = total_sq - rep_sq
             dist.all_reduce(uni_sq, group=self._norm_group)
             stage_sq = uni_sq + rep_sq
