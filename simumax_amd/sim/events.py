@@ -39,6 +39,8 @@ class Job:
     mb: int = -1                   # microbatch index
     mem: Optional[MemDelta] = None
     call_stack: str = ""
+    overlap: bool = False          # comm overlapped with compute: occupies
+                                   # only the comm lane (CommEvent.overlap)
 
 
 @dataclass
@@ -53,6 +55,7 @@ class LogEvent:
     call_stack: str = ""
     wait_start: float = 0.0   # time the job became head-of-lane
     mem: object = None        # the job's MemDelta (memory replay)
+    gid: Optional[str] = None  # rendezvous id (trace flow arrows)
 
 
 class DeadlockError(RuntimeError):
@@ -71,7 +74,10 @@ class SimuSystem:
     def run(self) -> float:
         ranks = sorted(self.jobs)
         ptr = {r: 0 for r in ranks}
-        lane_t = {r: {"comp": 0.0, "comm": 0.0} for r in ranks}
+        # comp = compute stream, comm = serialized sync-collective stream,
+        # async = overlapped-collective stream (own HIP stream/communicator,
+        # does not contend with sync comm — CommEvent.overlap)
+        lane_t = {r: {"comp": 0.0, "comm": 0.0, "async": 0.0} for r in ranks}
         # rendezvous state: gid -> {rank: arrival_time}
         arrivals: Dict[str, Dict[int, float]] = {}
         done: Dict[str, float] = {}  # gid -> completion time
@@ -88,11 +94,15 @@ class SimuSystem:
                         end = start + job.dur
                         lane_t[r][job.lane] = end
                         if job.lane == "comm":
-                            # sync collective: blocks compute too
-                            start = max(lane_t[r]["comp"], start)
+                            lane = "async" if job.overlap else "comm"
+                            # issued when compute reaches this point; each
+                            # stream serializes its own transfers
+                            start = max(lane_t[r]["comp"], lane_t[r][lane])
                             end = start + job.dur
-                            lane_t[r]["comm"] = end
-                            lane_t[r]["comp"] = end
+                            lane_t[r][lane] = end
+                            if not job.overlap:
+                                # sync collective: blocks compute too
+                                lane_t[r]["comp"] = end
                         self.log.append(LogEvent(r, job.name, job.kind, job.lane,
                                                  start, end, job.mb, job.call_stack,
                                                  start, job.mem))
@@ -108,7 +118,7 @@ class SimuSystem:
                         done.setdefault(job.gid + "#rposted", t)
                         self.log.append(LogEvent(r, job.name, "p2p", "comm",
                                                  t, t, job.mb, job.call_stack,
-                                                 t, None))
+                                                 t, None, gid=job.gid))
                         ptr[r] += 1
                         executed += 1
                         progressed = True
@@ -132,7 +142,7 @@ class SimuSystem:
                         self.log.append(LogEvent(r, job.name, "p2p", "comm",
                                                  start, end, job.mb,
                                                  job.call_stack, arrive,
-                                                 job.mem))
+                                                 job.mem, gid=job.gid))
                         ptr[r] += 1
                         executed += 1
                         progressed = True
@@ -149,7 +159,8 @@ class SimuSystem:
                         done.setdefault(job.gid + "#posted", post)
                         self.log.append(LogEvent(r, job.name, "p2p", job.lane,
                                                  post, post, job.mb,
-                                                 job.call_stack, post, job.mem))
+                                                 job.call_stack, post, job.mem,
+                                                 gid=job.gid))
                         ptr[r] += 1
                         executed += 1
                         progressed = True
@@ -166,7 +177,8 @@ class SimuSystem:
                         lane_t[r]["comp"] = max(lane_t[r]["comp"], end)
                         self.log.append(LogEvent(r, job.name, "p2p", job.lane,
                                                  start, end, job.mb,
-                                                 job.call_stack, arrive, job.mem))
+                                                 job.call_stack, arrive, job.mem,
+                                                 gid=job.gid))
                         ptr[r] += 1
                         executed += 1
                         progressed = True
@@ -176,10 +188,13 @@ class SimuSystem:
                     sim_peers = tuple(p for p in peers if p in self.jobs)
                     if len(sim_peers) <= 1:
                         # peers not simulated (lane-merged): local cost
-                        start = max(lane_t[r][job.lane], lane_t[r]["comp"])
+                        lane = ("async" if job.overlap and job.lane == "comm"
+                                else job.lane)
+                        start = max(lane_t[r][lane], lane_t[r]["comp"])
                         end = start + job.dur
-                        lane_t[r][job.lane] = end
-                        lane_t[r]["comp"] = max(lane_t[r]["comp"], end)
+                        lane_t[r][lane] = end
+                        if not (job.overlap and job.lane == "comm"):
+                            lane_t[r]["comp"] = max(lane_t[r]["comp"], end)
                         self.log.append(LogEvent(r, job.name, job.kind, job.lane,
                                                  start, end, job.mb, job.call_stack,
                                                  start, job.mem))
@@ -192,9 +207,12 @@ class SimuSystem:
                         arrive = arrivals[job.gid][r]
                         self.log.append(LogEvent(r, job.name, job.kind, job.lane,
                                                  end - job.dur, end, job.mb,
-                                                 job.call_stack, arrive, job.mem))
-                        lane_t[r][job.lane] = max(lane_t[r][job.lane], end)
-                        if job.lane == "comm":
+                                                 job.call_stack, arrive, job.mem,
+                                                 gid=job.gid))
+                        lane = ("async" if job.overlap and job.lane == "comm"
+                                else job.lane)
+                        lane_t[r][lane] = max(lane_t[r][lane], end)
+                        if job.lane == "comm" and not job.overlap:
                             # sync comm also blocks the compute lane
                             lane_t[r]["comp"] = max(lane_t[r]["comp"], end)
                         ptr[r] += 1
@@ -204,8 +222,14 @@ class SimuSystem:
                     # arrive and block
                     a = arrivals.setdefault(job.gid, {})
                     if r not in a:
-                        # sync p2p: the rank arrives once BOTH lanes reach it
-                        a[r] = max(lane_t[r].values())
+                        if job.overlap and job.lane == "comm":
+                            # issued at the compute front; queues on the
+                            # overlap stream only
+                            a[r] = max(lane_t[r]["comp"], lane_t[r]["async"])
+                        else:
+                            # sync: the rank arrives once its sync lanes
+                            # reach this job
+                            a[r] = max(lane_t[r]["comp"], lane_t[r]["comm"])
                         progressed = True
                     if all(p in a for p in sim_peers):
                         done[job.gid] = max(a.values()) + job.dur

@@ -40,7 +40,7 @@ def _leaf_jobs_fwd(chunk, stage, mb, p2p=None):
                     name=f"{leaf.full_name}.{ev.op_name}", kind="comm",
                     dur=ev.time_ms, lane="comm", mb=mb,
                     gid=f"mb{mb}-{leaf.full_name}-{ev.op_name}-{ev.comm_stage}",
-                    call_stack=leaf.full_name,
+                    call_stack=leaf.full_name, overlap=ev.overlap,
                 ))
     return jobs
 
@@ -94,7 +94,7 @@ def _leaf_jobs_bwd(chunk, stage, mb):
                     name=f"{leaf.full_name}.{ev.op_name}(bwd)", kind="comm",
                     dur=ev.time_ms, lane="comm", mb=mb,
                     gid=f"mb{mb}-{leaf.full_name}-{ev.op_name}-{ev.stage}",
-                    call_stack=leaf.full_name,
+                    call_stack=leaf.full_name, overlap=ev.overlap,
                 ))
     return jobs
 

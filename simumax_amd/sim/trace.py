@@ -38,7 +38,6 @@ def events_to_chrome_trace(log, save_path: str):
         for i, lane in enumerate(_LANE_ORDER):
             out.append(dict(ph="M", name="thread_sort_index", pid=f"rank{r}",
                             tid=lane, args={"sort_index": i}))
-    flow_id = 0
     for e in log:
         tid = _tid(e.kind, e.lane)
         cat = "comm" if e.lane == "comm" else "compute"
@@ -54,7 +53,31 @@ def events_to_chrome_trace(log, save_path: str):
             args={"call_stack": e.call_stack or e.name,
                   "stream_type": "comm" if e.lane == "comm" else "compute",
                   "microbatch": e.mb}))
+    # flow arrows sender-post -> receiver-transfer for every p2p rendezvous
+    # (reference parity: generate_tracing.py flow events). The sender's
+    # zero-duration post is the arrow tail; the receiver's transfer slice
+    # (positive duration, same gid, other rank) is the head.
+    flows = {}
+    for e in log:
+        if e.kind != "p2p" or not getattr(e, "gid", None):
+            continue
+        slot = flows.setdefault(e.gid, {})
+        if "send" in e.name:            # sender post (zero duration)
+            slot.setdefault("send", e)
+        elif e.end > e.start:           # receiver transfer slice
+            slot.setdefault("recv", e)
+    flow_id = 0
+    for gid, slot in flows.items():
+        snd, rcv = slot.get("send"), slot.get("recv")
+        if snd is None or rcv is None or snd.rank == rcv.rank:
+            continue
         flow_id += 1
+        out.append(dict(ph="s", id=flow_id, name="p2p", cat="flow",
+                        pid=f"rank{snd.rank}", tid=_tid("p2p", "comm"),
+                        ts=snd.start * 1e3))
+        out.append(dict(ph="f", id=flow_id, bp="e", name="p2p", cat="flow",
+                        pid=f"rank{rcv.rank}", tid=_tid("p2p", "comm"),
+                        ts=rcv.end * 1e3))
     payload = {"traceEvents": out, "displayTimeUnit": "ms"}
     with open(save_path, "w") as f:
         json.dump(payload, f)

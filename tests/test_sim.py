@@ -179,3 +179,62 @@ def test_assemble_units_async_ordering():
     kinds = [j.kind for j in a]
     assert kinds == ["p2p_post_recv", "p2p_post_recv", "p2p_wait", "fwd",
                      "p2p_wait", "fwd"]
+
+
+def test_overlap_comm_does_not_block_compute():
+    """A CommEvent.overlap collective occupies only the comm lane
+    (simulator honors overlap)."""
+    jobs = {
+        0: [
+            Job(name="c1", kind="fwd", dur=10.0),
+            Job(name="ag_overlap", kind="comm", dur=8.0, lane="comm",
+                overlap=True),
+            Job(name="c2", kind="fwd", dur=10.0),
+        ],
+    }
+    sim = SimuSystem(jobs)
+    total = sim.run()
+    ev = {e.name: e for e in sim.log}
+    # c2 starts right after c1; the overlapped all_gather runs beside it
+    assert ev["c2"].start == pytest.approx(10.0)
+    assert ev["ag_overlap"].start == pytest.approx(10.0)
+    assert total == pytest.approx(20.0)
+    # a sync version blocks: same stream without overlap
+    jobs_sync = {
+        0: [
+            Job(name="c1", kind="fwd", dur=10.0),
+            Job(name="ag_sync", kind="comm", dur=8.0, lane="comm"),
+            Job(name="c2", kind="fwd", dur=10.0),
+        ],
+    }
+    sim2 = SimuSystem(jobs_sync)
+    assert sim2.run() == pytest.approx(28.0)
+
+
+def test_sim_vs_perf_agreement_overlap_heavy(tmp_path):
+    """tp2 + sequence-parallel with overlap_grad_reduce: the SP wgrad
+    all_gathers are overlapped; simulate() must still match analysis_cost
+    (VERDICT r1 done-criterion: within ~1%)."""
+    p = build(strategy="tp2_pp1_dp4_mbs1", enable_sequence_parallel=True,
+              overlap_grad_reduce=True)
+    cost = p.analysis_cost()
+    res = p.simulate(str(tmp_path))
+    analytic = (cost["pipeline_time"] / cost["straggler_ratio"]
+                + cost["dp_time"] + cost["optim_time"])
+    assert res["total_time"] == pytest.approx(analytic, rel=0.01)
+
+
+def test_trace_flow_arrows(tmp_path):
+    p = build()  # pp2
+    p.simulate(str(tmp_path))
+    tr = json.loads((tmp_path / "tracing_logs.json").read_text())
+    flows = [e for e in tr["traceEvents"] if e.get("ph") in ("s", "f")]
+    assert flows, "p2p flow arrows missing from trace"
+    starts = [e for e in flows if e["ph"] == "s"]
+    ends = [e for e in flows if e["ph"] == "f"]
+    assert len(starts) == len(ends)
+    # arrows cross ranks
+    by_id = {}
+    for e in flows:
+        by_id.setdefault(e["id"], []).append(e["pid"])
+    assert all(len(set(v)) == 2 for v in by_id.values())
