@@ -106,6 +106,10 @@ class LLMBlock(MetaModule):
         if full_block:
             self.set_recompute(True)
             return
+        if self.strategy.recompute_granularity == "full_block":
+            # layers beyond recompute_layer_num are NOT recomputed at all
+            # (full_block is a per-layer, not per-submodule, choice)
+            return
         cfg_attn = self.strategy.parse_attention_recompute()
         cfg_mlp = self.strategy.parse_mlp_recompute()
         if cfg_attn.recompute_norm:

@@ -674,11 +674,163 @@ class PerfLLM(PerfBase):
             self.run_estimate()
         return best
 
+    # ---- recompute-aware search family (ref perf_llm.py:3213-3355) -------
+    def _stage_peak_gbytes(self, mem: Result) -> float:
+        return max(st["peak_mem"] for st in mem["stages_raw"]) / 1024**3
+
+    def _search_record(self, cost: Result, mem: Result) -> Result:
+        s = self.strategy
+        return Result(
+            parallelism=s.parallelism,
+            tp=s.tp_size, pp=s.pp_size, ep=s.ep_size, dp=s.dp_size,
+            micro_batch_size=s.micro_batch_size,
+            micro_batch_num=s.micro_batch_num,
+            recompute_granularity=s.recompute_granularity,
+            recompute_layer_num=s.recompute_layer_num,
+            attn_recompute=s.attn_recompute,
+            mla_rms_recompute=s.mla_rms_recompute,
+            mlp_recompute=s.mlp_recompute,
+            mlp_rms_recompute=s.mlp_rms_recompute,
+            mfu=cost["mfu"], iter_time=cost["iter_time"],
+            peak_mem=mem["max_peak_mem"],
+        )
+
+    def search_best_strategy_no_recompute(self, gmi_error: float = 6.0,
+                                          best_mfu: float = 0.0,
+                                          all_search_result: Optional[list] = None,
+                                          save_path: Optional[str] = None) -> Result:
+        """Evaluate the current parallel strategy without recompute; returns
+        the record if it fits in memory minus the gmi_error margin (GiB
+        reserved for RCCL buffers / allocator overhead the model doesn't
+        price — ref perf_llm.py:3330-3355)."""
+        s = self.strategy
+        s.enable_recompute = False
+        s.recompute_granularity = None
+        s.recompute_layer_num = 0
+        budget = self.system.accelerator.mem_gbs - gmi_error
+        self.run_estimate()
+        mem = self.analysis_mem()
+        cost = self.analysis_cost()
+        if self._stage_peak_gbytes(mem) > budget:
+            return Result()
+        rec = self._search_record(cost, mem)
+        if all_search_result is not None:
+            all_search_result.append(rec)
+        if cost["mfu"] > best_mfu and save_path is not None:
+            self.analysis(save_path)
+        return rec if cost["mfu"] > best_mfu else Result()
+
+    def search_best_recompute_layer_num(self, layer_num: Optional[int] = None,
+                                        use_reserved_memory: bool = True,
+                                        gmi_error: float = 6.0,
+                                        best_mfu: float = 0.0,
+                                        all_search_result: Optional[list] = None,
+                                        save_path: Optional[str] = None) -> Result:
+        """Binary-search the smallest per-stage full-recompute layer count
+        that fits memory (fewer recomputed layers = higher MFU, so the
+        smallest feasible count is MFU-optimal); tracks the best MFU seen
+        (ref perf_llm.py:3270-3330)."""
+        del use_reserved_memory  # mem_factor already covers reserved margin
+        s = self.strategy
+        layer_num = layer_num or self.model_config.layer_num
+        budget = self.system.accelerator.mem_gbs - gmi_error
+        ori = (s.enable_recompute, s.recompute_granularity, s.recompute_layer_num)
+        s.enable_recompute = True
+        s.recompute_granularity = "full_block"
+        best = Result()
+        left, right = 0, math.ceil(layer_num / s.pp_size)
+        while left <= right:
+            n = (left + right) // 2
+            # n == 0 probes the no-recompute point (the model treats
+            # recompute_layer_num 0 under full_block as "all layers", the
+            # config-parity default)
+            s.enable_recompute = n > 0
+            s.recompute_layer_num = n
+            self.run_estimate()
+            mem = self.analysis_mem()
+            cost = self.analysis_cost()
+            if self._stage_peak_gbytes(mem) > budget:
+                left = n + 1
+                continue
+            right = n - 1
+            rec = self._search_record(cost, mem)
+            if all_search_result is not None:
+                all_search_result.append(rec)
+            if cost["mfu"] >= best_mfu:
+                best_mfu = cost["mfu"]
+                best = rec
+                if save_path is not None:
+                    self.analysis(save_path)
+        (s.enable_recompute, s.recompute_granularity,
+         s.recompute_layer_num) = ori
+        return best
+
+    # the three selective combinations the reference's search walks
+    # (perf_llm.py:3227-3246): everything, attention-side only, mlp-side only
+    _SELECTIVE_COMBOS = (
+        dict(attn_recompute=True, mla_rms_recompute=True,
+             mlp_recompute=True, mlp_rms_recompute=True),
+        dict(attn_recompute=True, mla_rms_recompute=True,
+             mlp_recompute=False, mlp_rms_recompute=False),
+        dict(attn_recompute=False, mla_rms_recompute=False,
+             mlp_recompute=True, mlp_rms_recompute=True),
+    )
+
+    def search_best_selective_recompute(self, use_reserved_memory: bool = True,
+                                        gmi_error: float = 6.0,
+                                        best_mfu: float = 0.0,
+                                        all_search_result: Optional[list] = None,
+                                        save_path: Optional[str] = None) -> Result:
+        """Walk the curated selective-recompute combinations, keep the best
+        feasible MFU (ref perf_llm.py:3213-3267)."""
+        del use_reserved_memory
+        if self.strategy.megatron_recompute:
+            raise NotImplementedError(
+                "search does not support megatron_recompute; evaluate those "
+                "strategies explicitly")
+        s = self.strategy
+        budget = self.system.accelerator.mem_gbs - gmi_error
+        ori = (s.enable_recompute, s.recompute_granularity, s.attn_recompute,
+               s.mla_rms_recompute, s.mlp_recompute, s.mlp_rms_recompute)
+        s.enable_recompute = True
+        s.recompute_granularity = "selective_recompute"
+        s.recompute_layer_num = 0
+        best = Result()
+        for combo in self._SELECTIVE_COMBOS:
+            for k, v in combo.items():
+                setattr(s, k, v)
+            self.run_estimate()
+            mem = self.analysis_mem()
+            cost = self.analysis_cost()
+            if self._stage_peak_gbytes(mem) > budget:
+                continue
+            rec = self._search_record(cost, mem)
+            if all_search_result is not None:
+                all_search_result.append(rec)
+            if cost["mfu"] > best_mfu:
+                best_mfu = cost["mfu"]
+                best = rec
+                if save_path is not None:
+                    self.analysis(save_path)
+        (s.enable_recompute, s.recompute_granularity, s.attn_recompute,
+         s.mla_rms_recompute, s.mlp_recompute, s.mlp_rms_recompute) = ori
+        return best
+
     def search_best_parallel_strategy(self, world_size: int, global_batch_size: int,
                                       tp_candidates=(1, 2, 4, 8),
                                       pp_candidates=(1, 2, 4, 8),
                                       ep_candidates=(1,),
+                                      recompute_search_type=("no_recompute",),
+                                      gmi_error: float = 6.0,
+                                      probe_mbs: bool = False,
+                                      max_mbs: int = 32,
+                                      all_search_result: Optional[list] = None,
                                       verbose=False) -> Optional[Result]:
+        """Grid-search tp/pp/ep; per candidate optionally probe the largest
+        memory-feasible micro-batch size at fixed global batch, then search
+        the requested recompute families (ref perf_llm.py:3355-3578)."""
+        if isinstance(recompute_search_type, str):
+            recompute_search_type = (recompute_search_type,)
         base_strategy = deepcopy(self.strategy)
         model_cfg = deepcopy(self.model_config)
         system_cfg = deepcopy(self.system)
@@ -698,24 +850,42 @@ class PerfLLM(PerfBase):
                             continue
                         self.configure(st, model_cfg, system_cfg)
                         self.strategy.reset_global_batch_size(global_batch_size)
+                        if probe_mbs:
+                            # memory-guarded probe: the largest mbs whose
+                            # analysis_mem fits, at fixed global batch
+                            if self.search_max_micro_batch_size_fixed_gbs(
+                                    global_batch_size, max_mbs=max_mbs) is None:
+                                continue
                         self.run_estimate()
-                        mem = self.analysis_mem()
-                        if mem["oom"]:
-                            continue
-                        cost = self.analysis_cost()
                     except (AssertionError, ZeroDivisionError):
                         continue
-                    cand = Result(
-                        tp=tp, pp=pp, ep=ep, dp=st.dp_size,
-                        mfu=cost["mfu"], iter_time=cost["iter_time"],
-                        peak_mem=mem["max_peak_mem"],
-                        parallelism=self.strategy.parallelism,
-                    )
-                    if verbose:
-                        print(f"tp{tp} pp{pp} ep{ep}: MFU {cost['mfu']*100:.2f}% "
-                              f"iter {cost['iter_time']:.1f} ms")
-                    if best is None or cand["mfu"] > best["mfu"]:
-                        best = cand
+                    best_mfu = best["mfu"] if best else 0.0
+                    for kind in recompute_search_type:
+                        try:
+                            if kind == "no_recompute":
+                                rec = self.search_best_strategy_no_recompute(
+                                    gmi_error=gmi_error, best_mfu=best_mfu,
+                                    all_search_result=all_search_result)
+                            elif kind == "full_block":
+                                rec = self.search_best_recompute_layer_num(
+                                    gmi_error=gmi_error, best_mfu=best_mfu,
+                                    all_search_result=all_search_result)
+                            elif kind == "selective_recompute":
+                                rec = self.search_best_selective_recompute(
+                                    gmi_error=gmi_error, best_mfu=best_mfu,
+                                    all_search_result=all_search_result)
+                            else:
+                                raise ValueError(f"unknown recompute search "
+                                                 f"type: {kind}")
+                        except (AssertionError, ZeroDivisionError):
+                            continue
+                        if rec and (best is None or rec["mfu"] > best["mfu"]):
+                            best = rec
+                            best_mfu = rec["mfu"]
+                        if verbose and rec:
+                            print(f"tp{tp} pp{pp} ep{ep} {kind}: "
+                                  f"MFU {rec['mfu']*100:.2f}% "
+                                  f"iter {rec['iter_time']:.1f} ms")
         return best
 
 

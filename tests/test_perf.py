@@ -313,3 +313,50 @@ def test_strategy_search_moe_with_ep():
     best = res.best
     assert best["mfu"] > 0.1
     assert best["peak_mem"] < 288 * 1024**3
+
+
+def test_search_recompute_layer_num():
+    """70B on 288 GB OOMs without recompute at tp2/pp2 mbs1xmbc8; the
+    binary search finds the smallest feasible full-recompute layer count
+    (ref perf_llm.py:3270-3330)."""
+    p = build(model="llama3-70b", strategy="tp2_pp2_dp2_mbs1_selective",
+              enable_recompute=False, recompute_granularity=None,
+              micro_batch_num=8, micro_batch_size=2)
+    p.run_estimate()
+    assert p.analysis_mem()["oom"]  # needs recompute
+    rec = p.search_best_recompute_layer_num(gmi_error=6.0)
+    assert rec, "no feasible recompute layer count found"
+    assert 0 < rec["recompute_layer_num"] <= 40
+    assert rec["peak_mem"] <= (288 - 6.0) * 1024**3
+    # fewer recomputed layers than full recompute => better MFU than
+    # recomputing every layer
+    p.strategy.enable_recompute = True
+    p.strategy.recompute_granularity = "full_block"
+    p.strategy.recompute_layer_num = 40  # ceil(80/2): everything
+    p.run_estimate()
+    full = p.analysis_cost()
+    assert rec["mfu"] >= full["mfu"]
+
+
+def test_search_selective_recompute():
+    p = build(model="llama3-70b", strategy="tp2_pp2_dp2_mbs1_selective",
+              micro_batch_num=8)
+    results = []
+    rec = p.search_best_selective_recompute(all_search_result=results)
+    # at least one curated combo evaluated; feasible ones recorded
+    assert isinstance(results, list)
+    if rec:
+        assert rec["recompute_granularity"] == "selective_recompute"
+        assert rec["peak_mem"] <= (288 - 6.0) * 1024**3
+
+
+def test_search_best_parallel_strategy_with_recompute_probe():
+    p = build(model="llama3-70b-l12", strategy="tp1_pp1_dp8_mbs1")
+    best = p.search_best_parallel_strategy(
+        world_size=8, global_batch_size=8,
+        tp_candidates=(1, 2), pp_candidates=(1, 2), ep_candidates=(1,),
+        recompute_search_type=("no_recompute", "full_block"),
+        probe_mbs=True)
+    assert best is not None
+    assert best["mfu"] > 0
+    assert best["tp"] * best["pp"] * best["dp"] == 8
