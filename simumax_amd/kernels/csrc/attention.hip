@@ -27,9 +27,11 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define KVTILE 32
 #define DHEAD 128
 #define WAVES 4
-#define KS 136              // padded K row stride (elems)
-#define VTS 40              // padded V^T row stride
-#define PS 40               // padded P row stride
+#define KS 144              // padded K row stride (elems); stride%128==16
+// makes (col*stride + kgrp*8) distinct banks within every ds_read_b128
+// 16-lane group (a +8 pad leaves a 2-way conflict: stride dwords % 16 == 4)
+#define VTS 48              // padded V^T row stride
+#define PS 48               // padded P row stride
 
 DEV bf16x8v ld_frag(const bf16raw *p) {
     uint4 r = *reinterpret_cast<const uint4 *>(p);
@@ -78,23 +80,37 @@ DEV bf16x8v tr_frag(const bf16raw *sub_base, int lane) {
     return r;
 }
 
+// v3 forward geometry (guide-structure: 8-wave/512-thread workgroup,
+// 2 waves per SIMD, one workgroup per CU):
+//   * the workgroup owns a 256-row q block; wave w owns TWO 16-row
+//     subtiles, w and 15-w — with causal masking each wave then sees the
+//     same total kv work (load balance across the 2-wave SIMD pairing)
+//   * 64-key kv tiles (half the barriers of v2's 32), double-buffered with
+//     T14 split staging (issue loads early, LDS writes after compute)
+//   * static s_setprio(1) for the younger wave half (guide T5 static form)
+//   * V (and the second half of K's tr image) keep the v2 ds_read_b64_tr_b16
+//     image: two 32-key groups per 64-key tile
 // DQK: q/k head dim (128 dense/GQA, 192 MLA); v/o stay 128 (DHEAD).
-// NQS: 16-row q subtiles per wave (2 at DQK=128; 1 at 192 for registers).
-template <int DQK, int NQS>
-__global__ __launch_bounds__(FA_BLOCK, 2)
+#define FWD_BLOCK 512
+#define FWD_WAVES 8
+#define KVT2 64             // kv tile (two 32-key tr-image groups)
+#define PS2 80              // padded P row stride (64 cols + 16)
+
+template <int DQK>
+__global__ __launch_bounds__(FWD_BLOCK, 2)
 void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
                    const bf16raw *__restrict__ v, bf16raw *__restrict__ o,
                    float *__restrict__ lse, int B, int S, int Hq, int Hkv,
                    int causal) {
+    constexpr int NQS = 2;                // two causal-balanced subtiles
     constexpr int KC = DQK / 32;          // 32-deep k chunks of QK^T
-    constexpr int KS_T = DQK + 8;         // padded K row stride
-    constexpr int QT = NQS * 16;          // q rows per wave
-    constexpr int BUF_ELEMS = KVTILE * KS_T + 8 * VSUB;
+    constexpr int KS_T = DQK + 16;        // padded K row stride
+    constexpr int BUF_ELEMS = KVT2 * KS_T + 2 * 8 * VSUB;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     bf16raw *buf0 = reinterpret_cast<bf16raw *>(smem);
     bf16raw *P_all = buf0 + 2 * BUF_ELEMS;
 
-    const int qblk = blockIdx.x;            // 128-row q block
+    const int qblk = blockIdx.x;            // 256-row q block
     const int h = blockIdx.y;
     const int b = blockIdx.z;
     const int hkv = h / (Hq / Hkv);
@@ -103,7 +119,12 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     const int col = lane & 15;
     const int kgrp = lane >> 4;
 
-    const int qbase = qblk * (WAVES * QT) + wave * QT;
+    // wave w owns subtiles w (low rows) and 15-w (high rows)
+    const int sub[NQS] = {wave, 2 * FWD_WAVES - 1 - wave};
+    const int blk_rows = FWD_WAVES * 2 * 16;   // 256
+    int qsb[NQS];
+#pragma unroll
+    for (int qs = 0; qs < NQS; ++qs) qsb[qs] = qblk * blk_rows + sub[qs] * 16;
     const float scale = rsqrtf((float)DQK);
 
     const long q_row = (long)Hq * DQK;        // q layout [B,S,Hq,DQK]
@@ -113,12 +134,12 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     const bf16raw *qp = q + ((long)b * S) * q_row + (long)h * DQK;
     const bf16raw *kp = k + ((long)b * S) * k_row + (long)hkv * DQK;
     const bf16raw *vp = v + ((long)b * S) * v_row + (long)hkv * DHEAD;
-    bf16raw *Pw = P_all + wave * QT * PS;
+    bf16raw *Pw = P_all + wave * 2 * 16 * PS2;
 
     bf16x8v a_q[NQS][KC];
 #pragma unroll
     for (int qs = 0; qs < NQS; ++qs) {
-        const int qrow = qbase + qs * 16 + col;
+        const int qrow = qsb[qs] + col;
 #pragma unroll
         for (int kc = 0; kc < KC; ++kc) {
             bf16x8 raw = load8(qp + (long)min(qrow, S - 1) * q_row + kc * 32 + kgrp * 8);
@@ -143,28 +164,27 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
 #pragma unroll
         for (int dt = 0; dt < 8; ++dt) acc[qs][dt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    const int blk_rows = WAVES * QT;
     const int kv_limit = causal ? min(S, qblk * blk_rows + blk_rows) : S;
-    const int n_tiles = CDIV(kv_limit, KVTILE);
+    const int n_tiles = CDIV(kv_limit, KVT2);
 
-    // staging: T14 split — issue global loads into registers early, write
-    // to LDS after compute. Per thread: 2 chunks of (K 8 elems + V 8 elems).
-    constexpr int KCHUNKS = KVTILE * DQK / (FA_BLOCK * 8);   // per-thread K pieces
-    bf16x8 st_k[KCHUNKS], st_v[2];
+    // T14 split staging over the 512-thread workgroup
+    constexpr int KCHUNKS = KVT2 * DQK / (FWD_BLOCK * 8);   // per-thread K pieces
+    constexpr int VCHUNKS = KVT2 * DHEAD / (FWD_BLOCK * 8);
+    bf16x8 st_k[KCHUNKS], st_v[VCHUNKS];
     auto stage_load = [&](int t) {
         if (t >= n_tiles) return;
-        const int kv = t * KVTILE;
+        const int kv = t * KVT2;
         const int tid = threadIdx.x;
 #pragma unroll
         for (int c = 0; c < KCHUNKS; ++c) {
-            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int e = tid * 8 + c * FWD_BLOCK * 8;
             const int kvr = e / DQK, d0 = e % DQK;
             const int src = min(kv + kvr, S - 1);
             st_k[c] = load8(kp + (long)src * k_row + d0);
         }
 #pragma unroll
-        for (int c = 0; c < 2; ++c) {
-            const int e = tid * 8 + c * FA_BLOCK * 8;
+        for (int c = 0; c < VCHUNKS; ++c) {
+            const int e = tid * 8 + c * FWD_BLOCK * 8;
             const int kvr = e / DHEAD, d0 = e % DHEAD;
             const int src = min(kv + kvr, S - 1);
             st_v[c] = load8(vp + (long)src * v_row + d0);
@@ -173,20 +193,21 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     auto stage_write = [&](int t) {
         if (t >= n_tiles) return;
         bf16raw *K_lds = buf0 + (t & 1) * BUF_ELEMS;
-        bf16raw *V_img = K_lds + KVTILE * KS_T;
+        bf16raw *V_img = K_lds + KVT2 * KS_T;
         const int tid = threadIdx.x;
 #pragma unroll
         for (int c = 0; c < KCHUNKS; ++c) {
-            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int e = tid * 8 + c * FWD_BLOCK * 8;
             const int kvr = e / DQK, d0 = e % DQK;
             store8(K_lds + kvr * KS_T + d0, st_k[c]);
         }
 #pragma unroll
-        for (int c = 0; c < 2; ++c) {
-            const int e = tid * 8 + c * FA_BLOCK * 8;
+        for (int c = 0; c < VCHUNKS; ++c) {
+            const int e = tid * 8 + c * FWD_BLOCK * 8;
             const int kvr = e / DHEAD, d0 = e % DHEAD;
-            // V image: subtile d0/16, row perm(kvr), col d0%16
-            store8(V_img + (d0 >> 4) * VSUB + v_img_row(kvr) * 16 + (d0 & 15),
+            // 32-key group (kvr>>5), subtile d0/16, row perm(kvr&31)
+            store8(V_img + (kvr >> 5) * 8 * VSUB + (d0 >> 4) * VSUB
+                       + v_img_row(kvr & 31) * 16 + (d0 & 15),
                    st_v[c]);
         }
     };
@@ -195,81 +216,82 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     stage_write(0);
     __syncthreads();
     stage_load(1);
+    // static priority for the younger wave half (T5 static form): the
+    // condition must be provably wave-uniform or s_setprio goes under exec
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= FWD_BLOCK / 2)
+        __builtin_amdgcn_s_setprio(1);
 
     for (int t = 0; t < n_tiles; ++t) {
-        const int kv = t * KVTILE;
+        const int kv = t * KVT2;
         bf16raw *K_lds = buf0 + (t & 1) * BUF_ELEMS;
-        bf16raw *V_img = K_lds + KVTILE * KS_T;
+        bf16raw *V_img = K_lds + KVT2 * KS_T;
 
-        if (!(causal && kv >= qbase + QT)) {
-            // QK^T for ALL q-subtiles per K fragment: each B-fragment is
-            // read once and feeds NQS independent MFMA chains
-            f32x4 sq[NQS][2];
 #pragma unroll
-            for (int qs = 0; qs < NQS; ++qs) {
-                sq[qs][0] = f32x4{0, 0, 0, 0};
-                sq[qs][1] = f32x4{0, 0, 0, 0};
-            }
+        for (int qs = 0; qs < NQS; ++qs) {
+            if (causal && kv >= qsb[qs] + 16) continue;
+            // QK^T: 4 16-col K subtiles x KC k-chunks
+            f32x4 sq[4];
+#pragma unroll
+            for (int ks = 0; ks < 4; ++ks) sq[ks] = f32x4{0, 0, 0, 0};
 #pragma unroll
             for (int kc = 0; kc < KC; ++kc) {
-                bf16x8v b0 = ld_frag(K_lds + col * KS_T + kc * 32 + kgrp * 8);
-                bf16x8v b1 = ld_frag(K_lds + (16 + col) * KS_T + kc * 32 + kgrp * 8);
 #pragma unroll
-                for (int qs = 0; qs < NQS; ++qs) {
-                    sq[qs][0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], b0, sq[qs][0], 0, 0, 0);
-                    sq[qs][1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], b1, sq[qs][1], 0, 0, 0);
+                for (int ks = 0; ks < 4; ++ks) {
+                    bf16x8v bk = ld_frag(K_lds + (ks * 16 + col) * KS_T + kc * 32 + kgrp * 8);
+                    sq[ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], bk, sq[ks], 0, 0, 0);
                 }
             }
-            // softmax per subtile, then P->LDS
-            bf16x8v a_p[NQS];
+            // online softmax over the 64 kv cols
+            float tile_max[4];
 #pragma unroll
-            for (int qs = 0; qs < NQS; ++qs) {
-                f32x4 s0 = sq[qs][0], s1 = sq[qs][1];
-                float tile_max[4];
+            for (int j = 0; j < 4; ++j) {
+                const int row = qsb[qs] + kgrp * 4 + j;
 #pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    const int row = qbase + qs * 16 + kgrp * 4 + j;
-                    const int c0 = kv + col, c1 = kv + 16 + col;
-                    if (row >= S || c0 >= S || (causal && c0 > row)) s0[j] = -INFINITY;
-                    if (row >= S || c1 >= S || (causal && c1 > row)) s1[j] = -INFINITY;
-                    tile_max[j] = group16_max(fmaxf(s0[j], s1[j]));
+                for (int ks = 0; ks < 4; ++ks) {
+                    const int c0 = kv + ks * 16 + col;
+                    if (row >= S || c0 >= S || (causal && c0 > row))
+                        sq[ks][j] = -INFINITY;
                 }
-                float alpha[4];
-#pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    float mn = fmaxf(m[qs][j], tile_max[j]);
-                    alpha[j] = (m[qs][j] == -INFINITY) ? 0.f : __expf(m[qs][j] - mn);
-                    m[qs][j] = mn;
-                    float p0 = (s0[j] == -INFINITY) ? 0.f : __expf(s0[j] - mn);
-                    float p1 = (s1[j] == -INFINITY) ? 0.f : __expf(s1[j] - mn);
-                    s0[j] = p0;
-                    s1[j] = p1;
-                    // l kept as PER-LANE partials (this lane's 2 columns);
-                    // the cross-lane reduction happens once in the epilogue
-                    l[qs][j] = l[qs][j] * alpha[j] + p0 + p1;
-                }
-#pragma unroll
-                for (int dt = 0; dt < 8; ++dt)
-#pragma unroll
-                    for (int j = 0; j < 4; ++j) acc[qs][dt][j] *= alpha[j];
-                bf16raw *pw = Pw + qs * 16 * PS;
-#pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    pw[(kgrp * 4 + j) * PS + col] = f2bf(s0[j]);
-                    pw[(kgrp * 4 + j) * PS + 16 + col] = f2bf(s1[j]);
-                }
-                a_p[qs] = ld_frag(pw + col * PS + kgrp * 8);
+                tile_max[j] = group16_max(
+                    fmaxf(fmaxf(sq[0][j], sq[1][j]), fmaxf(sq[2][j], sq[3][j])));
             }
-            // PV for all subtiles per V fragment (one tr-read pair each)
-            __builtin_amdgcn_s_setprio(1);
+            float alpha[4];
 #pragma unroll
-            for (int dt = 0; dt < 8; ++dt) {
-                bf16x8v b_v = tr_frag(V_img + dt * VSUB, lane);
+            for (int j = 0; j < 4; ++j) {
+                float mn = fmaxf(m[qs][j], tile_max[j]);
+                alpha[j] = (m[qs][j] == -INFINITY) ? 0.f : __expf(m[qs][j] - mn);
+                m[qs][j] = mn;
+                float ps = 0.f;
 #pragma unroll
-                for (int qs = 0; qs < NQS; ++qs)
-                    acc[qs][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_p[qs], b_v, acc[qs][dt], 0, 0, 0);
+                for (int ks = 0; ks < 4; ++ks) {
+                    float p = (sq[ks][j] == -INFINITY) ? 0.f : __expf(sq[ks][j] - mn);
+                    sq[ks][j] = p;
+                    ps += p;
+                }
+                // l kept per-lane; reduced once in the epilogue
+                l[qs][j] = l[qs][j] * alpha[j] + ps;
             }
-            __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+            for (int dt = 0; dt < 8; ++dt)
+#pragma unroll
+                for (int j = 0; j < 4; ++j) acc[qs][dt][j] *= alpha[j];
+            bf16raw *pw = Pw + qs * 16 * PS2;
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+#pragma unroll
+                for (int ks = 0; ks < 4; ++ks)
+                    pw[(kgrp * 4 + j) * PS2 + ks * 16 + col] = f2bf(sq[ks][j]);
+            // PV over the two 32-key groups
+#pragma unroll
+            for (int g = 0; g < 2; ++g) {
+                bf16x8v a_p = ld_frag(pw + col * PS2 + g * 32 + kgrp * 8);
+#pragma unroll
+                for (int dt = 0; dt < 8; ++dt) {
+                    bf16x8v b_v = tr_frag(V_img + g * 8 * VSUB + dt * VSUB, lane);
+                    acc[qs][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_p, b_v, acc[qs][dt], 0, 0, 0);
+                }
+            }
         }
         stage_write(t + 1);
         __syncthreads();
@@ -277,8 +299,7 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     }
 
     bf16raw *op = o + ((long)b * S) * o_row + (long)h * DHEAD;
-    // reduce the per-lane l partials across the 16-lane column group
-    // exactly once (the main loop keeps l lane-local)
+    // reduce the per-lane l partials across the 16-lane column group once
 #pragma unroll
     for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
@@ -287,7 +308,7 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-            const int row = qbase + qs * 16 + kgrp * 4 + j;
+            const int row = qsb[qs] + kgrp * 4 + j;
             if (row >= S) continue;
             const float inv = (l[qs][j] > 0.f) ? 1.f / l[qs][j] : 0.f;
 #pragma unroll
@@ -308,8 +329,8 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
 //   3. fa_bwd_dq:   q-parallel like the forward; dQ written directly (bf16)
 // ==========================================================================
 #define BW_QT 32
-#define QS2 136   // padded [32][128] row stride
-#define TS2 40    // padded [..][32] row stride
+#define QS2 144   // padded [32][128] row stride
+#define TS2 48    // padded [..][32] row stride
 #define BKV 16    // keys per wave in dkv kernel
 
 extern "C" __global__ __launch_bounds__(FA_BLOCK)
@@ -349,9 +370,9 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
                        float *__restrict__ dkv,  // [B,S,Hkv,2,D] fp32
                        int B, int S, int Hq, int Hkv, int causal) {
     constexpr int KC = DQK / 32;
-    constexpr int KS_T = DQK + 8;
+    constexpr int KS_T = DQK + 16;
     constexpr int QSUB = DQK / 16;     // q-side tr-image d subtiles for Q
-    constexpr int QS_T = DQK + 8;      // padded Q row stride
+    constexpr int QS_T = DQK + 16;     // padded Q row stride
     extern __shared__ __attribute__((aligned(16))) char smem[];
     bf16raw *Q_lds = reinterpret_cast<bf16raw *>(smem);      // [32][QS_T]
     bf16raw *dO_lds = Q_lds + BW_QT * QS_T;                  // [32][QS2]
@@ -411,32 +432,64 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
 
     const int q_start = causal
         ? (kvblk * (WAVES * BKV) / BW_QT) * BW_QT : 0;
-    for (int qt = q_start; qt < S; qt += BW_QT) {
-        {
-            const int tid = threadIdx.x;
-            for (int e = tid * 8; e < BW_QT * DQK; e += FA_BLOCK * 8) {
-                const int r = e / DQK, d0 = e % DQK;
-                const int src = min(qt + r, S - 1);
-                bf16x8 qq = load8(qp + (long)src * q_row + d0);
-                store8(Q_lds + r * QS_T + d0, qq);
-                store8(Q_img + (d0 >> 4) * VSUB + v_img_row(r) * 16 + (d0 & 15),
-                       qq);
-            }
-            for (int e = tid * 8; e < BW_QT * DHEAD; e += FA_BLOCK * 8) {
-                const int r = e / DHEAD, d0 = e % DHEAD;
-                const int src = min(qt + r, S - 1);
-                bf16x8 dd = load8(dop + (long)src * o_row + d0);
-                store8(dO_lds + r * QS2 + d0, dd);
-                store8(dO_img + (d0 >> 4) * VSUB + v_img_row(r) * 16 + (d0 & 15),
-                       dd);
-            }
-            for (int r = tid; r < BW_QT; r += FA_BLOCK) {
-                const int src = min(qt + r, S - 1);
-                lse_lds[r] = lse[((long)b * Hq + h) * S + src];
-                D_lds[r] = Dsum[((long)b * Hq + h) * S + src];
-            }
+    // T14 split staging: tile qt+BW_QT's global loads are issued into
+    // registers while tile qt computes; the LDS write happens between the
+    // two barriers (r1 PMC: the serial stage->sync->compute loop left the
+    // q-tile loads fully exposed at 1 block/CU)
+    constexpr int QCH = BW_QT * DQK / (FA_BLOCK * 8);
+    constexpr int OCH = BW_QT * DHEAD / (FA_BLOCK * 8);
+    bf16x8 st_q[QCH], st_do[OCH];
+    float st_lse = 0.f, st_D = 0.f;
+    auto q_stage_load = [&](int qt) {
+        if (qt >= S) return;
+        const int tid = threadIdx.x;
+#pragma unroll
+        for (int c = 0; c < QCH; ++c) {
+            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int r = e / DQK, d0 = e % DQK;
+            st_q[c] = load8(qp + (long)min(qt + r, S - 1) * q_row + d0);
         }
-        __syncthreads();
+#pragma unroll
+        for (int c = 0; c < OCH; ++c) {
+            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int r = e / DHEAD, d0 = e % DHEAD;
+            st_do[c] = load8(dop + (long)min(qt + r, S - 1) * o_row + d0);
+        }
+        if (tid < BW_QT) {
+            const int src = min(qt + tid, S - 1);
+            st_lse = lse[((long)b * Hq + h) * S + src];
+            st_D = Dsum[((long)b * Hq + h) * S + src];
+        }
+    };
+    auto q_stage_write = [&](int qt) {
+        if (qt >= S) return;
+        const int tid = threadIdx.x;
+#pragma unroll
+        for (int c = 0; c < QCH; ++c) {
+            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int r = e / DQK, d0 = e % DQK;
+            store8(Q_lds + r * QS_T + d0, st_q[c]);
+            store8(Q_img + (d0 >> 4) * VSUB + v_img_row(r) * 16 + (d0 & 15),
+                   st_q[c]);
+        }
+#pragma unroll
+        for (int c = 0; c < OCH; ++c) {
+            const int e = tid * 8 + c * FA_BLOCK * 8;
+            const int r = e / DHEAD, d0 = e % DHEAD;
+            store8(dO_lds + r * QS2 + d0, st_do[c]);
+            store8(dO_img + (d0 >> 4) * VSUB + v_img_row(r) * 16 + (d0 & 15),
+                   st_do[c]);
+        }
+        if (tid < BW_QT) {
+            lse_lds[tid] = st_lse;
+            D_lds[tid] = st_D;
+        }
+    };
+    q_stage_load(q_start);
+    q_stage_write(q_start);
+    __syncthreads();
+    q_stage_load(q_start + BW_QT);
+    for (int qt = q_start; qt < S; qt += BW_QT) {
 
 #pragma unroll
         for (int qs = 0; qs < 2; ++qs) {
@@ -484,7 +537,10 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
                     a_dst, b_q, dk_acc[dt], 0, 0, 0);
             }
         }
-        __syncthreads();
+        __syncthreads();          // all waves done reading this q tile
+        q_stage_write(qt + BW_QT);
+        __syncthreads();          // next tile visible
+        q_stage_load(qt + 2 * BW_QT);
     }
 
     // dkv layout: [B, S, Hkv, DQK + DHEAD] fp32 (dk then dv)
@@ -514,7 +570,7 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
                       bf16raw *__restrict__ dq,
                       int B, int S, int Hq, int Hkv, int causal) {
     constexpr int KC = DQK / 32;
-    constexpr int KS_T = DQK + 8;
+    constexpr int KS_T = DQK + 16;
     constexpr int QSUB = DQK / 16;
     constexpr int QT = NQS * 16;
     constexpr int BUF = KVTILE * KS_T + KVTILE * KS + QSUB * VSUB;
@@ -543,9 +599,10 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
     const bf16raw *dop = dout + ((long)b * S) * o_row + (long)h * DHEAD;
     bf16raw *dSw = dS_all + wave * QT * PS;
 
-    // per-wave q-side registers: scaled Q frags, lse, D; dO fragments are
-    // re-read from global per tile (the WG's dO tile stays L2-resident)
-    bf16x8v a_q[NQS][KC];
+    // per-wave q-side registers: scaled Q frags, dO frags, lse, D — all
+    // kv-tile-invariant, so they are loaded ONCE (r1 PMC: re-reading dO
+    // from global inside the kv loop parked the wave 51.5% of its cycles)
+    bf16x8v a_q[NQS][KC], a_do[NQS][4];
     float lse_r[NQS][4], D_r[NQS][4];
 #pragma unroll
     for (int qs = 0; qs < NQS; ++qs) {
@@ -558,6 +615,10 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
             for (int j = 0; j < 8; ++j) sc.set(j, raw.get(j) * scale);
             a_q[qs][kc] = *reinterpret_cast<bf16x8v *>(&sc.raw);
         }
+#pragma unroll
+        for (int kc = 0; kc < 4; ++kc)
+            a_do[qs][kc] = ld_frag(
+                dop + (long)min(qrow, S - 1) * o_row + kc * 32 + kgrp * 8);
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
             const int row = qbase + qs * 16 + kgrp * 4 + j;
@@ -648,9 +709,8 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
                 for (int kc = 0; kc < 4; ++kc) {
                     bf16x8v bv0 = ld_frag(V_lds + col * KS + kc * 32 + kgrp * 8);
                     bf16x8v bv1 = ld_frag(V_lds + (16 + col) * KS + kc * 32 + kgrp * 8);
-                    bf16x8v a_do = ld_frag(dop + (long)min(qbase + qs * 16 + col, S - 1) * o_row + kc * 32 + kgrp * 8);
-                    dp0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do, bv0, dp0, 0, 0, 0);
-                    dp1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do, bv1, dp1, 0, 0, 0);
+                    dp0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do[qs][kc], bv0, dp0, 0, 0, 0);
+                    dp1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do[qs][kc], bv1, dp1, 0, 0, 0);
                 }
 #pragma unroll
                 for (int j = 0; j < 4; ++j) {
@@ -700,21 +760,18 @@ extern "C" void fa_fwd_launch(const void *q, const void *k, const void *v,
                               int Hkv, int dqk, int causal,
                               hipStream_t stream) {
     if (dqk == 128) {
-        // NQS=1 -> 3 waves/SIMD: measured +18-30% over NQS=2
-        constexpr int QT = 16;
-        dim3 grid(CDIV(S, WAVES * QT), Hq, B);
-        size_t smem = (2 * (KVTILE * (128 + 8) + 8 * VSUB) + WAVES * QT * PS)
-                      * sizeof(bf16raw);
-        hipLaunchKernelGGL((fa_fwd_kernel<128, 1>), grid, dim3(FA_BLOCK),
+        dim3 grid(CDIV(S, 256), Hq, B);
+        size_t smem = (2 * (KVT2 * (128 + 16) + 2 * 8 * VSUB)
+                       + FWD_WAVES * 2 * 16 * PS2) * sizeof(bf16raw);
+        hipLaunchKernelGGL((fa_fwd_kernel<128>), grid, dim3(FWD_BLOCK),
                            smem, stream, (const bf16raw *)q,
                            (const bf16raw *)k, (const bf16raw *)v,
                            (bf16raw *)o, (float *)lse, B, S, Hq, Hkv, causal);
     } else if (dqk == 192) {
-        constexpr int QT = 16;
-        dim3 grid(CDIV(S, WAVES * QT), Hq, B);
-        size_t smem = (2 * (KVTILE * (192 + 8) + 8 * VSUB) + WAVES * QT * PS)
-                      * sizeof(bf16raw);
-        hipLaunchKernelGGL((fa_fwd_kernel<192, 1>), grid, dim3(FA_BLOCK),
+        dim3 grid(CDIV(S, 256), Hq, B);
+        size_t smem = (2 * (KVT2 * (192 + 16) + 2 * 8 * VSUB)
+                       + FWD_WAVES * 2 * 16 * PS2) * sizeof(bf16raw);
+        hipLaunchKernelGGL((fa_fwd_kernel<192>), grid, dim3(FWD_BLOCK),
                            smem, stream, (const bf16raw *)q,
                            (const bf16raw *)k, (const bf16raw *)v,
                            (bf16raw *)o, (float *)lse, B, S, Hq, Hkv, causal);
@@ -742,10 +799,10 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
         dim3 grid(CDIV(S, WAVES * BKV), Hq, B);
         size_t smem;
         if (dqk == 128) {
-            smem = (BW_QT * (128 + 8) + BW_QT * QS2 + (128 / 16 + 8) * VSUB)
+            smem = (BW_QT * (128 + 16) + BW_QT * QS2 + (128 / 16 + 8) * VSUB)
                        * sizeof(bf16raw)
                    + 2 * BW_QT * sizeof(float)
-                   + WAVES * (BKV * (128 + 8) + 2 * BKV * TS2) * sizeof(bf16raw);
+                   + WAVES * (BKV * (128 + 16) + 2 * BKV * TS2) * sizeof(bf16raw);
             hipLaunchKernelGGL((fa_bwd_dkv_kernel<128>), grid, dim3(FA_BLOCK),
                                smem, stream, (const bf16raw *)dout,
                                (const bf16raw *)q, (const bf16raw *)k,
@@ -753,10 +810,10 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
                                (const float *)dsum, (float *)dkv, B, S, Hq,
                                Hkv, causal);
         } else if (dqk == 192) {
-            smem = (BW_QT * (192 + 8) + BW_QT * QS2 + (192 / 16 + 8) * VSUB)
+            smem = (BW_QT * (192 + 16) + BW_QT * QS2 + (192 / 16 + 8) * VSUB)
                        * sizeof(bf16raw)
                    + 2 * BW_QT * sizeof(float)
-                   + WAVES * (BKV * (192 + 8) + 2 * BKV * TS2) * sizeof(bf16raw);
+                   + WAVES * (BKV * (192 + 16) + 2 * BKV * TS2) * sizeof(bf16raw);
             hipLaunchKernelGGL((fa_bwd_dkv_kernel<192>), grid, dim3(FA_BLOCK),
                                smem, stream, (const bf16raw *)dout,
                                (const bf16raw *)q, (const bf16raw *)k,
@@ -771,7 +828,7 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
     if (dqk == 128) {
         constexpr int QT = 16;   // NQS=1: occupancy over per-wave work
         dim3 grid(CDIV(S, WAVES * QT), Hq, B);
-        size_t smem = (2 * (KVTILE * (128 + 8) + KVTILE * KS + 8 * VSUB)
+        size_t smem = (2 * (KVTILE * (128 + 16) + KVTILE * KS + 8 * VSUB)
                        + WAVES * QT * PS) * sizeof(bf16raw);
         hipLaunchKernelGGL((fa_bwd_dq_kernel<128, 1>), grid, dim3(FA_BLOCK),
                            smem, stream, (const bf16raw *)dout,
@@ -782,7 +839,7 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
     } else {
         constexpr int QT = 16;
         dim3 grid(CDIV(S, WAVES * QT), Hq, B);
-        size_t smem = (2 * (KVTILE * (192 + 8) + KVTILE * KS + 12 * VSUB)
+        size_t smem = (2 * (KVTILE * (192 + 16) + KVTILE * KS + 12 * VSUB)
                        + WAVES * QT * PS) * sizeof(bf16raw);
         hipLaunchKernelGGL((fa_bwd_dq_kernel<192, 1>), grid, dim3(FA_BLOCK),
                            smem, stream, (const bf16raw *)dout,
