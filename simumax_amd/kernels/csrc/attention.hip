@@ -229,6 +229,12 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
 #pragma unroll
         for (int qs = 0; qs < NQS; ++qs) {
             if (causal && kv >= qsb[qs] + 16) continue;
+            // interior tiles (every row/col in range, fully below the
+            // causal diagonal) skip the per-element masking entirely —
+            // the masking VALU, not the MFMAs, dominated the issued
+            // instructions in the r1 PMC profile
+            const bool clean = (qsb[qs] + 16 <= S) && (kv + KVT2 <= S)
+                && (!causal || kv + KVT2 - 1 <= qsb[qs]);
             // QK^T: 4 16-col K subtiles x KC k-chunks
             f32x4 sq[4];
 #pragma unroll
@@ -243,28 +249,33 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
             }
             // online softmax over the 64 kv cols
             float tile_max[4];
+            if (!clean) {
 #pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                const int row = qsb[qs] + kgrp * 4 + j;
+                for (int j = 0; j < 4; ++j) {
+                    const int row = qsb[qs] + kgrp * 4 + j;
 #pragma unroll
-                for (int ks = 0; ks < 4; ++ks) {
-                    const int c0 = kv + ks * 16 + col;
-                    if (row >= S || c0 >= S || (causal && c0 > row))
-                        sq[ks][j] = -INFINITY;
+                    for (int ks = 0; ks < 4; ++ks) {
+                        const int c0 = kv + ks * 16 + col;
+                        if (row >= S || c0 >= S || (causal && c0 > row))
+                            sq[ks][j] = -INFINITY;
+                    }
                 }
+            }
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
                 tile_max[j] = group16_max(
                     fmaxf(fmaxf(sq[0][j], sq[1][j]), fmaxf(sq[2][j], sq[3][j])));
-            }
             float alpha[4];
 #pragma unroll
             for (int j = 0; j < 4; ++j) {
                 float mn = fmaxf(m[qs][j], tile_max[j]);
-                alpha[j] = (m[qs][j] == -INFINITY) ? 0.f : __expf(m[qs][j] - mn);
+                // exp(-inf - mn) = 0, so the -INFINITY selects are free
+                alpha[j] = __expf(m[qs][j] - mn);
                 m[qs][j] = mn;
                 float ps = 0.f;
 #pragma unroll
                 for (int ks = 0; ks < 4; ++ks) {
-                    float p = (sq[ks][j] == -INFINITY) ? 0.f : __expf(sq[ks][j] - mn);
+                    float p = __expf(sq[ks][j] - mn);
                     sq[ks][j] = p;
                     ps += p;
                 }
@@ -506,14 +517,23 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
                 bf16x8v a_do = ld_frag(dO_lds + (qs * 16 + col) * QS2 + kc * 32 + kgrp * 8);
                 dp0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do, b_v[kc], dp0, 0, 0, 0);
             }
+            // interior tiles: every row/col real and strictly below the
+            // diagonal -> no per-element masking (PMC: mask VALU-bound)
+            const bool clean = (qt + qs * 16 + 16 <= S) && (kvbase + BKV <= S)
+                && (!causal || kvbase + BKV - 1 <= qt + qs * 16);
 #pragma unroll
             for (int j = 0; j < 4; ++j) {
-                const int row = qt + qs * 16 + kgrp * 4 + j;
-                const int c0 = kvbase + col;
                 const float ls = lse_lds[qs * 16 + kgrp * 4 + j];
                 const float Dv = D_lds[qs * 16 + kgrp * 4 + j];
-                float p0 = (row < S && c0 < S && (!causal || c0 <= row) && ls != -INFINITY)
-                               ? __expf(s0[j] - ls) : 0.f;
+                float p0;
+                if (clean) {
+                    p0 = __expf(s0[j] - ls);
+                } else {
+                    const int row = qt + qs * 16 + kgrp * 4 + j;
+                    const int c0 = kvbase + col;
+                    p0 = (row < S && c0 < S && (!causal || c0 <= row)
+                          && ls != -INFINITY) ? __expf(s0[j] - ls) : 0.f;
+                }
                 float ds0 = p0 * (dp0[j] - Dv) * scale;
                 const int qrow = qs * 16 + kgrp * 4 + j;
                 Pt_l[col * TS2 + qrow] = f2bf(p0);
@@ -712,16 +732,25 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
                     dp0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do[qs][kc], bv0, dp0, 0, 0, 0);
                     dp1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do[qs][kc], bv1, dp1, 0, 0, 0);
                 }
+                const bool clean = (qbase + qs * 16 + 16 <= S)
+                    && (kv + KVTILE <= S)
+                    && (!causal || kv + KVTILE - 1 <= qbase + qs * 16);
 #pragma unroll
                 for (int j = 0; j < 4; ++j) {
-                    const int row = qbase + qs * 16 + kgrp * 4 + j;
-                    const int c0 = kv + col, c1 = kv + 16 + col;
                     const float ls = lse_r[qs][j];
                     const float Dv = D_r[qs][j];
-                    float p0 = (row < S && c0 < S && (!causal || c0 <= row) && ls != -INFINITY)
-                                   ? __expf(s0[j] - ls) : 0.f;
-                    float p1 = (row < S && c1 < S && (!causal || c1 <= row) && ls != -INFINITY)
-                                   ? __expf(s1[j] - ls) : 0.f;
+                    float p0, p1;
+                    if (clean) {
+                        p0 = __expf(s0[j] - ls);
+                        p1 = __expf(s1[j] - ls);
+                    } else {
+                        const int row = qbase + qs * 16 + kgrp * 4 + j;
+                        const int c0 = kv + col, c1 = kv + 16 + col;
+                        p0 = (row < S && c0 < S && (!causal || c0 <= row)
+                              && ls != -INFINITY) ? __expf(s0[j] - ls) : 0.f;
+                        p1 = (row < S && c1 < S && (!causal || c1 <= row)
+                              && ls != -INFINITY) ? __expf(s1[j] - ls) : 0.f;
+                    }
                     float ds0 = p0 * (dp0[j] - Dv) * scale;
                     float ds1 = p1 * (dp1[j] - Dv) * scale;
                     bf16raw *pw = dSw + qs * 16 * PS;
