@@ -226,28 +226,41 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
         bf16raw *K_lds = buf0 + (t & 1) * BUF_ELEMS;
         bf16raw *V_img = K_lds + KVT2 * KS_T;
 
+        // subtile 1 (high rows) is always active (its last row is the
+        // block's last row >= kv_limit); subtile 0 (low rows) goes inactive
+        // once the causal diagonal passes it. In the common both-active case
+        // the K and V LDS fragments are read ONCE and feed both subtiles'
+        // MFMA chains (halves the LDS read traffic).
+        const bool act0 = !(causal && kv >= qsb[0] + 16);
+
+        f32x4 sq[NQS][4];
 #pragma unroll
-        for (int qs = 0; qs < NQS; ++qs) {
-            if (causal && kv >= qsb[qs] + 16) continue;
-            // interior tiles (every row/col in range, fully below the
-            // causal diagonal) skip the per-element masking entirely —
-            // the masking VALU, not the MFMAs, dominated the issued
-            // instructions in the r1 PMC profile
-            const bool clean = (qsb[qs] + 16 <= S) && (kv + KVT2 <= S)
-                && (!causal || kv + KVT2 - 1 <= qsb[qs]);
-            // QK^T: 4 16-col K subtiles x KC k-chunks
-            f32x4 sq[4];
+        for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
-            for (int ks = 0; ks < 4; ++ks) sq[ks] = f32x4{0, 0, 0, 0};
+            for (int ks = 0; ks < 4; ++ks) sq[qs][ks] = f32x4{0, 0, 0, 0};
+        if (act0) {
 #pragma unroll
-            for (int kc = 0; kc < KC; ++kc) {
+            for (int kc = 0; kc < KC; ++kc)
 #pragma unroll
                 for (int ks = 0; ks < 4; ++ks) {
                     bf16x8v bk = ld_frag(K_lds + (ks * 16 + col) * KS_T + kc * 32 + kgrp * 8);
-                    sq[ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], bk, sq[ks], 0, 0, 0);
+                    sq[0][ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[0][kc], bk, sq[0][ks], 0, 0, 0);
+                    sq[1][ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[1][kc], bk, sq[1][ks], 0, 0, 0);
                 }
-            }
-            // online softmax over the 64 kv cols
+        } else {
+#pragma unroll
+            for (int kc = 0; kc < KC; ++kc)
+#pragma unroll
+                for (int ks = 0; ks < 4; ++ks) {
+                    bf16x8v bk = ld_frag(K_lds + (ks * 16 + col) * KS_T + kc * 32 + kgrp * 8);
+                    sq[1][ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[1][kc], bk, sq[1][ks], 0, 0, 0);
+                }
+        }
+
+        // online softmax + P->LDS per active subtile
+        auto softmax_p = [&](int qs) {
+            const bool clean = (qsb[qs] + 16 <= S) && (kv + KVT2 <= S)
+                && (!causal || kv + KVT2 - 1 <= qsb[qs]);
             float tile_max[4];
             if (!clean) {
 #pragma unroll
@@ -257,14 +270,15 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
                     for (int ks = 0; ks < 4; ++ks) {
                         const int c0 = kv + ks * 16 + col;
                         if (row >= S || c0 >= S || (causal && c0 > row))
-                            sq[ks][j] = -INFINITY;
+                            sq[qs][ks][j] = -INFINITY;
                     }
                 }
             }
 #pragma unroll
             for (int j = 0; j < 4; ++j)
                 tile_max[j] = group16_max(
-                    fmaxf(fmaxf(sq[0][j], sq[1][j]), fmaxf(sq[2][j], sq[3][j])));
+                    fmaxf(fmaxf(sq[qs][0][j], sq[qs][1][j]),
+                          fmaxf(sq[qs][2][j], sq[qs][3][j])));
             float alpha[4];
 #pragma unroll
             for (int j = 0; j < 4; ++j) {
@@ -275,8 +289,8 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
                 float ps = 0.f;
 #pragma unroll
                 for (int ks = 0; ks < 4; ++ks) {
-                    float p = __expf(sq[ks][j] - mn);
-                    sq[ks][j] = p;
+                    float p = __expf(sq[qs][ks][j] - mn);
+                    sq[qs][ks][j] = p;
                     ps += p;
                 }
                 // l kept per-lane; reduced once in the epilogue
@@ -291,16 +305,31 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
             for (int j = 0; j < 4; ++j)
 #pragma unroll
                 for (int ks = 0; ks < 4; ++ks)
-                    pw[(kgrp * 4 + j) * PS2 + ks * 16 + col] = f2bf(sq[ks][j]);
-            // PV over the two 32-key groups
+                    pw[(kgrp * 4 + j) * PS2 + ks * 16 + col] = f2bf(sq[qs][ks][j]);
+        };
+        if (act0) softmax_p(0);
+        softmax_p(1);
+
+        // PV over the two 32-key groups; V fragments shared across subtiles
 #pragma unroll
-            for (int g = 0; g < 2; ++g) {
-                bf16x8v a_p = ld_frag(pw + col * PS2 + g * 32 + kgrp * 8);
+        for (int g = 0; g < 2; ++g) {
+            bf16x8v a_p0 = ld_frag(Pw + col * PS2 + g * 32 + kgrp * 8);
+            bf16x8v a_p1 = ld_frag(Pw + 16 * PS2 + col * PS2 + g * 32 + kgrp * 8);
+            if (act0) {
 #pragma unroll
                 for (int dt = 0; dt < 8; ++dt) {
                     bf16x8v b_v = tr_frag(V_img + g * 8 * VSUB + dt * VSUB, lane);
-                    acc[qs][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a_p, b_v, acc[qs][dt], 0, 0, 0);
+                    acc[0][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_p0, b_v, acc[0][dt], 0, 0, 0);
+                    acc[1][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_p1, b_v, acc[1][dt], 0, 0, 0);
+                }
+            } else {
+#pragma unroll
+                for (int dt = 0; dt < 8; ++dt) {
+                    bf16x8v b_v = tr_frag(V_img + g * 8 * VSUB + dt * VSUB, lane);
+                    acc[1][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a_p1, b_v, acc[1][dt], 0, 0, 0);
                 }
             }
         }
@@ -417,8 +446,10 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
     const bf16raw *dop = dout + ((long)b * S) * o_row + (long)h * DHEAD;
     const float scale = rsqrtf((float)DQK);
 
-    // stage this wave's scaled K; V as B-fragments in registers
-    bf16x8v b_v[4];
+    // stage this wave's scaled K; K and V B-fragments held in REGISTERS
+    // across the whole q loop (kv-tile-invariant; K_l LDS keeps the tr
+    // image source for dK)
+    bf16x8v b_v[4], b_k[KC];
     {
         for (int e = lane * 8; e < BKV * DQK; e += WAVE * 8) {
             const int kvr = e / DQK, d0 = e % DQK;
@@ -433,6 +464,10 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
 #pragma unroll
         for (int kc = 0; kc < 4; ++kc)
             b_v[kc] = ld_frag(vp + (long)src * v_row + kc * 32 + kgrp * 8);
+        __syncthreads();   // K_l visible (same wave wrote it, but be strict)
+#pragma unroll
+        for (int kc = 0; kc < KC; ++kc)
+            b_k[kc] = ld_frag(K_l + col * KS_T + kc * 32 + kgrp * 8);
     }
 
     f32x4 dv_acc[8], dk_acc[QSUB];
@@ -509,8 +544,7 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
 #pragma unroll
             for (int kc = 0; kc < KC; ++kc) {
                 bf16x8v a_qf = ld_frag(Q_lds + (qs * 16 + col) * QS_T + kc * 32 + kgrp * 8);
-                bf16x8v b0 = ld_frag(K_l + col * KS_T + kc * 32 + kgrp * 8);
-                s0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_qf, b0, s0, 0, 0, 0);
+                s0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_qf, b_k[kc], s0, 0, 0, 0);
             }
 #pragma unroll
             for (int kc = 0; kc < 4; ++kc) {

@@ -297,3 +297,21 @@ def test_grouped_gemm_kernels():
                             for e in range(E)])
     E_.grouped_wgrad(dout, x, g)
     assert relerr(g, gref) < 1e-2
+
+
+@pytest.mark.gpu
+def test_flash_attention_fwd_rescale_branch():
+    """Force the defer-max rescale branch mid-stream (guide T13 rule 26):
+    spike one K row against one Q row so the running max jumps past the
+    threshold at a late tile; compare against the fp32 reference."""
+    B, S, Hq, Hkv, D = 1, 1024, 4, 4, 128
+    torch.manual_seed(7)
+    q = torch.randn(B, S, Hq, D, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(B, S, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(B, S, Hkv, D, device=DEV, dtype=torch.bfloat16)
+    # spike: key 900 strongly aligned with queries >= 900 (causal-visible)
+    k[:, 900, :, :] = 30.0
+    q[:, 950:, :, :] += 3.0
+    o, lse = ext().fa_fwd(q, k, v, True)
+    oref = K._sdpa_torch(q, k, v, True)
+    assert relerr(o, oref) < 3e-2, f"rescale-branch rel err {relerr(o, oref)}"
