@@ -40,6 +40,8 @@ def parse_args():
                     help="tensor-parallel degree (world must be tp*pp*dp)")
     ap.add_argument("--pp", type=int, default=1,
                     help="pipeline-parallel degree (1F1B)")
+    ap.add_argument("--cp", type=int, default=1,
+                    help="Ulysses context-parallel degree (seq sharded)")
     return ap.parse_args()
 
 
@@ -53,7 +55,7 @@ def predict(model_cfg, world, args):
         micro_batch_size=args.mbs,
         micro_batch_num=args.mbc,
         world_size=world,
-        tp_size=args.tp, pp_size=args.pp, ep_size=1,
+        tp_size=args.tp, pp_size=args.pp, ep_size=1, cp_size=args.cp,
         enable_sequence_parallel=False,
         zero_state=0,                # trainer replicates optimizer state
         use_fp32_accum_grad=True,
@@ -101,7 +103,7 @@ def main():
 
     tc = TrainConfig(seq_len=args.seq_len, micro_batch_size=args.mbs,
                      micro_batch_num=args.mbc, tp_size=args.tp,
-                     pp_size=args.pp)
+                     pp_size=args.pp, cp_size=args.cp)
     device = f"cuda:{local_rank}"
     t0 = time.time()
     ps = None
@@ -123,12 +125,19 @@ def main():
         print(f"[bench] built {args.model} ({n_params/1e9:.2f}B params"
               f"{' on this rank' if args.tp * args.pp > 1 else ''}) "
               f"in {time.time()-t0:.1f}s", file=sys.stderr)
-    # each DATA-parallel column gets distinct data (tp/pp peers share it)
+    # each DATA-parallel column gets distinct data (tp/pp/cp peers share it)
+    mp_deg = args.tp * args.cp
     dp_rank = ps.dp_rank if ps is not None else (
-        rank // args.tp if args.tp > 1 else rank)
+        rank // mp_deg if mp_deg > 1 else rank)
     toks, labels = make_synthetic_batch(model_cfg.vocab_size, args.mbc,
                                         args.mbs, args.seq_len, device,
                                         seed=1000 + dp_rank)
+    if args.cp > 1:
+        # cp ranks train on their seq slice of the SAME batch
+        s_loc = args.seq_len // args.cp
+        cp_rank = rank % args.cp
+        sl = slice(cp_rank * s_loc, (cp_rank + 1) * s_loc)
+        toks, labels = toks[:, :, sl].contiguous(), labels[:, :, sl].contiguous()
 
     for _ in range(args.warmup):
         step_fn(model, opt, reducer, toks, labels, args.mbc)
@@ -154,7 +163,7 @@ def main():
     peak_bytes = pk.item()
 
     if rank == 0:
-        dp = world // (args.tp * args.pp)
+        dp = world // (args.tp * args.pp * args.cp)
         tokens_per_iter = args.mbs * args.mbc * dp * args.seq_len
         flops_token = model_cfg.flops_per_token(args.seq_len)
         peak_tflops = 2500.0
@@ -191,6 +200,7 @@ def main():
                 "global_batch": args.mbs * args.mbc * dp,
                 "seq_len": args.seq_len,
                 "parallelism": (f"tp{args.tp}." if args.tp > 1 else "")
+                               + (f"cp{args.cp}." if args.cp > 1 else "")
                                + (f"pp{args.pp}." if args.pp > 1 else "")
                                + f"dp{dp}",
                 "layers": model_cfg.layer_num,

@@ -1,0 +1,119 @@
+"""Ulysses-style context parallelism for the reference trainer.
+
+Mirrors the simulator's CP a2a cost model (ops/dense.py CoreAttention
+cp_comm_type="a2a", reference dense_module.py:1158-1338): the sequence is
+sharded over the cp group outside attention; inside attention an
+all-to-all scatters heads and gathers sequence (q, k, v), flash attention
+runs on the full sequence with head_num/cp local heads, and a reverse
+all-to-all restores the seq-sharded layout for the out projection.
+
+Gradient semantics: cp ranks see the same batch but different seq slices,
+so parameter grads are averaged over the dp*cp group — the trainer's
+reducer already spans the whole world when tp=ep=1, which IS dp_cp.
+
+Process-group layout matches core/utils.get_rank_group: cp consecutive
+(tp fastest, but the trainer composes cp only with pure DP for now).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+_CP_GROUPS = {}
+
+
+def get_cp_groups(cp_size):
+    """cp group = cp_size consecutive ranks. Returns (cp_group, cp_rank)."""
+    if cp_size <= 1 or not dist.is_initialized():
+        return None, 0
+    key = (cp_size, dist.get_world_size())
+    if key not in _CP_GROUPS:
+        world = dist.get_world_size()
+        assert world % cp_size == 0
+        groups = {}
+        for start in range(0, world, cp_size):
+            g = dist.new_group(list(range(start, start + cp_size)))
+            for r in range(start, start + cp_size):
+                groups[r] = g
+        _CP_GROUPS[key] = groups
+    groups = _CP_GROUPS[key]
+    r = dist.get_rank()
+    return groups[r], r % cp_size
+
+
+def _a2a_exchange(chunks, group):
+    """all_to_all of equal-shaped stacked chunks ([cp, ...] tensor).
+    gloo has no all_to_all: fall back to all_gather + select."""
+    cp = dist.get_world_size(group)
+    out = torch.empty_like(chunks)
+    backend = dist.get_backend(group)
+    if backend == "nccl":
+        dist.all_to_all_single(out, chunks.contiguous(), group=group)
+    else:
+        gathered = [torch.empty_like(chunks) for _ in range(cp)]
+        dist.all_gather(gathered, chunks.contiguous(), group=group)
+        me = dist.get_rank(group)
+        for j in range(cp):
+            out[j] = gathered[j][me]
+    return out
+
+
+def _scatter_head_gather_seq(x, group):
+    """[B, s, H, d] -> [B, s*cp, H/cp, d] (s local seq, H full heads)."""
+    cp = dist.get_world_size(group)
+    B, s, H, d = x.shape
+    Hc = H // cp
+    # [cp, B, s, Hc, d]: chunk i = heads [i*Hc, (i+1)*Hc)
+    chunks = x.view(B, s, cp, Hc, d).permute(2, 0, 1, 3, 4).contiguous()
+    out = _a2a_exchange(chunks, group)
+    # out[j] = seq slice j of my head block -> concat along seq
+    return out.permute(1, 0, 2, 3, 4).reshape(B, cp * s, Hc, d)
+
+
+def _scatter_seq_gather_head(x, group):
+    """[B, S, Hc, d] -> [B, S/cp, Hc*cp, d] (inverse of the above)."""
+    cp = dist.get_world_size(group)
+    B, S, Hc, d = x.shape
+    s = S // cp
+    # [cp, B, s, Hc, d]: chunk j = seq slice j
+    chunks = x.view(B, cp, s, Hc, d).permute(1, 0, 2, 3, 4).contiguous()
+    out = _a2a_exchange(chunks, group)
+    # out[i] = my seq slice of head block i -> concat along heads
+    return out.permute(1, 2, 0, 3, 4).reshape(B, s, cp * Hc, d)
+
+
+class _HeadScatterSeqGather(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return _scatter_head_gather_seq(x, group)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return _scatter_seq_gather_head(dy.contiguous(), ctx.group), None
+
+
+class _SeqScatterHeadGather(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return _scatter_seq_gather_head(x, group)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return _scatter_head_gather_seq(dy.contiguous(), ctx.group), None
+
+
+def cp_pre_attention(x, group):
+    """q/k/v [B, s, H, d] -> [B, S, H/cp, d] before flash attention."""
+    if group is None:
+        return x
+    return _HeadScatterSeqGather.apply(x, group)
+
+
+def cp_post_attention(x, group):
+    """o [B, S, H/cp, d] -> [B, s, H, d] after flash attention."""
+    if group is None:
+        return x
+    return _SeqScatterHeadGather.apply(x, group)

@@ -25,14 +25,19 @@ class GQAAttention(nn.Module):
     head sections), row-parallel out projection (Megatron layout)."""
 
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
-                 tp_group=None, tp_size=1, sp=False):
+                 tp_group=None, tp_size=1, sp=False, cp_group=None,
+                 cp_size=1):
         super().__init__()
         h = cfg.hidden_size
         assert cfg.head_num % tp_size == 0 and cfg.kv_head_num % tp_size == 0
         self.heads = cfg.head_num // tp_size
         self.kv_heads = cfg.kv_head_num // tp_size
+        if cp_size > 1:
+            # Ulysses CP: heads scattered over cp inside attention
+            assert self.heads % cp_size == 0 and self.kv_heads % cp_size == 0
         self.head_size = cfg.head_size
         self.tp_group = tp_group
+        self.cp_group = cp_group
         self.sp = sp
         qkv_out = (self.heads + 2 * self.kv_heads) * cfg.head_size
         self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype, device=device)
@@ -59,7 +64,18 @@ class GQAAttention(nn.Module):
         q = q.view(B, S, self.heads, d)
         k = k.view(B, S, self.kv_heads, d)
         v = v.reshape(B, S, self.kv_heads, d)
-        ctx = K.flash_attention(q, k, v, causal=True)
+        if self.cp_group is not None:
+            # a2a: scatter heads / gather sequence (Ulysses), flash on the
+            # full sequence, inverse a2a on the context
+            from .cp import cp_post_attention, cp_pre_attention
+
+            q = cp_pre_attention(q, self.cp_group)
+            k = cp_pre_attention(k, self.cp_group)
+            v = cp_pre_attention(v, self.cp_group)
+            ctx = K.flash_attention(q, k, v, causal=True)
+            ctx = cp_post_attention(ctx, self.cp_group)
+        else:
+            ctx = K.flash_attention(q, k, v, causal=True)
         out = self.out_proj(ctx.reshape(B, S, self.heads * d))
         if self.sp:
             return scatter_seq(out, self.tp_group)
@@ -116,7 +132,7 @@ class MLAAttention(nn.Module):
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
                  layer_idx=0, ep_group=None, ep_size=1, tp_group=None,
-                 tp_size=1, sp=False):
+                 tp_size=1, sp=False, cp_group=None, cp_size=1):
         super().__init__()
         h = cfg.hidden_size
         self.tp_group = tp_group
@@ -124,11 +140,13 @@ class LlamaDecoderLayer(nn.Module):
         self.input_norm = K.RMSNorm(h, dtype=dtype, device=device)
         if getattr(cfg, "attention_type", "gqa") == "mla":
             assert tp_size == 1, "MLA requires tp_size == 1 (simulator parity)"
+            assert cp_size == 1, "trainer CP supports GQA attention"
             self.attention = MLAAttention(cfg, dtype=dtype, device=device)
         else:
             self.attention = GQAAttention(cfg, dtype=dtype, device=device,
                                           tp_group=tp_group, tp_size=tp_size,
-                                          sp=sp)
+                                          sp=sp, cp_group=cp_group,
+                                          cp_size=cp_size)
         if sp:
             # SP norms see only the local seq shard: their weight grads
             # are partial sums and need a tp all_reduce (reducer handles
@@ -180,10 +198,13 @@ class LlamaDecoderLayer(nn.Module):
 class LlamaForTraining(nn.Module):
     def __init__(self, cfg: ModelConfig, seq_len: int, dtype=torch.bfloat16,
                  rope_base=500000.0, device=None, ep_group=None, ep_size=1,
-                 tp_group=None, tp_size=1, tp_rank=0, sp=False):
+                 tp_group=None, tp_size=1, tp_rank=0, sp=False,
+                 cp_group=None, cp_rank=0, cp_size=1):
         super().__init__()
         self.cfg = cfg
-        self.seq_len = seq_len
+        self.seq_len = seq_len          # FULL sequence (rope cache size)
+        self.cp_rank = cp_rank
+        self.cp_size = cp_size
         self.tp_group = tp_group
         self.tp_size = tp_size
         self.sp = sp and tp_size > 1
@@ -196,7 +217,8 @@ class LlamaForTraining(nn.Module):
             [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i,
                                ep_group=ep_group, ep_size=ep_size,
                                tp_group=tp_group, tp_size=tp_size,
-                               sp=sp and tp_size > 1)
+                               sp=sp and tp_size > 1, cp_group=cp_group,
+                               cp_size=cp_size)
              for i in range(cfg.layer_num)])
         self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
         if self.sp:
@@ -213,9 +235,10 @@ class LlamaForTraining(nn.Module):
         self.register_buffer("rope_cs", cs, persistent=False)
 
     def forward(self, tokens, labels):
-        # tokens/labels: [B, S] int64
+        # tokens/labels: [B, S] int64 (S = the LOCAL seq slice under CP)
         B, S = tokens.shape
-        pos = (torch.arange(S, device=tokens.device, dtype=torch.int32)
+        pos = (torch.arange(self.cp_rank * S, (self.cp_rank + 1) * S,
+                            device=tokens.device, dtype=torch.int32)
                .repeat(B))
         x = self.embedding(tokens)
         if self.sp:

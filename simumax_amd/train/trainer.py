@@ -37,6 +37,7 @@ class TrainConfig:
     ep_size: int = 1
     tp_size: int = 1
     pp_size: int = 1
+    cp_size: int = 1   # Ulysses context parallel (seq_len = FULL sequence)
     sequence_parallel: bool = False
     zero_state: int = 0   # 1 = ZeRO-1 distributed optimizer (sharded state)
 
@@ -371,6 +372,9 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                   tp_size: int = 1):
     torch.manual_seed(1234)
     assert not (cfg.ep_size > 1 and cfg.tp_size > 1), "trainer: tp XOR ep"
+    if cfg.cp_size > 1:
+        assert cfg.tp_size == 1 and cfg.ep_size == 1 and cfg.pp_size == 1, \
+            "trainer CP composes with pure DP for now"
     if cfg.zero_state == 1:
         assert cfg.tp_size == 1 and cfg.ep_size == 1 and cfg.pp_size == 1, \
             "ZeRO-1 composes with pure DP in the trainer for now"
@@ -382,16 +386,22 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
     edp_size = 1
     if cfg.ep_size > 1 and dist.is_initialized():
         edp_size = dist.get_world_size() // cfg.ep_size
+    from .cp import get_cp_groups
     from .tp import get_tp_groups
 
     tp_group, dp_group, tp_rank = get_tp_groups(tp_size)
+    cp_group, cp_rank = get_cp_groups(cfg.cp_size)
     dp_size = None
     if tp_group is not None:
         dp_size = dist.get_world_size() // tp_size
+    # cp ranks average grads with the dp group (dp_cp): the default
+    # world-spanning reducer already does that when tp = ep = 1
     model = LlamaForTraining(model_cfg, cfg.seq_len, device=device,
                              ep_group=ep_group, ep_size=cfg.ep_size,
                              tp_group=tp_group, tp_size=tp_size,
-                             tp_rank=tp_rank, sp=cfg.sequence_parallel)
+                             tp_rank=tp_rank, sp=cfg.sequence_parallel,
+                             cp_group=cp_group, cp_rank=cp_rank,
+                             cp_size=cfg.cp_size)
     opt = MixedPrecisionAdam(model.parameters(), cfg)
     if tp_group is not None:
         # tp shards are unique; norms/embedding are replicated across tp
