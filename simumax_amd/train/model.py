@@ -26,7 +26,7 @@ class GQAAttention(nn.Module):
 
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
                  tp_group=None, tp_size=1, sp=False, cp_group=None,
-                 cp_size=1):
+                 cp_size=1, fp8=False):
         super().__init__()
         h = cfg.hidden_size
         assert cfg.head_num % tp_size == 0 and cfg.kv_head_num % tp_size == 0
@@ -40,9 +40,12 @@ class GQAAttention(nn.Module):
         self.cp_group = cp_group
         self.sp = sp
         qkv_out = (self.heads + 2 * self.kv_heads) * cfg.head_size
-        self.qkv_proj = K.FusedLinear(h, qkv_out, dtype=dtype, device=device)
-        self.out_proj = K.FusedLinear(self.heads * cfg.head_size, h,
-                                      dtype=dtype, device=device)
+        Lin = K.FusedLinear
+        if fp8:
+            from ..kernels.fp8 import Fp8Linear as Lin
+        self.qkv_proj = Lin(h, qkv_out, dtype=dtype, device=device)
+        self.out_proj = Lin(self.heads * cfg.head_size, h,
+                            dtype=dtype, device=device)
         if tp_size > 1:
             self.qkv_proj.weight._is_tp_shard = True
             self.out_proj.weight._is_tp_shard = True
@@ -132,7 +135,7 @@ class MLAAttention(nn.Module):
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
                  layer_idx=0, ep_group=None, ep_size=1, tp_group=None,
-                 tp_size=1, sp=False, cp_group=None, cp_size=1):
+                 tp_size=1, sp=False, cp_group=None, cp_size=1, fp8=False):
         super().__init__()
         h = cfg.hidden_size
         self.tp_group = tp_group
@@ -146,7 +149,7 @@ class LlamaDecoderLayer(nn.Module):
             self.attention = GQAAttention(cfg, dtype=dtype, device=device,
                                           tp_group=tp_group, tp_size=tp_size,
                                           sp=sp, cp_group=cp_group,
-                                          cp_size=cp_size)
+                                          cp_size=cp_size, fp8=fp8)
         if sp:
             # SP norms see only the local seq shard: their weight grads
             # are partial sums and need a tp all_reduce (reducer handles
@@ -169,8 +172,11 @@ class LlamaDecoderLayer(nn.Module):
             # parallel over I/tp with one fwd all_reduce
             assert cfg.intermediate_size % tp_size == 0
             i_local = cfg.intermediate_size // tp_size
-            self.fc1 = K.FusedLinear(h, 2 * i_local, dtype=dtype, device=device)
-            self.fc2 = K.FusedLinear(i_local, h, dtype=dtype, device=device)
+            Lin = K.FusedLinear
+            if fp8:
+                from ..kernels.fp8 import Fp8Linear as Lin
+            self.fc1 = Lin(h, 2 * i_local, dtype=dtype, device=device)
+            self.fc2 = Lin(i_local, h, dtype=dtype, device=device)
             if tp_size > 1:
                 self.fc1.weight._is_tp_shard = True
                 self.fc2.weight._is_tp_shard = True
@@ -199,7 +205,7 @@ class LlamaForTraining(nn.Module):
     def __init__(self, cfg: ModelConfig, seq_len: int, dtype=torch.bfloat16,
                  rope_base=500000.0, device=None, ep_group=None, ep_size=1,
                  tp_group=None, tp_size=1, tp_rank=0, sp=False,
-                 cp_group=None, cp_rank=0, cp_size=1):
+                 cp_group=None, cp_rank=0, cp_size=1, fp8=False):
         super().__init__()
         self.cfg = cfg
         self.seq_len = seq_len          # FULL sequence (rope cache size)
@@ -218,7 +224,7 @@ class LlamaForTraining(nn.Module):
                                ep_group=ep_group, ep_size=ep_size,
                                tp_group=tp_group, tp_size=tp_size,
                                sp=sp and tp_size > 1, cp_group=cp_group,
-                               cp_size=cp_size)
+                               cp_size=cp_size, fp8=fp8)
              for i in range(cfg.layer_num)])
         self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
         if self.sp:

@@ -38,6 +38,7 @@ class TrainConfig:
     tp_size: int = 1
     pp_size: int = 1
     cp_size: int = 1   # Ulysses context parallel (seq_len = FULL sequence)
+    fp8: bool = False  # e4m3/e5m2 GEMMs via _scaled_mm (decoder linears)
     sequence_parallel: bool = False
     zero_state: int = 0   # 1 = ZeRO-1 distributed optimizer (sharded state)
 
@@ -401,7 +402,7 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                              tp_group=tp_group, tp_size=tp_size,
                              tp_rank=tp_rank, sp=cfg.sequence_parallel,
                              cp_group=cp_group, cp_rank=cp_rank,
-                             cp_size=cfg.cp_size)
+                             cp_size=cfg.cp_size, fp8=cfg.fp8)
     opt = MixedPrecisionAdam(model.parameters(), cfg)
     if tp_group is not None:
         # tp shards are unique; norms/embedding are replicated across tp
