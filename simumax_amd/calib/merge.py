@@ -77,6 +77,13 @@ def main():
             statistics.median(group.values()), 4)
         print(f"group_matmul: {len(group)} shapes")
 
+    fp8g = _load("fp8_group_matmul.json")
+    if fp8g:
+        acc["op"]["fp8_group_matmul"]["accurate_efficient_factor"] = fp8g
+        acc["op"]["fp8_group_matmul"]["efficient_factor"] = round(
+            statistics.median(fp8g.values()), 4)
+        print(f"fp8_group_matmul: {len(fp8g)} shapes")
+
     bw = _load("bandwidth.json")
     bw.update(_load("bandwidth_insitu.json"))
     if bw:

@@ -215,6 +215,47 @@ def sweep_fp8_gemms(shape_keys, path):
     return table
 
 
+def sweep_fp8_grouped(shape_keys, path):
+    """fp8 e4m3 grouped GEMM (per-expert _scaled_mm loop) for the
+    fp8_group_matmul table; keys are the bf16 grouped keys with
+    dtype=fp8. Efficiency vs the 5 PF dense fp8 peak."""
+    from simumax_amd.kernels.fp8 import _quant
+
+    tab = _load(path)
+    for bkey in shape_keys:
+        desc = bkey.replace("dtype=bf16", "dtype=fp8")
+        if desc in tab and not OVERWRITE:
+            continue
+        ng, m, n, k = parse_group_key(bkey)
+        try:
+            xs, ws, sx, sw = [], [], [], []
+            for _ in range(ng):
+                x = torch.randn(m, k, device="cuda", dtype=torch.bfloat16) / 8
+                w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16) / 8
+                xq, xsc = _quant(x, "e4m3")
+                wq, wsc = _quant(w, "e4m3")
+                xs.append(xq); ws.append(wq.t().contiguous().t())
+                sx.append(xsc); sw.append(wsc)
+
+            def fn():
+                for i in range(ng):
+                    torch._scaled_mm(xs[i], ws[i], scale_a=sx[i],
+                                     scale_b=sw[i], out_dtype=torch.bfloat16)
+            t_ms = _timeit(fn, warmup=2, iters=5)
+            flops = 2.0 * ng * m * n * k
+            eff = flops / (t_ms / 1e3) / (2 * PEAK_BF16)
+        except (RuntimeError, torch.cuda.OutOfMemoryError) as e:
+            print(f"[fp8group] skip {desc}: {str(e)[:70]}", flush=True)
+            torch.cuda.empty_cache()
+            continue
+        tab[desc] = round(eff, 4)
+        print(f"[fp8group] {desc[:60]} -> eff {eff:.4f}", flush=True)
+        _save(path, tab)
+        del xs, ws
+        torch.cuda.empty_cache()
+    return tab
+
+
 def parse_group_key(desc):
     m = re.match(r"ng=(\d+), M=(\d+), N=(\d+), K=(\d+), dtype=(\w+)", desc)
     assert m, desc
@@ -438,6 +479,9 @@ def main():
         sweep_grouped(group_keys, os.path.join(OUT_DIR, "group_matmul.json"))
     if which in ("all", "fp8"):
         sweep_fp8_gemms(gemm_keys, os.path.join(OUT_DIR, "fp8_matmul.json"))
+    if which in ("all", "fp8group"):
+        sweep_fp8_grouped(group_keys,
+                          os.path.join(OUT_DIR, "fp8_group_matmul.json"))
     print("[calib] done", flush=True)
 
 
