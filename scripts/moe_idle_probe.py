@@ -76,7 +76,10 @@ def main():
                    idle_per_layer_mb_ms=round(per_lmb, 3))
         print(row, flush=True)
         out[name] = row
+        red.remove_hooks()   # C++-side hook storage pins the whole trainer
         del m, opt, red
+        import gc
+        gc.collect()
         torch.cuda.empty_cache()
     os.makedirs("gpurun_out", exist_ok=True)
     with open("gpurun_out/moe_idle.json", "w") as f:

@@ -223,7 +223,7 @@ def sweep_fp8_grouped(shape_keys, path):
 
     tab = _load(path)
     for bkey in shape_keys:
-        desc = bkey.replace("dtype=bf16", "dtype=fp8")
+        desc = bkey.replace(" dtype=bf16,", " dtype=fp8,")
         if desc in tab and not OVERWRITE:
             continue
         ng, m, n, k = parse_group_key(bkey)
@@ -234,12 +234,14 @@ def sweep_fp8_grouped(shape_keys, path):
                 w = torch.randn(n, k, device="cuda", dtype=torch.bfloat16) / 8
                 xq, xsc = _quant(x, "e4m3")
                 wq, wsc = _quant(w, "e4m3")
-                xs.append(xq); ws.append(wq.t().contiguous().t())
+                # wq row-major [N,K]: wq.t() is [K,N] column-major, the
+                # layout _scaled_mm requires for mat2
+                xs.append(xq); ws.append(wq)
                 sx.append(xsc); sw.append(wsc)
 
             def fn():
                 for i in range(ng):
-                    torch._scaled_mm(xs[i], ws[i], scale_a=sx[i],
+                    torch._scaled_mm(xs[i], ws[i].t(), scale_a=sx[i],
                                      scale_b=sw[i], out_dtype=torch.bfloat16)
             t_ms = _timeit(fn, warmup=2, iters=5)
             flops = 2.0 * ng * m * n * k

@@ -45,8 +45,8 @@ def test_fp8_trainer_step():
 
     if not fp8_available():
         pytest.skip("no _scaled_mm fp8 support")
-    cfg = ModelConfig(hidden_size=512, head_num=8, kv_head_num=4,
-                      head_size=64, intermediate_size=1024, layer_num=2,
+    cfg = ModelConfig(hidden_size=512, head_num=4, kv_head_num=2,
+                      head_size=128, intermediate_size=1024, layer_num=2,
                       vocab_size=2048, use_swiglu=True)
     tc = TrainConfig(seq_len=512, micro_batch_size=2, micro_batch_num=2,
                      fp8=True)
