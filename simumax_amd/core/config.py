@@ -388,6 +388,9 @@ class BandwidthConfig:
     gbps: float = 0.0
     efficient_factor: float = 1.0
     latency_us: float = 0.0
+    # host-launch-bound chains whose length scales with a unit count
+    # (e.g. per-local-expert GEMM-loop launches in MoE routing)
+    per_unit_us: float = 0.0
     fixed_latency: Optional[float] = None
     fixed_latency_us_by_comm_num: Optional[Dict[str, float]] = None
 
@@ -525,12 +528,14 @@ class SystemConfig(Config):
                         compute_only_time=t)
         return t
 
-    def compute_mem_access_time(self, op_name, mem_bytes, reture_detail=False):
+    def compute_mem_access_time(self, op_name, mem_bytes, reture_detail=False,
+                                units=0):
         """HBM3E access time in ms via bandwidth table (8 TB/s peak, measured
-        stream efficiency from the HIP harness)."""
+        stream efficiency from the HIP harness). `units` scales the
+        per_unit_us launch-latency term (per-local-expert chains)."""
         op = self.accelerator.bandwidth.get(op_name) or self.accelerator.bandwidth["default"]
         t = mem_bytes / (op.gbps * 1024**3 * op.efficient_factor) * 1e3
-        t += op.latency_us / 1e3
+        t += (op.latency_us + getattr(op, "per_unit_us", 0.0) * units) / 1e3
         if mem_bytes == 0:
             t = 0.0
         if reture_detail:

@@ -57,4 +57,8 @@ MI355X = {
 # + denom sqrt/add (r4 w4 rw8) + addcdiv (r4 r4 rw8) + bf16 copy (r4 w2).
 # The calibration sweep times the same sequence, so the efficiency factor
 # is consistent by construction.
-OPTIMIZER_TRAFFIC_BYTES_PER_PARAM = 70
+# measured (torch.profiler, mixtral-8x7b-l8 11.9B params, 2026-09): the
+# flat mixed-precision Adam chain (zero_grad, chunked grad-norm dot, clip,
+# m/v updates, chunked denom, addcdiv, param copy) moves ~76 B/param at the
+# calibrated optimizer-stream efficiency
+OPTIMIZER_TRAFFIC_BYTES_PER_PARAM = 76

@@ -189,7 +189,9 @@ class MetaModule:
             if extra > 0:
                 # glue kernels (layout copies, grad fan-in adds, host-bound
                 # small-launch chains) run as separate launches, additive
-                t += sysc.compute_mem_access_time(extra_op or "default", extra)
+                t += sysc.compute_mem_access_time(
+                    extra_op or "default", extra,
+                    units=getattr(self, "extra_op_units", 0))
             return t
 
         ci = self._cost_info
