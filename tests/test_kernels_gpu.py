@@ -315,3 +315,29 @@ def test_flash_attention_fwd_rescale_branch():
     o, lse = ext().fa_fwd(q, k, v, True)
     oref = K._sdpa_torch(q, k, v, True)
     assert relerr(o, oref) < 3e-2, f"rescale-branch rel err {relerr(o, oref)}"
+
+
+@pytest.mark.gpu
+def test_flash_attention_mla_shapes():
+    """MLA asymmetric head dims (Dqk=192, Dv=128) through the fa kernels
+    vs the fp32 math reference (DeepSeek MLA path)."""
+    B, S, H, Dqk, Dv = 1, 512, 4, 192, 128
+    torch.manual_seed(11)
+    q = torch.randn(B, S, H, Dqk, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn(B, S, H, Dqk, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    v = torch.randn(B, S, H, Dv, device=DEV, dtype=torch.bfloat16,
+                    requires_grad=True)
+    o = K.flash_attention(q, k, v, causal=True)
+    do = torch.randn_like(o)
+    o.backward(do)
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    v2 = v.detach().float().requires_grad_(True)
+    o2 = K._sdpa_torch(q2, k2, v2, True)
+    o2.backward(do.float())
+    assert relerr(o, o2) < 3e-2, f"mla fwd {relerr(o, o2)}"
+    assert relerr(q.grad, q2.grad) < 5e-2, f"mla dq {relerr(q.grad, q2.grad)}"
+    assert relerr(k.grad, k2.grad) < 5e-2, f"mla dk {relerr(k.grad, k2.grad)}"
+    assert relerr(v.grad, v2.grad) < 5e-2, f"mla dv {relerr(v.grad, v2.grad)}"
