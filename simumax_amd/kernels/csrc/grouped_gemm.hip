@@ -97,13 +97,26 @@ void gg_fwd_kernel(const bf16raw *__restrict__ A, const bf16raw *__restrict__ W,
 #pragma unroll
         for (int cs = 0; cs < 8; ++cs) acc[rs][cs] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-    for (int k0 = 0; k0 < K; k0 += GG_KSTEP) {
-        __syncthreads();
+    // T14 split staging: tile k0+GG_KSTEP's loads are issued into
+    // registers while tile k0 computes (r1 PMC: the synchronous
+    // stage->barrier->compute loop left fwd 3x more wait-parked than
+    // dgrad at identical MFMA work)
+    bf16x8 st_a[8], st_w[8];
+    auto stage_load = [&](int k0) {
+        if (k0 >= K) return;
+        const int arow = srow < mrem ? srow : 0;   // clamp; zeroed on write
+#pragma unroll
+        for (int c = 0; c < 8; ++c)
+            st_a[c] = load8(Ae + (long)arow * K + k0 + scol + c * 8);
+#pragma unroll
+        for (int c = 0; c < 8; ++c)
+            st_w[c] = load8(We + (long)srow * K + k0 + scol + c * 8);
+    };
+    auto stage_write = [&]() {
         if (srow < mrem) {
 #pragma unroll
             for (int c = 0; c < 8; ++c)
-                store8(a_lds + srow * GG_KPAD + scol + c * 8,
-                       load8(Ae + (long)srow * K + k0 + scol + c * 8));
+                store8(a_lds + srow * GG_KPAD + scol + c * 8, st_a[c]);
         } else {
             const uint4 z{0, 0, 0, 0};
 #pragma unroll
@@ -113,9 +126,13 @@ void gg_fwd_kernel(const bf16raw *__restrict__ A, const bf16raw *__restrict__ W,
         }
 #pragma unroll
         for (int c = 0; c < 8; ++c)
-            store8(w_lds + srow * GG_KPAD + scol + c * 8,
-                   load8(We + (long)srow * K + k0 + scol + c * 8));
-        __syncthreads();
+            store8(w_lds + srow * GG_KPAD + scol + c * 8, st_w[c]);
+    };
+    stage_load(0);
+    stage_write();
+    __syncthreads();
+    for (int k0 = 0; k0 < K; k0 += GG_KSTEP) {
+        stage_load(k0 + GG_KSTEP);
 #pragma unroll
         for (int kc = 0; kc < GG_KSTEP / 32; ++kc) {
             bf16x8v a0 = gg_frag(a_lds + (wave * 32 + (lane & 15)) * GG_KPAD
@@ -132,6 +149,9 @@ void gg_fwd_kernel(const bf16raw *__restrict__ A, const bf16raw *__restrict__ W,
                     a1, b, acc[1][cs], 0, 0, 0);
             }
         }
+        __syncthreads();   // all waves done reading this k tile
+        stage_write();
+        __syncthreads();   // next tile visible
     }
 #pragma unroll
     for (int rs = 0; rs < 2; ++rs)
