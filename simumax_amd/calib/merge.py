@@ -109,16 +109,22 @@ def main():
                 acc["bandwidth"][key]["efficient_factor"] = round(
                     bw[f"{key}_eff"], 4)
         if "moe_routing_ms" in bw:
+            cur = acc["bandwidth"].get("moe_routing", {})
+            # refresh the chain-wall base; the per-local-expert launch
+            # term (scripts/moe_idle_probe fit) rides on top
             acc["bandwidth"]["moe_routing"] = {
                 "gbps": 8000.0,
                 "efficient_factor": 0.55,
                 "latency_us": round(bw["moe_routing_ms"] * 1e3, 1),
+                "per_unit_us": cur.get("per_unit_us", 0.0),
             }
         if "moe_routing_bwd_ms" in bw:
+            cur = acc["bandwidth"].get("moe_routing_bwd", {})
             acc["bandwidth"]["moe_routing_bwd"] = {
                 "gbps": 8000.0,
                 "efficient_factor": 0.55,
                 "latency_us": round(bw["moe_routing_bwd_ms"] * 1e3, 1),
+                "per_unit_us": cur.get("per_unit_us", 0.0),
             }
         if "optimizer_eff" in bw:
             acc["bandwidth"]["optimizer"] = {
