@@ -219,7 +219,7 @@ def sweep_fp8_grouped(shape_keys, path):
     """fp8 e4m3 grouped GEMM (per-expert _scaled_mm loop) for the
     fp8_group_matmul table; keys are the bf16 grouped keys with
     dtype=fp8. Efficiency vs the 5 PF dense fp8 peak."""
-    from simumax_amd.kernels.fp8 import _quant
+    from simumax_amd.kernels.fp8 import _quant_dynamic as _quant
 
     tab = _load(path)
     for bkey in shape_keys:

@@ -18,7 +18,7 @@ CSRC = os.path.join(HERE, "csrc")
 OUT = os.path.join(HERE, "simumax_hip.so")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
-KERNEL_SOURCES = ["rmsnorm.hip", "rope.hip", "swiglu.hip", "cross_entropy.hip",
+KERNEL_SOURCES = ["rmsnorm.hip", "rope.hip", "swiglu.hip", "cross_entropy.hip", "fp8_cast.hip",
                   "attention.hip", "mfma_probe.hip", "gemm_bench.hip",
                   "grouped_gemm.hip"]
 

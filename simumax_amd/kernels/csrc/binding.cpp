@@ -27,6 +27,8 @@ void ce_fwd_launch(const void *, const void *, void *, void *, void *, long,
                    int, hipStream_t);
 void ce_bwd_launch(const void *, const void *, const void *, const void *,
                    const void *, void *, long, int, hipStream_t);
+void fp8_cast_launch(const void *, void *, void *, const void *, long, int,
+                     float, hipStream_t);
 }
 
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
@@ -243,7 +245,23 @@ void grouped_wgrad(torch::Tensor dout, torch::Tensor x, torch::Tensor g) {
              cur_stream());
 }
 
+// fused bf16 -> fp8 cast with delayed-scaling amax update; out must be a
+// float8 tensor of x's shape, amax a zeroed fp32 [1], scale fp32 [1]
+void fp8_cast(torch::Tensor x, torch::Tensor out, torch::Tensor amax,
+              torch::Tensor scale, bool e5m2, double fmax) {
+    CHECK_IN(x);
+    TORCH_CHECK(out.is_cuda() && out.is_contiguous(), "out");
+    TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "x must be bf16");
+    TORCH_CHECK(x.numel() % 8 == 0, "numel % 8");
+    TORCH_CHECK(out.numel() == x.numel(), "out size");
+    fp8_cast_launch(x.data_ptr(), out.data_ptr(), amax.data_ptr(),
+                    scale.data_ptr(), (long)x.numel(), e5m2 ? 1 : 0,
+                    (float)fmax, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("fp8_cast", &fp8_cast,
+          "fused bf16->fp8 cast + running amax (delayed scaling)");
     m.def("wgrad_accum", &wgrad_accum,
           "main_grad(fp32) += dout^T @ x (bf16 in, hipBLAS GemmEx)");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 fragment-layout probe");

@@ -1,16 +1,23 @@
 """fp8 (OCP e4m3/e5m2) linear path for the trainer — gfx950's 5 PF dense
 fp8 MFMA rate through hipBLASLt's _scaled_mm.
 
-TransformerEngine-style recipe, simplified to per-tensor DYNAMIC scaling
-(amax of the current tensor, no history window): activations and weights
-quantize to e4m3 in forward; gradients to e5m2 in backward (wider exponent
-range); all GEMMs accumulate in fp32 and emit bf16. The simulator prices
-this path with the fp8_matmul efficiency table (ops/dense.py
-QuantizedColLinear/QuantizedRowLinear; reference dense_module.py:2365-2453).
+TransformerEngine-style recipe with per-tensor DELAYED scaling: each
+tensor role (activation, gradient) keeps a running amax buffer; the fused
+gfx950 cast kernel (kernels/csrc/fp8_cast.hip) converts bf16 -> fp8 in
+ONE pass using last call's amax as the scale while recording this call's
+amax for the next — a naive dynamic recipe needs 3 passes (amax reduce,
+scale-mul, cast) and measured 41% SLOWER than bf16 end to end. Weights
+re-quantize only when the parameter version changes (once per optimizer
+step, shared by all microbatches), caching both the row-major fp8 weight
+and the column-major copy _scaled_mm wants for dgrad.
+
+Numerics: activations/weights e4m3, gradients e5m2 (wider exponent),
+fp32 accumulation, bf16 GEMM outputs. The simulator prices this path via
+the fp8_matmul efficiency table (ops/dense.py fp8 branch; reference
+dense_module.py:2365-2453).
 
 torch._scaled_mm contract (ROCm/hipBLASLt): A row-major [M,K], B
-column-major [K,N] (i.e. pass w.t() of a row-major [N,K] weight), both fp8,
-per-tensor fp32 scales, out_dtype bf16.
+column-major [K,N], both fp8, per-tensor fp32 scale tensors on device.
 """
 
 from __future__ import annotations
@@ -21,52 +28,80 @@ E4M3_MAX = 448.0
 E5M2_MAX = 57344.0
 
 
-def _quant(t, fmt):
-    """Per-tensor dynamic scaling quantize. Returns (fp8 tensor, descale)."""
+def _ext():
+    from .ops import ext
+
+    return ext()
+
+
+def _quant_dynamic(t, fmt):
+    """Two-pass fallback (CPU tensors / first call priming)."""
     dt = torch.float8_e4m3fn if fmt == "e4m3" else torch.float8_e5m2
     fmax = E4M3_MAX if fmt == "e4m3" else E5M2_MAX
     amax = t.abs().amax().float().clamp(min=1e-12)
     scale = fmax / amax
     q = (t.float() * scale).clamp(-fmax, fmax).to(dt)
-    return q, (1.0 / scale).view(1)
+    return q, (amax / fmax).reshape(1)
+
+
+def _quant_delayed(t, fmt, amax_buf, primed):
+    """One-pass fused cast using last call's amax; records this call's
+    amax into amax_buf. Returns (fp8 tensor, descale [1])."""
+    if not t.is_cuda or t.numel() % 8 != 0:
+        return _quant_dynamic(t, fmt)
+    if not primed[0]:
+        # prime the amax with a one-off reduction so call 1 is well-scaled
+        amax_buf.copy_(t.abs().amax().float().reshape(()))
+        primed[0] = True
+    dt = torch.float8_e4m3fn if fmt == "e4m3" else torch.float8_e5m2
+    fmax = E4M3_MAX if fmt == "e4m3" else E5M2_MAX
+    a_prev = amax_buf.clamp(min=1e-12).reshape(1)
+    scale = fmax / a_prev
+    descale = a_prev / fmax
+    amax_buf.zero_()
+    out = torch.empty(t.shape, dtype=dt, device=t.device)
+    _ext().fp8_cast(t, out, amax_buf, scale, fmt == "e5m2", fmax)
+    return out, descale
 
 
 class _Fp8LinearFn(torch.autograd.Function):
     """y = x @ w^T with all three GEMMs (fwd / dgrad / wgrad) in fp8."""
 
     @staticmethod
-    def forward(ctx, x, weight):
+    def forward(ctx, x, weight, mod):
         ishape = x.shape
-        x2 = x.reshape(-1, ishape[-1])
-        xq, xs = _quant(x2, "e4m3")
-        wq, ws = _quant(weight, "e4m3")
+        x2 = x.reshape(-1, ishape[-1]).contiguous()
+        xq, xs = _quant_delayed(x2, "e4m3", mod.x_amax, mod._x_primed)
+        wq, ws, w_cm = mod._weight_quant()
         y = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
                              out_dtype=torch.bfloat16)
-        ctx.save_for_backward(xq, xs, wq, ws)
+        ctx.save_for_backward(xq, xs, ws, w_cm)
+        ctx.mod = mod
         ctx.ishape = ishape
         return y.reshape(*ishape[:-1], weight.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
-        xq, xs, wq, ws = ctx.saved_tensors
-        dy2 = dy.reshape(-1, dy.shape[-1])
-        gq, gs = _quant(dy2, "e5m2")
-        # _scaled_mm wants mat2 COLUMN-major: materialize the transposed
-        # copies (TE keeps a cached transposed weight the same way)
-        # dgrad: dx[M,K] = dy[M,N] @ w[N,K]
-        w_cm = wq.t().contiguous().t()
+        xq, xs, ws, w_cm = ctx.saved_tensors
+        mod = ctx.mod
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        gq, gs = _quant_delayed(dy2, "e5m2", mod.g_amax, mod._g_primed)
+        # dgrad: dx[M,K] = dy[M,N] @ w[N,K] (w_cm is the cached column-major
+        # fp8 weight)
         dx = torch._scaled_mm(gq, w_cm, scale_a=gs, scale_b=ws,
                               out_dtype=torch.bfloat16)
-        # wgrad: dw[N,K] = dy^T[N,M] @ x[M,K]
+        # wgrad: dw[N,K] = dy^T[N,M] @ x[M,K]; fp8 transposes are cheap
+        # byte copies (half the bf16 traffic)
         gqt = gq.t().contiguous()
         x_cm = xq.t().contiguous().t()
         dw = torch._scaled_mm(gqt, x_cm, scale_a=gs, scale_b=xs,
                               out_dtype=torch.float32)
-        return dx.reshape(ctx.ishape), dw
+        return dx.reshape(ctx.ishape), dw, None
 
 
 class Fp8Linear(torch.nn.Module):
-    """Drop-in fp8 linear (bias-free, Megatron-style)."""
+    """Drop-in fp8 linear (bias-free, Megatron-style) with delayed
+    scaling and per-step weight-quant caching."""
 
     def __init__(self, in_features, out_features, dtype=torch.bfloat16,
                  device=None):
@@ -74,9 +109,27 @@ class Fp8Linear(torch.nn.Module):
         w = torch.empty(out_features, in_features, dtype=dtype, device=device)
         torch.nn.init.normal_(w, std=0.02)
         self.weight = torch.nn.Parameter(w)
+        self.register_buffer("x_amax", torch.zeros((), dtype=torch.float32,
+                                                   device=device),
+                             persistent=False)
+        self.register_buffer("g_amax", torch.zeros((), dtype=torch.float32,
+                                                   device=device),
+                             persistent=False)
+        self._x_primed = [False]
+        self._g_primed = [False]
+        self._wcache = None          # (version, wq, ws, w_cm)
+
+    def _weight_quant(self):
+        ver = self.weight._version
+        if self._wcache is not None and self._wcache[0] == ver:
+            return self._wcache[1], self._wcache[2], self._wcache[3]
+        wq, ws = _quant_dynamic(self.weight.detach(), "e4m3")
+        w_cm = wq.t().contiguous().t()
+        self._wcache = (ver, wq, ws, w_cm)
+        return wq, ws, w_cm
 
     def forward(self, x):
-        return _Fp8LinearFn.apply(x, self.weight)
+        return _Fp8LinearFn.apply(x, self.weight, self)
 
 
 def fp8_available():
@@ -85,8 +138,8 @@ def fp8_available():
             return False
         a = torch.randn(16, 32, device="cuda", dtype=torch.bfloat16)
         b = torch.randn(16, 32, device="cuda", dtype=torch.bfloat16)
-        aq, asc = _quant(a, "e4m3")
-        bq, bsc = _quant(b, "e4m3")
+        aq, asc = _quant_dynamic(a, "e4m3")
+        bq, bsc = _quant_dynamic(b, "e4m3")
         torch._scaled_mm(aq, bq.t(), scale_a=asc, scale_b=bsc,
                          out_dtype=torch.bfloat16)
         return True
