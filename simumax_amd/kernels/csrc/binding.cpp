@@ -29,6 +29,8 @@ void ce_bwd_launch(const void *, const void *, const void *, const void *,
                    const void *, void *, long, int, hipStream_t);
 void fp8_cast_launch(const void *, void *, void *, const void *, long, int,
                      float, hipStream_t);
+void fp8_cast_t_launch(const void *, void *, void *, void *, const void *,
+                       int, int, int, float, hipStream_t);
 }
 
 std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
@@ -259,9 +261,26 @@ void fp8_cast(torch::Tensor x, torch::Tensor out, torch::Tensor amax,
                     (float)fmax, cur_stream());
 }
 
+// cast + transpose in one pass: out [M,N] fp8 + out_t [N,M] fp8
+void fp8_cast_t(torch::Tensor x, torch::Tensor out, torch::Tensor out_t,
+                torch::Tensor amax, torch::Tensor scale, bool e5m2,
+                double fmax) {
+    CHECK_IN(x);
+    TORCH_CHECK(x.dim() == 2 && x.scalar_type() == torch::kBFloat16, "x");
+    const int M = x.size(0), N = x.size(1);
+    TORCH_CHECK(M % 64 == 0 && N % 64 == 0, "dims must be multiples of 64");
+    TORCH_CHECK(out.numel() == x.numel() && out_t.numel() == x.numel(),
+                "out sizes");
+    fp8_cast_t_launch(x.data_ptr(), out.data_ptr(), out_t.data_ptr(),
+                      amax.data_ptr(), scale.data_ptr(), M, N,
+                      e5m2 ? 1 : 0, (float)fmax, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("fp8_cast", &fp8_cast,
           "fused bf16->fp8 cast + running amax (delayed scaling)");
+    m.def("fp8_cast_t", &fp8_cast_t,
+          "fused bf16->fp8 cast + transpose + running amax");
     m.def("wgrad_accum", &wgrad_accum,
           "main_grad(fp32) += dout^T @ x (bf16 in, hipBLAS GemmEx)");
     m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 fragment-layout probe");

@@ -57,3 +57,81 @@ extern "C" void fp8_cast_launch(const void *x, void *out, void *amax,
                        0, stream, (const bf16raw *)x, (unsigned char *)out,
                        (float *)amax, (const float *)scale, n, e5m2, fmax);
 }
+
+// cast + TRANSPOSE in one pass (TE cast_transpose): emits the row-major
+// fp8 image AND the [N,M] transposed image (wgrad's A / dgrad's B want
+// the other layout; a torch .t().contiguous() on fp8 bytes runs at
+// ~0.5 TB/s uncoalesced and costs more than the GEMM it feeds).
+// 64x64 tiles staged through LDS at byte granularity.
+#define CT_TILE 64
+#define CT_PAD 72
+
+__global__ __launch_bounds__(CAST_BLOCK)
+void fp8_cast_t_kernel(const bf16raw *__restrict__ x,
+                       unsigned char *__restrict__ out,
+                       unsigned char *__restrict__ out_t,
+                       float *__restrict__ amax,
+                       const float *__restrict__ scale_p,
+                       int M, int N, int e5m2, float fmax) {
+    __shared__ unsigned char tile[CT_TILE * CT_PAD];
+    const float scale = *scale_p;
+    const int tiles_n = N / CT_TILE;
+    float local = 0.f;
+    for (int t = blockIdx.x; t < (M / CT_TILE) * tiles_n; t += gridDim.x) {
+        const int tm = (t / tiles_n) * CT_TILE;
+        const int tn = (t % tiles_n) * CT_TILE;
+        __syncthreads();   // previous tile fully read
+#pragma unroll
+        for (int it = 0; it < 2; ++it) {
+            const int e = (threadIdx.x + it * CAST_BLOCK) * 8;
+            const int r = e / CT_TILE, c = e % CT_TILE;
+            bf16x8 v = load8(x + (long)(tm + r) * N + tn + c);
+            unsigned char o[8];
+#pragma unroll
+            for (int p = 0; p < 4; ++p) {
+                float a = v.get(2 * p), b = v.get(2 * p + 1);
+                local = fmaxf(local, fmaxf(fabsf(a), fabsf(b)));
+                float as = fminf(fmaxf(a * scale, -fmax), fmax);
+                float bs = fminf(fmaxf(b * scale, -fmax), fmax);
+                int packed;
+                if (e5m2)
+                    packed = __builtin_amdgcn_cvt_pk_bf8_f32(as, bs, 0, false);
+                else
+                    packed = __builtin_amdgcn_cvt_pk_fp8_f32(as, bs, 0, false);
+                o[2 * p] = packed & 0xff;
+                o[2 * p + 1] = (packed >> 8) & 0xff;
+            }
+            *reinterpret_cast<uint2 *>(out + (long)(tm + r) * N + tn + c) =
+                *reinterpret_cast<uint2 *>(o);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) tile[r * CT_PAD + c + j] = o[j];
+        }
+        __syncthreads();
+        // transposed write: thread covers out_t[tn + r'][tm + c'..+8]
+#pragma unroll
+        for (int it = 0; it < 2; ++it) {
+            const int e = (threadIdx.x + it * CAST_BLOCK) * 8;
+            const int r2 = e / CT_TILE, c2 = e % CT_TILE;
+            unsigned char o[8];
+#pragma unroll
+            for (int j = 0; j < 8; ++j) o[j] = tile[(c2 + j) * CT_PAD + r2];
+            *reinterpret_cast<uint2 *>(out_t + (long)(tn + r2) * M + tm + c2)
+                = *reinterpret_cast<uint2 *>(o);
+        }
+    }
+    local = wave_max(local);
+    if ((threadIdx.x % WAVE) == 0 && local > 0.f)
+        atomicMax(reinterpret_cast<int *>(amax), __float_as_int(local));
+}
+
+extern "C" void fp8_cast_t_launch(const void *x, void *out, void *out_t,
+                                  void *amax, const void *scale, int M, int N,
+                                  int e5m2, float fmax, hipStream_t stream) {
+    int tiles = (M / CT_TILE) * (N / CT_TILE);
+    int blocks = tiles < 2048 ? tiles : 2048;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(fp8_cast_t_kernel, dim3(blocks), dim3(CAST_BLOCK),
+                       0, stream, (const bf16raw *)x, (unsigned char *)out,
+                       (unsigned char *)out_t, (float *)amax,
+                       (const float *)scale, M, N, e5m2, fmax);
+}

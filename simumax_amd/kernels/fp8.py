@@ -64,6 +64,28 @@ def _quant_delayed(t, fmt, amax_buf, primed):
     return out, descale
 
 
+def _quant_delayed_t(t, fmt, amax_buf, primed):
+    """Cast + transpose in one fused pass: returns (q [M,N], q_t [N,M],
+    descale). Falls back to cast + torch transpose off the fast path."""
+    if (not t.is_cuda or t.dim() != 2 or t.size(0) % 64 or t.size(1) % 64):
+        q, d = _quant_delayed(t, fmt, amax_buf, primed) if t.is_cuda else \
+            _quant_dynamic(t, fmt)
+        return q, q.t().contiguous(), d
+    if not primed[0]:
+        amax_buf.copy_(t.abs().amax().float().reshape(()))
+        primed[0] = True
+    dt = torch.float8_e4m3fn if fmt == "e4m3" else torch.float8_e5m2
+    fmax = E4M3_MAX if fmt == "e4m3" else E5M2_MAX
+    a_prev = amax_buf.clamp(min=1e-12).reshape(1)
+    scale = fmax / a_prev
+    descale = a_prev / fmax
+    amax_buf.zero_()
+    q = torch.empty(t.shape, dtype=dt, device=t.device)
+    qt = torch.empty((t.size(1), t.size(0)), dtype=dt, device=t.device)
+    _ext().fp8_cast_t(t, q, qt, amax_buf, scale, fmt == "e5m2", fmax)
+    return q, qt, descale
+
+
 class _Fp8LinearFn(torch.autograd.Function):
     """y = x @ w^T with all three GEMMs (fwd / dgrad / wgrad) in fp8."""
 
@@ -71,31 +93,33 @@ class _Fp8LinearFn(torch.autograd.Function):
     def forward(ctx, x, weight, mod):
         ishape = x.shape
         x2 = x.reshape(-1, ishape[-1]).contiguous()
-        xq, xs = _quant_delayed(x2, "e4m3", mod.x_amax, mod._x_primed)
+        # cast_transpose: the transposed image is wgrad's column-major B
+        # operand (a torch fp8 .t().contiguous() costs more than the GEMM)
+        xq, xq_t, xs = _quant_delayed_t(x2, "e4m3", mod.x_amax, mod._x_primed)
         wq, ws, w_cm = mod._weight_quant()
         y = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
                              out_dtype=torch.bfloat16)
-        ctx.save_for_backward(xq, xs, ws, w_cm)
+        ctx.save_for_backward(xq_t, xs, ws, w_cm)
         ctx.mod = mod
         ctx.ishape = ishape
         return y.reshape(*ishape[:-1], weight.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
-        xq, xs, ws, w_cm = ctx.saved_tensors
+        xq_t, xs, ws, w_cm = ctx.saved_tensors
         mod = ctx.mod
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
-        gq, gs = _quant_delayed(dy2, "e5m2", mod.g_amax, mod._g_primed)
-        # dgrad: dx[M,K] = dy[M,N] @ w[N,K] (w_cm is the cached column-major
-        # fp8 weight)
+        gq, gqt, gs = _quant_delayed_t(dy2, "e5m2", mod.g_amax,
+                                       mod._g_primed)
+        # dgrad: dx[M,K] = dy[M,N] @ w[N,K] (w_cm: cached col-major weight)
         dx = torch._scaled_mm(gq, w_cm, scale_a=gs, scale_b=ws,
                               out_dtype=torch.bfloat16)
-        # wgrad: dw[N,K] = dy^T[N,M] @ x[M,K]; fp8 transposes are cheap
-        # byte copies (half the bf16 traffic)
-        gqt = gq.t().contiguous()
-        x_cm = xq.t().contiguous().t()
-        dw = torch._scaled_mm(gqt, x_cm, scale_a=gs, scale_b=xs,
-                              out_dtype=torch.float32)
+        # wgrad: dw[N,K] = dy^T[N,M] @ x[M,K]; both operands come from the
+        # fused cast_transpose (gqt row-major [N,M]; xq_t.t() col-major
+        # [M,K]). bf16 out — the post-accumulate hook adds into the fp32
+        # main_grad (Megatron grad_reduce_in_bf16 semantics)
+        dw = torch._scaled_mm(gqt, xq_t.t(), scale_a=gs, scale_b=xs,
+                              out_dtype=torch.bfloat16)
         return dx.reshape(ctx.ishape), dw, None
 
 
