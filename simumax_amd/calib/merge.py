@@ -63,6 +63,7 @@ def main():
             print(f"{key}: {len(tab)} shapes, median eff "
                   f"{acc['op'][key]['efficient_factor']}")
     fp8 = _load("fp8_matmul.json")
+    fp8 = _overlay_insitu(fp8, "fp8_matmul_insitu.json")
     if fp8:
         acc["op"]["fp8_matmul"]["accurate_efficient_factor"] = fp8
         acc["op"]["fp8_matmul"]["efficient_factor"] = round(

@@ -24,6 +24,8 @@ from __future__ import annotations
 
 import torch
 
+from . import insitu
+
 E4M3_MAX = 448.0
 E5M2_MAX = 57344.0
 
@@ -97,11 +99,22 @@ class _Fp8LinearFn(torch.autograd.Function):
         # operand (a torch fp8 .t().contiguous() costs more than the GEMM)
         xq, xq_t, xs = _quant_delayed_t(x2, "e4m3", mod.x_amax, mod._x_primed)
         wq, ws, w_cm = mod._weight_quant()
-        y = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
-                             out_dtype=torch.bfloat16)
+        if insitu.ENABLED and x.is_cuda:
+            b, m = (x.shape[0], x.shape[1]) if x.ndim == 3 else (1, x.shape[0])
+            stop = insitu.start("fp8_matmul", insitu.gemm_key(
+                b, m, x.shape[-1], weight.shape[0], "TN", False, "bf16"))
+            y = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
+                                 out_dtype=torch.bfloat16)
+            stop()
+        else:
+            y = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
+                                 out_dtype=torch.bfloat16)
         ctx.save_for_backward(xq_t, xs, ws, w_cm)
         ctx.mod = mod
         ctx.ishape = ishape
+        # simulator shape-key (b, m) convention: 3D tensors key as (B, S)
+        ctx.bm = ((x.shape[0], x.shape[1]) if x.ndim == 3
+                  else (1, x.shape[0]))
         return y.reshape(*ishape[:-1], weight.shape[0])
 
     @staticmethod
@@ -111,15 +124,29 @@ class _Fp8LinearFn(torch.autograd.Function):
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         gq, gqt, gs = _quant_delayed_t(dy2, "e5m2", mod.g_amax,
                                        mod._g_primed)
+        timing = insitu.ENABLED and dy.is_cuda
+        M, N = gq.shape
+        K = w_cm.shape[1]
+        kb, km = ctx.bm
+        if timing:
+            stop = insitu.start("fp8_matmul", insitu.gemm_key(
+                kb, km, N, K, "NN", False, "bf16"))
         # dgrad: dx[M,K] = dy[M,N] @ w[N,K] (w_cm: cached col-major weight)
         dx = torch._scaled_mm(gq, w_cm, scale_a=gs, scale_b=ws,
                               out_dtype=torch.bfloat16)
+        if timing:
+            stop()
         # wgrad: dw[N,K] = dy^T[N,M] @ x[M,K]; both operands come from the
         # fused cast_transpose (gqt row-major [N,M]; xq_t.t() col-major
         # [M,K]). bf16 out — the post-accumulate hook adds into the fp32
         # main_grad (Megatron grad_reduce_in_bf16 semantics)
+        if timing:
+            stop = insitu.start("fp8_matmul", insitu.gemm_key(
+                1, N, M, K, "NT", True, "fp32"))  # wgrad keys flatten (b,m)
         dw = torch._scaled_mm(gqt, xq_t.t(), scale_a=gs, scale_b=xs,
                               out_dtype=torch.bfloat16)
+        if timing:
+            stop()
         return dx.reshape(ctx.ishape), dw, None
 
 

@@ -112,6 +112,8 @@ def summarize():
         row = dict(t_ms=round(t, 5), n=len(ts))
         if table == "matmul":
             row["eff"] = _gemm_flops(key) / (t / 1e3) / (PEAK_BF16_TFLOPS * 1e12)
+        elif table == "fp8_matmul":
+            row["eff"] = _gemm_flops(key) / (t / 1e3) / (2 * PEAK_BF16_TFLOPS * 1e12)
         elif table in ("sdp_fwd", "sdp_bwd"):
             stage = "fwd" if table == "sdp_fwd" else "bwd"
             row["eff"] = _sdp_flops(key, stage) / (t / 1e3) / (PEAK_BF16_TFLOPS * 1e12)

@@ -189,6 +189,18 @@ class LinearCol(LinearBase, ParamMixin):
         info.fwd_accessed_mem = in_b + w_b + out_b
         info.bwd_grad_act_accessed_mem = out_b + w_b + in_b
         info.bwd_grad_w_accessed_mem = out_b + in_b + k["K"] * k["N"] * self.grad_element_size
+        if self.strategy.fp8:
+            # trainer fp8 path (kernels/fp8.py): fused cast_transpose of
+            # the input (fwd) and of dy (bwd) — read bf16, write BOTH fp8
+            # images = 2x the bf16 bytes each; wgrad lands in a bf16
+            # p.grad that the hook adds into the fp32 main_grad
+            # (2+4+4 B/elem); the per-step weight quant amortizes over
+            # the microbatches
+            nw = k["K"] * k["N"]
+            mbc = max(1, self.strategy.micro_batch_num)
+            info.fwd_extra_mem += 2 * in_b + 2 * w_b // mbc
+            info.bwd_grad_act_extra_mem += 2 * out_b
+            info.bwd_grad_w_extra_mem += 10 * nw
 
     def _leaf_intra_net_info(self):
         if self.tp <= 1:
@@ -260,6 +272,18 @@ class LinearRow(LinearBase, ParamMixin):
         info.fwd_accessed_mem = in_b + w_b + out_b
         info.bwd_grad_act_accessed_mem = out_b + w_b + in_b
         info.bwd_grad_w_accessed_mem = out_b + in_b + k["K"] * k["N"] * self.grad_element_size
+        if self.strategy.fp8:
+            # trainer fp8 path (kernels/fp8.py): fused cast_transpose of
+            # the input (fwd) and of dy (bwd) — read bf16, write BOTH fp8
+            # images = 2x the bf16 bytes each; wgrad lands in a bf16
+            # p.grad that the hook adds into the fp32 main_grad
+            # (2+4+4 B/elem); the per-step weight quant amortizes over
+            # the microbatches
+            nw = k["K"] * k["N"]
+            mbc = max(1, self.strategy.micro_batch_num)
+            info.fwd_extra_mem += 2 * in_b + 2 * w_b // mbc
+            info.bwd_grad_act_extra_mem += 2 * out_b
+            info.bwd_grad_w_extra_mem += 10 * nw
 
     def _leaf_intra_net_info(self):
         if self.tp <= 1:
