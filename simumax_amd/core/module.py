@@ -209,7 +209,9 @@ class MetaModule:
         ci.bwd_grad_w_time = stage_time(self.bwd_w_op, "bwd_grad_w",
                                         comp.bwd_grad_w_flops,
                                         comp.bwd_grad_w_accessed_mem,
-                                        comp.bwd_grad_w_extra_mem)
+                                        comp.bwd_grad_w_extra_mem,
+                                        getattr(self, "bwd_w_mem_op", None),
+                                        getattr(self, "bwd_w_extra_op", None))
         self._price_comm()
         if self.enable_recompute and not self.is_variance_node:
             ci.recompute_compute_time = ci.fwd_compute_time

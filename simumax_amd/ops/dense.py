@@ -141,6 +141,9 @@ class LinearCol(LinearBase, ParamMixin):
         self.sp = (strategy.enable_sequence_parallel and self.tp > 1) if sp_gather is None else sp_gather
         if strategy.fp8:
             self.fwd_op = self.bwd_act_op = self.bwd_w_op = "fp8_matmul"
+            self.fwd_extra_op = "fp8_quant"
+            self.bwd_act_extra_op = "fp8_quant"
+            self.bwd_w_extra_op = "fp8_quant"
 
     @property
     def micro_input_tensor(self):
@@ -163,10 +166,22 @@ class LinearCol(LinearBase, ParamMixin):
 
     def _leaf_model_info(self, info):
         self.add_param(info, self.input_size * self.output_size, self.is_expert)
+        if self.strategy.fp8:
+            # per-step weight-quant cache: wq + column-major copy, 1 B/elem
+            # each (kernels/fp8.py Fp8Linear._weight_quant)
+            nw = self.input_size * self.output_size
+            if self.is_expert:
+                info.moe_weight_bytes += 2 * nw
+            else:
+                info.dense_weight_bytes += 2 * nw
 
     def _leaf_act_info(self, info):
         # caches the (sharded under SP) input; the gathered copy is transient
         info.activation_mem_cache = self.input_info.first.mem_bytes()
+        if self.strategy.fp8:
+            # the fp8 path saves the 1 B/elem transposed fp8 input instead
+            # of the bf16 input
+            info.activation_mem_cache //= 2
         # bwd transient: the freshly allocated placeholder wgrad that
         # AccumulateGrad steals (freed by the post-accumulate hook)
         info.bwd_peak_mem_no_cache = (
@@ -237,6 +252,9 @@ class LinearRow(LinearBase, ParamMixin):
         self.sp = (strategy.enable_sequence_parallel and self.tp > 1) if sp_scatter is None else sp_scatter
         if strategy.fp8:
             self.fwd_op = self.bwd_act_op = self.bwd_w_op = "fp8_matmul"
+            self.fwd_extra_op = "fp8_quant"
+            self.bwd_act_extra_op = "fp8_quant"
+            self.bwd_w_extra_op = "fp8_quant"
 
     def create_output_info(self, input_info):
         t = input_info.tensors[0]
@@ -249,9 +267,17 @@ class LinearRow(LinearBase, ParamMixin):
 
     def _leaf_model_info(self, info):
         self.add_param(info, self.input_size * self.output_size, self.is_expert)
+        if self.strategy.fp8:
+            nw = self.input_size * self.output_size
+            if self.is_expert:
+                info.moe_weight_bytes += 2 * nw
+            else:
+                info.dense_weight_bytes += 2 * nw
 
     def _leaf_act_info(self, info):
         info.activation_mem_cache = self.input_info.first.mem_bytes()
+        if self.strategy.fp8:
+            info.activation_mem_cache //= 2
         info.bwd_peak_mem_no_cache = (
             self.input_size * self.output_size * self.element_size)
         # full (pre-scatter) output is transient under SP
