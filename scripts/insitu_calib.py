@@ -36,6 +36,7 @@ CASES = [
     ("llama3-8b", 16384, 1, 1),
     ("qwen3-32b-l12", 4096, 1, 2),
     ("deepseekv2-l4", 4096, 1, 2),
+    ("llama3-8b", 4096, 1, 2, True),   # fp8 decoder linears
 ]
 
 OUTDIR = "gpurun_out/calib"
@@ -95,11 +96,13 @@ def measure_routing_chain(device="cuda:0"):
 
 def main():
     opt_samples = []
-    for model, seq, mbs, mbc in sys.argv[1:] and [
+    for case in sys.argv[1:] and [
             c for c in CASES if c[0] in sys.argv[1:]] or CASES:
+        model, seq, mbs, mbc = case[:4]
+        fp8 = len(case) > 4 and case[4]
         mc = ModelConfig.init_from_config_file(get_simu_model_config(model))
         tc = TrainConfig(seq_len=seq, micro_batch_size=mbs,
-                         micro_batch_num=mbc)
+                         micro_batch_num=mbc, fp8=fp8)
         m, opt, red = build_trainer(mc, tc, "cuda:0")
         toks, labels = make_synthetic_batch(mc.vocab_size, mbc, mbs, seq,
                                             "cuda:0")

@@ -30,12 +30,13 @@ CASES = [
     ("8b_seq16384_mbc1", "llama3-8b", 16384, 1, 1, 3),
     ("qwen32b_l12_seq4096", "qwen3-32b-l12", 4096, 1, 2, 3),
     ("deepseekv2_l4_mla_moe", "deepseekv2-l4", 4096, 1, 2, 3),
+    ("8b_fp8_seq4096_mbc4", "llama3-8b", 4096, 1, 4, 3, True),
 ]
 
 OUT = "gpurun_out/validation.jsonl"
 
 
-def predict(model_cfg, seq, mbs, mbc):
+def predict(model_cfg, seq, mbs, mbc, fp8=False):
     import copy
 
     from simumax_amd import (PerfLLM, StrategyConfig, SystemConfig,
@@ -43,7 +44,7 @@ def predict(model_cfg, seq, mbs, mbc):
 
     st = StrategyConfig(
         seq_len=seq, micro_batch_size=mbs, micro_batch_num=mbc,
-        world_size=1, tp_size=1, pp_size=1,
+        world_size=1, tp_size=1, pp_size=1, fp8=fp8,
         enable_sequence_parallel=False, zero_state=0,
         use_fp32_accum_grad=True, enable_recompute=False,
         cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
@@ -56,9 +57,10 @@ def predict(model_cfg, seq, mbs, mbc):
     return p.analysis_cost(), p.analysis_mem()
 
 
-def run_case(name, model, seq, mbs, mbc, steps, warmup=1):
+def run_case(name, model, seq, mbs, mbc, steps, fp8=False, warmup=1):
     mc = ModelConfig.init_from_config_file(get_simu_model_config(model))
-    tc = TrainConfig(seq_len=seq, micro_batch_size=mbs, micro_batch_num=mbc)
+    tc = TrainConfig(seq_len=seq, micro_batch_size=mbs, micro_batch_num=mbc,
+                     fp8=fp8)
     t0 = time.time()
     m, opt, red = build_trainer(mc, tc, "cuda:0")
     toks, labels = make_synthetic_batch(mc.vocab_size, mbc, mbs, seq, "cuda:0")
@@ -72,7 +74,7 @@ def run_case(name, model, seq, mbs, mbc, steps, warmup=1):
     torch.cuda.synchronize()
     ms = (time.time() - t1) / steps * 1e3
     peak = torch.cuda.max_memory_allocated()
-    cost, mem = predict(mc, seq, mbs, mbc)
+    cost, mem = predict(mc, seq, mbs, mbc, fp8=fp8)
     row = dict(
         case=name, model=model, seq=seq, mbs=mbs, mbc=mbc,
         measured_ms=round(ms, 2), predicted_ms=round(cost["iter_time"], 2),
@@ -80,6 +82,7 @@ def run_case(name, model, seq, mbs, mbc, steps, warmup=1):
         measured_gib=round(peak / 2**30, 2),
         predicted_gib=round(mem["max_peak_mem"] / 2**30, 2),
         mem_err_pct=round((mem["max_peak_mem"] - peak) / peak * 100, 2),
+        fp8=fp8,
         measured_mfu=round(
             mc.flops_per_token(seq) * mbs * mbc * seq / (ms / 1e3) / 2.5e15, 4),
         predicted_mfu=round(cost["mfu"], 4),
