@@ -28,6 +28,78 @@ import torch.distributed as dist
 import torch.nn as nn
 
 
+
+
+class _PermuteGlue(nn.Module):
+    """Dispatch/combine index glue as a module so in-situ calibration can
+    event-time BOTH directions via backward hooks; bytes follow the
+    simulator's Permutation/UnPermutation accessed-mem conventions
+    (tests/test_glue_model.py::test_permutation_traffic_formulas)."""
+
+    def __init__(self, kind):
+        super().__init__()
+        self.kind = kind          # "dispatch" | "combine"
+        self._h = []
+
+    def _bytes(self, in_b, out_b, topk, stage):
+        if self.kind == "dispatch":
+            if stage == "fwd":
+                return 2 * out_b + 3 * max(in_b * topk, out_b)
+            return 2 * out_b + 3 * in_b
+        if stage == "fwd":
+            return 4 * out_b + 3 * in_b
+        return 6 * out_b + 2 * in_b
+
+    def arm_insitu(self, in_b, out_b, topk):
+        """Register timing hooks once (idempotent)."""
+        from ..kernels import insitu
+
+        if self._h or not insitu.ENABLED:
+            return
+        state = {}
+
+        def fpre(m, inp):
+            state["f"] = insitu.start(
+                "bw_permute_fwd", str(self._bytes(in_b, out_b, topk, "fwd")))
+
+        def fpost(m, inp, out):
+            state["f"]()
+
+        def bpre(m, gout):
+            state["b"] = insitu.start(
+                "bw_permute_bwd", str(self._bytes(in_b, out_b, topk, "bwd")))
+
+        def bpost(m, gin, gout):
+            state["b"]()
+
+        self._h = [self.register_forward_pre_hook(fpre),
+                   self.register_forward_hook(fpost),
+                   self.register_full_backward_pre_hook(bpre),
+                   self.register_full_backward_hook(bpost)]
+
+
+class _DispatchGlue(_PermuteGlue):
+    def __init__(self):
+        super().__init__("dispatch")
+
+    def forward(self, xf, src_tok, slot_index, total_slots):
+        xp = torch.zeros(total_slots, xf.shape[1], dtype=xf.dtype,
+                         device=xf.device)
+        xp.index_copy_(0, slot_index, xf.index_select(0, src_tok))
+        return xp
+
+
+class _CombineGlue(_PermuteGlue):
+    def __init__(self):
+        super().__init__("combine")
+
+    def forward(self, y_flat, src_tok, slot_index, w_kept, n_tokens):
+        out = y_flat.new_zeros(n_tokens, y_flat.shape[1])
+        out.index_add_(0, src_tok,
+                       y_flat.index_select(0, slot_index) * w_kept[:, None])
+        return out
+
+
 class _AllToAll(torch.autograd.Function):
     """Equal-split all_to_all_single with autograd (backward = a2a of the
     incoming grads — exact adjoint for symmetric splits)."""
@@ -185,6 +257,8 @@ class MoEMLP(nn.Module):
         self.ep = ep_size
         assert self.E % self.ep == 0
         self.le = self.E // self.ep          # local experts
+        self.dispatch_glue = _DispatchGlue()
+        self.combine_glue = _CombineGlue()
         self.router = nn.Linear(h, self.E, bias=False, dtype=dtype,
                                 device=device)
         # LOCAL expert weights in the natural Linear layout [le, out, in]
@@ -240,8 +314,10 @@ class MoEMLP(nn.Module):
 
         # dispatch: [E*cap, H] padded buffer, ordered by GLOBAL expert
         # (Permutation permute1)
-        xp = torch.zeros(self.E * cap, H, dtype=x.dtype, device=x.device)
-        xp.index_copy_(0, slot_index, xf.index_select(0, src_tok))
+        in_b = N * H * xf.element_size()
+        out_b = self.E * cap * H * xf.element_size()
+        self.dispatch_glue.arm_insitu(in_b, out_b, self.topk)
+        xp = self.dispatch_glue(xf, src_tok, slot_index, self.E * cap)
 
         if self.ep > 1:
             # EP dispatch a2a: the block for dest rank d = experts
@@ -266,9 +342,8 @@ class MoEMLP(nn.Module):
             y_flat = all_to_all(yb, self.ep_group)           # [E*cap, H]
         else:
             y_flat = y.reshape(self.E * cap, H)
-        out = torch.zeros_like(xf)
-        out.index_add_(0, src_tok,
-                       y_flat.index_select(0, slot_index) * w_kept[:, None])
+        self.combine_glue.arm_insitu(in_b, out_b, self.topk)
+        out = self.combine_glue(y_flat, src_tok, slot_index, w_kept, N)
         out = out.view(B, S, H)
         if self.shared:
             out = out + self.shared_fc2(
