@@ -377,8 +377,8 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
         assert cfg.tp_size == 1 and cfg.ep_size == 1 and cfg.pp_size == 1, \
             "trainer CP composes with pure DP for now"
     if cfg.zero_state == 1:
-        assert cfg.tp_size == 1 and cfg.ep_size == 1 and cfg.pp_size == 1, \
-            "ZeRO-1 composes with pure DP in the trainer for now"
+        assert cfg.ep_size == 1 and cfg.pp_size == 1, \
+            "trainer ZeRO-1 composes with DP and TP (not EP/PP yet)"
     tp_size = max(tp_size, cfg.tp_size)
     # Megatron-style vocab padding (keeps CE vocab a GPU-friendly multiple
     # and makes GEMM shape keys match the calibration tables)
@@ -403,7 +403,11 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                              tp_rank=tp_rank, sp=cfg.sequence_parallel,
                              cp_group=cp_group, cp_rank=cp_rank,
                              cp_size=cfg.cp_size, fp8=cfg.fp8)
-    opt = MixedPrecisionAdam(model.parameters(), cfg)
+    # ZeRO-1 shards the fp32 optimizer state over the DATA-parallel group
+    # (Megatron distributed optimizer); with tp > 1 that is dp_group, not
+    # the world
+    opt = MixedPrecisionAdam(model.parameters(), cfg,
+                             zero_group=dp_group if tp_size > 1 else None)
     if tp_group is not None:
         # tp shards are unique; norms/embedding are replicated across tp
         for p in opt.params:

@@ -156,3 +156,54 @@ def test_tp2_sp_matches_tp1_gradients():
     """Sequence parallelism: gather/scatter seq shards + partial-grad tp
     reduction for the norms must still reproduce TP1 gradients."""
     _run(29521, sp=True)
+
+
+def _zero_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from simumax_amd.train.trainer import (TrainConfig, build_trainer,
+                                               make_synthetic_batch,
+                                               train_step)
+
+        cfg = _tiny_cfg()
+        results = {}
+        for zero in (0, 1):
+            tc = TrainConfig(seq_len=32, micro_batch_size=2,
+                             micro_batch_num=1, overlap_grad_reduce=False,
+                             tp_size=2, zero_state=zero, grad_clip=1e9)
+            model, opt, red = build_trainer(cfg, tc, "cpu")
+            dp_rank = rank // 2
+            toks, labels = make_synthetic_batch(cfg.vocab_size, 1, 2, 32,
+                                                "cpu", seed=500 + dp_rank)
+            for _ in range(3):
+                train_step(model, opt, red, toks, labels, 1)
+            results[zero] = torch.cat(
+                [p.data.reshape(-1).float() for p in model.parameters()])
+            red.remove_hooks()
+        err = (results[0] - results[1]).abs().max().item()
+        q.put((rank, err))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(420)
+def test_zero1_with_tp2_matches_zero0():
+    """ZeRO-1 (fp32 state sharded over the dp group) composed with TP2 on
+    4 ranks produces the same post-step parameters as ZeRO-0."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29637
+    ps = [ctx.Process(target=_zero_worker, args=(r, 4, port, q))
+          for r in range(4)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=400) for _ in range(4)]
+    for p in ps:
+        p.join(timeout=60)
+    for rank, err in results:
+        assert err < 3e-3, f"rank {rank} param divergence {err}"
