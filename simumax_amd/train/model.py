@@ -100,12 +100,20 @@ class MLAAttention(nn.Module):
         self.dv = cfg.v_head_dim
         self.kv_lora = cfg.kv_lora_rank
         qk_total = self.dn + self.dp
-        assert cfg.q_lora_rank, "MLA trainer assumes q_lora_rank (DeepSeek-V2+)"
-        self.q_down = K.FusedLinear(h, cfg.q_lora_rank, dtype=dtype,
+        self.q_lora = bool(cfg.q_lora_rank)
+        if self.q_lora:
+            # DeepSeek-V2+: low-rank q projection
+            self.q_down = K.FusedLinear(h, cfg.q_lora_rank, dtype=dtype,
+                                        device=device)
+            self.q_norm = K.RMSNorm(cfg.q_lora_rank, dtype=dtype,
                                     device=device)
-        self.q_norm = K.RMSNorm(cfg.q_lora_rank, dtype=dtype, device=device)
-        self.q_up = K.FusedLinear(cfg.q_lora_rank, self.heads * qk_total,
-                                  dtype=dtype, device=device)
+            self.q_up = K.FusedLinear(cfg.q_lora_rank, self.heads * qk_total,
+                                      dtype=dtype, device=device)
+        else:
+            # DeepSeek-V2-Lite: direct q projection (simulator q_proj
+            # branch, ops/dense.py MLAAttention)
+            self.q_proj = K.FusedLinear(h, self.heads * qk_total,
+                                        dtype=dtype, device=device)
         self.kv_down = K.FusedLinear(h, self.kv_lora + self.dp, dtype=dtype,
                                      device=device)
         self.kv_norm = K.RMSNorm(self.kv_lora, dtype=dtype, device=device)
@@ -118,7 +126,10 @@ class MLAAttention(nn.Module):
     def forward(self, y, rope_cs, pos):
         B, S, _ = y.shape
         H, dn, dp, dv = self.heads, self.dn, self.dp, self.dv
-        q = self.q_up(self.q_norm(self.q_down(y))).view(B * S, H, dn + dp)
+        if self.q_lora:
+            q = self.q_up(self.q_norm(self.q_down(y))).view(B * S, H, dn + dp)
+        else:
+            q = self.q_proj(y).view(B * S, H, dn + dp)
         q_pe = K.apply_rope(q[..., dn:].contiguous(), rope_cs, pos)
         kv = self.kv_down(y)
         k_pe = K.apply_rope(
