@@ -175,9 +175,18 @@ class LlamaDecoderLayer(nn.Module):
         if self.use_moe:
             from .moe import MoEMLP
 
-            assert tp_size == 1, "trainer supports tp XOR ep for now"
+            assert tp_size == 1 or sp, \
+                "MoE with tp requires sequence parallel (tp x ep)"
             self.moe_mlp = MoEMLP(cfg, dtype=dtype, device=device,
                                   ep_group=ep_group, ep_size=ep_size)
+            if tp_size > 1 and sp:
+                # the router (and any shared-expert fcs) see only this tp
+                # rank's seq shard: their weight grads are partial sums
+                # and need the tp all_reduce (like the SP norms)
+                self.moe_mlp.router.weight._needs_tp_grad_reduce = True
+                if getattr(self.moe_mlp, "shared", False):
+                    self.moe_mlp.shared_fc1.weight._needs_tp_grad_reduce = True
+                    self.moe_mlp.shared_fc2.weight._needs_tp_grad_reduce = True
         else:
             # Megatron MLP split: gate and up each sharded I/tp; fc2 row-
             # parallel over I/tp with one fwd all_reduce

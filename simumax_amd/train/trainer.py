@@ -372,7 +372,11 @@ def accumulate_main_grads(params):
 def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                   tp_size: int = 1):
     torch.manual_seed(1234)
-    assert not (cfg.ep_size > 1 and cfg.tp_size > 1), "trainer: tp XOR ep"
+    if cfg.ep_size > 1 and cfg.tp_size > 1:
+        # tp x ep composition: tp ranks must hold DISTINCT token shards
+        # (sequence parallel), so each rank routes its own tokens and the
+        # EP a2a carries no duplicates (Megatron etp=1 semantics)
+        assert cfg.sequence_parallel, "tp x ep requires sequence_parallel"
     if cfg.cp_size > 1:
         assert cfg.tp_size == 1 and cfg.ep_size == 1 and cfg.pp_size == 1, \
             "trainer CP composes with pure DP for now"
