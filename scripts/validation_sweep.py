@@ -31,12 +31,15 @@ CASES = [
     ("qwen32b_l12_seq4096", "qwen3-32b-l12", 4096, 1, 2, 3),
     ("deepseekv2_l4_mla_moe", "deepseekv2-l4", 4096, 1, 2, 3),
     ("8b_fp8_seq4096_mbc4", "llama3-8b", 4096, 1, 4, 3, True),
+    # full-block activation recompute (trainer torch.utils.checkpoint vs
+    # simulator full_block/recompute_layer_num)
+    ("70b_l12_rc_seq8192", "llama3-70b-l12", 8192, 1, 2, 3, False, 12),
 ]
 
 OUT = "gpurun_out/validation.jsonl"
 
 
-def predict(model_cfg, seq, mbs, mbc, fp8=False):
+def predict(model_cfg, seq, mbs, mbc, fp8=False, rc=0):
     import copy
 
     from simumax_amd import (PerfLLM, StrategyConfig, SystemConfig,
@@ -45,8 +48,11 @@ def predict(model_cfg, seq, mbs, mbc, fp8=False):
     st = StrategyConfig(
         seq_len=seq, micro_batch_size=mbs, micro_batch_num=mbc,
         world_size=1, tp_size=1, pp_size=1, fp8=fp8,
+        enable_recompute=rc > 0,
+        recompute_granularity="full_block" if rc else None,
+        recompute_layer_num=rc,
         enable_sequence_parallel=False, zero_state=0,
-        use_fp32_accum_grad=True, enable_recompute=False,
+        use_fp32_accum_grad=True,
         cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
         mem_factor=1.0)
     p = PerfLLM()
@@ -57,10 +63,10 @@ def predict(model_cfg, seq, mbs, mbc, fp8=False):
     return p.analysis_cost(), p.analysis_mem()
 
 
-def run_case(name, model, seq, mbs, mbc, steps, fp8=False, warmup=1):
+def run_case(name, model, seq, mbs, mbc, steps, fp8=False, rc=0, warmup=1):
     mc = ModelConfig.init_from_config_file(get_simu_model_config(model))
     tc = TrainConfig(seq_len=seq, micro_batch_size=mbs, micro_batch_num=mbc,
-                     fp8=fp8)
+                     fp8=fp8, recompute_layers=rc)
     t0 = time.time()
     m, opt, red = build_trainer(mc, tc, "cuda:0")
     toks, labels = make_synthetic_batch(mc.vocab_size, mbc, mbs, seq, "cuda:0")
@@ -74,7 +80,7 @@ def run_case(name, model, seq, mbs, mbc, steps, fp8=False, warmup=1):
     torch.cuda.synchronize()
     ms = (time.time() - t1) / steps * 1e3
     peak = torch.cuda.max_memory_allocated()
-    cost, mem = predict(mc, seq, mbs, mbc, fp8=fp8)
+    cost, mem = predict(mc, seq, mbs, mbc, fp8=fp8, rc=rc)
     row = dict(
         case=name, model=model, seq=seq, mbs=mbs, mbc=mbc,
         measured_ms=round(ms, 2), predicted_ms=round(cost["iter_time"], 2),

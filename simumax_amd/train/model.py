@@ -225,12 +225,14 @@ class LlamaForTraining(nn.Module):
     def __init__(self, cfg: ModelConfig, seq_len: int, dtype=torch.bfloat16,
                  rope_base=500000.0, device=None, ep_group=None, ep_size=1,
                  tp_group=None, tp_size=1, tp_rank=0, sp=False,
-                 cp_group=None, cp_rank=0, cp_size=1, fp8=False):
+                 cp_group=None, cp_rank=0, cp_size=1, fp8=False,
+                 recompute_layers=0):
         super().__init__()
         self.cfg = cfg
         self.seq_len = seq_len          # FULL sequence (rope cache size)
         self.cp_rank = cp_rank
         self.cp_size = cp_size
+        self.recompute_layers = recompute_layers
         self.tp_group = tp_group
         self.tp_size = tp_size
         self.sp = sp and tp_size > 1
@@ -271,8 +273,17 @@ class LlamaForTraining(nn.Module):
             from .tp import slice_seq
 
             x = slice_seq(x, self.tp_group)
-        for layer in self.layers:
-            x = layer(x, self.rope_cs, pos)
+        for i, layer in enumerate(self.layers):
+            if self.training and i < self.recompute_layers:
+                # full-block activation recompute: only the block input is
+                # held; the forward reruns during backward (simulator
+                # full_block semantics, models/llm.py apply_recompute)
+                import torch.utils.checkpoint as ckpt
+
+                x = ckpt.checkpoint(layer, x, self.rope_cs, pos,
+                                    use_reentrant=False)
+            else:
+                x = layer(x, self.rope_cs, pos)
         x = self.final_norm(x)
         if self.tp_size > 1:
             from .tp import copy_to_tp, gather_seq, vocab_parallel_ce

@@ -39,6 +39,9 @@ class TrainConfig:
     pp_size: int = 1
     cp_size: int = 1   # Ulysses context parallel (seq_len = FULL sequence)
     fp8: bool = False  # e4m3/e5m2 GEMMs via _scaled_mm (decoder linears)
+    recompute_layers: int = 0  # full-block activation recompute for the
+                               # first N layers (torch.utils.checkpoint;
+                               # simulator full_block + recompute_layer_num)
     sequence_parallel: bool = False
     zero_state: int = 0   # 1 = ZeRO-1 distributed optimizer (sharded state)
 
@@ -406,7 +409,8 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                              tp_group=tp_group, tp_size=tp_size,
                              tp_rank=tp_rank, sp=cfg.sequence_parallel,
                              cp_group=cp_group, cp_rank=cp_rank,
-                             cp_size=cfg.cp_size, fp8=cfg.fp8)
+                             cp_size=cfg.cp_size, fp8=cfg.fp8,
+                             recompute_layers=cfg.recompute_layers)
     # ZeRO-1 shards the fp32 optimizer state over the DATA-parallel group
     # (Megatron distributed optimizer); with tp > 1 that is dp_group, not
     # the world

@@ -127,3 +127,33 @@ def test_zero1_matches_zero0_params():
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
+
+
+def test_recompute_layers_bitexact():
+    """Full-block activation recompute (torch.utils.checkpoint) must be
+    gradient-bitexact with the plain forward (no dropout, same RNG-free
+    path)."""
+    import torch
+
+    from simumax_amd.core.config import ModelConfig
+    from simumax_amd.train.trainer import (TrainConfig, build_trainer,
+                                           accumulate_main_grads,
+                                           make_synthetic_batch)
+
+    cfg = ModelConfig(hidden_size=128, head_num=4, kv_head_num=2,
+                      head_size=32, intermediate_size=256, layer_num=2,
+                      vocab_size=512, use_swiglu=True)
+    grads = {}
+    for rc in (0, 2):
+        torch.manual_seed(0)
+        tc = TrainConfig(seq_len=64, micro_batch_size=2, micro_batch_num=1,
+                         recompute_layers=rc, grad_clip=1e9)
+        m, opt, red = build_trainer(cfg, tc, "cpu")
+        toks, labels = make_synthetic_batch(cfg.vocab_size, 1, 2, 64,
+                                            "cpu", seed=3)
+        opt.zero_grad()
+        m(toks[0], labels[0]).backward()
+        accumulate_main_grads(opt.params)
+        grads[rc] = torch.cat([p.main_grad.reshape(-1) for p in opt.params])
+        red.remove_hooks()
+    assert torch.equal(grads[0], grads[2])
