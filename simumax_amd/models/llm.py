@@ -191,11 +191,18 @@ class LLMModel(MetaModule):
             if seg["recompute"]:
                 seg_input = seg["leaves"][0].input_info.total_bytes()
                 seg["held"] = seg_input
-                # transient: caches live while the segment executes fwd
-                transient = 0.0
+                # the checkpointed forward runs in NO-GRAD mode
+                # (torch.utils.checkpoint, use_reentrant=False):
+                # intermediates are freed as soon as consumed, so the
+                # transient is each leaf's LIVE working set (input +
+                # output + workspace), not the segment's summed caches
                 for leaf in seg["leaves"]:
-                    transient += leaf.get_act_info().activation_mem_cache
-                    peak_here = cache + seg_input + transient + leaf.get_act_info().fwd_peak_mem_no_cache
+                    ai = leaf.get_act_info()
+                    out_b = (leaf.output_info.total_bytes()
+                             if leaf.output_info is not None else 0.0)
+                    live = (leaf.input_info.total_bytes() + out_b
+                            + ai.fwd_peak_mem_no_cache)
+                    peak_here = cache + seg_input + live
                     if peak_here > pp.fwd_peak_mem:
                         pp.fwd_peak_mem = peak_here
                         pp.fwd_peak_point = leaf.full_name
@@ -252,14 +259,25 @@ class LLMModel(MetaModule):
 
     @staticmethod
     def _segments(leaves):
-        """Group consecutive leaves by recompute status."""
+        """Group consecutive leaves by recompute status, splitting at
+        layer boundaries: each checkpointed LAYER is its own segment
+        (torch.utils.checkpoint wraps one block at a time, so only one
+        block's activations rematerialize during its backward — merging
+        consecutive recomputed layers would claim every block's caches
+        live at once)."""
+        import re
+
         segs = []
         cur: Optional[dict] = None
+        cur_key = None
         for leaf in leaves:
             rc = bool(leaf.enable_recompute)
-            if cur is None or cur["recompute"] != rc:
+            m = re.search(r"\.layer(\d+)\.", leaf.full_name)
+            key = (rc, m.group(1) if (rc and m) else None)
+            if cur is None or key != cur_key:
                 cur = {"recompute": rc, "leaves": [], "held": 0.0}
                 segs.append(cur)
+                cur_key = key
             cur["leaves"].append(leaf)
         return segs
 
