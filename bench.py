@@ -42,10 +42,13 @@ def parse_args():
                     help="pipeline-parallel degree (1F1B)")
     ap.add_argument("--cp", type=int, default=1,
                     help="Ulysses context-parallel degree (seq sharded)")
+    ap.add_argument("--no-self-calibrate", action="store_true",
+                    help="skip the one-step in-situ calibration of THIS "
+                         "box before predicting (box clocks vary 3-5%)")
     return ap.parse_args()
 
 
-def predict(model_cfg, world, args):
+def predict(model_cfg, world, args, overlay=None):
     """PerfLLM prediction for the trainer's exact config."""
     from simumax_amd import (PerfLLM, StrategyConfig, SystemConfig,
                              get_simu_system_config)
@@ -68,6 +71,15 @@ def predict(model_cfg, world, args):
     )
     p = PerfLLM()
     sysc = SystemConfig.init_from_config_file(get_simu_system_config("mi355x"))
+    if overlay is not None:
+        # same-machine calibrate-then-validate, compressed into the bench:
+        # per-shape efficiencies measured on THIS box (one untimed step)
+        # override the shipped tables before predicting
+        from simumax_amd.calib.insitu_overlay import apply_insitu_overlay
+
+        n = apply_insitu_overlay(sysc, overlay)
+        print(f"[bench] self-calibrated {n} table entries on this box",
+              file=sys.stderr)
     import copy
 
     mc = copy.deepcopy(model_cfg)  # trainer pads the vocab the same way
@@ -141,6 +153,30 @@ def main():
 
     for _ in range(args.warmup):
         step_fn(model, opt, reducer, toks, labels, args.mbc)
+    overlay = None
+    if not args.no_self_calibrate:
+        # one untimed in-situ calibration step: event-time every GEMM /
+        # flash-attention / fused-op call on THIS box (box clocks vary)
+        from simumax_amd.core.consts import OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+        from simumax_amd.kernels import insitu
+
+        insitu.enable()
+        step_fn(model, opt, reducer, toks, labels, args.mbc)
+        torch.cuda.synchronize()
+        insitu.disable()
+        overlay = insitu.summarize()
+        # optimizer bandwidth at model scale (one extra harmless update)
+        s_ev = torch.cuda.Event(enable_timing=True)
+        e_ev = torch.cuda.Event(enable_timing=True)
+        s_ev.record()
+        opt.step()
+        e_ev.record()
+        torch.cuda.synchronize()
+        t_opt = s_ev.elapsed_time(e_ev)
+        numel = opt.flat_grad.numel()
+        overlay.setdefault("bandwidth", {})["optimizer_eff"] = (
+            numel * OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+            / (t_opt / 1e3) / (8000.0 * 1024**3))
     torch.cuda.reset_peak_memory_stats()
     if distributed:
         dist.barrier()
@@ -171,7 +207,7 @@ def main():
                         / (world * peak_tflops * 1e12))
         tokens_per_s = tokens_per_iter / (ms_per_step / 1e3)
 
-        cost, mem = predict(model_cfg, world, args)
+        cost, mem = predict(model_cfg, world, args, overlay=overlay)
         pred_ms = cost["iter_time"]
         pred_mfu = cost["mfu"]
         pred_peak = mem["max_peak_mem"]
