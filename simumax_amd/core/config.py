@@ -423,6 +423,11 @@ class AcceleratorConfig:
     op: Dict[str, CompOpConfig] = None
     bandwidth: Dict[str, BandwidthConfig] = None
     mode: str = "roofline"
+    # checkpointed-recompute rerun time as a fraction of a normal forward
+    # (measured: the rerun's consumers read just-produced activations from
+    # L2/L3 and the no-grad pass skips autograd recording); optional,
+    # schema-compatible extension ignored by the reference
+    recompute_factor: float = 1.0
 
 
 @dataclass
@@ -459,6 +464,7 @@ class SystemConfig(Config):
             op={k: CompOpConfig(**v) for k, v in acc["op"].items()},
             bandwidth={k: BandwidthConfig(**v) for k, v in acc["bandwidth"].items()},
             mode=acc.get("mode", "roofline"),
+            recompute_factor=acc.get("recompute_factor", 1.0),
         )
         nets = {
             name: NetworkConfig(

@@ -291,13 +291,13 @@ class PerfLLM(PerfBase):
                     chunk.compute_activations()
         self._estimated = True
 
-    @staticmethod
-    def _refresh_recompute_costs(chunk: LLMModel):
+    def _refresh_recompute_costs(self, chunk: LLMModel):
+        rf = getattr(self.system.accelerator, "recompute_factor", 1.0) or 1.0
         for leaf in chunk.leaf_modules():
             if leaf.enable_recompute:
                 ci = leaf._cost_info
                 if not leaf.is_variance_node:
-                    ci.recompute_compute_time = ci.fwd_compute_time
+                    ci.recompute_compute_time = ci.fwd_compute_time * rf
                     ci.recompute_net_time = ci.fwd_net_time
                     ci.recompute_net_exposed_time = ci.fwd_net_exposed_time
                     leaf._compute_info.recompute_flops = leaf._compute_info.fwd_flops
