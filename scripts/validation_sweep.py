@@ -34,12 +34,14 @@ CASES = [
     # full-block activation recompute (trainer torch.utils.checkpoint vs
     # simulator full_block/recompute_layer_num)
     ("70b_l12_rc_seq8192", "llama3-70b-l12", 8192, 1, 2, 3, False, 12),
+    ("8b_seq32768_mbc1", "llama3-8b", 32768, 1, 1, 2),
+    ("llama2_7b_seq4096_mbc4", "llama2-7b", 4096, 1, 4, 3),
 ]
 
 OUT = "gpurun_out/validation.jsonl"
 
 
-def predict(model_cfg, seq, mbs, mbc, fp8=False, rc=0):
+def predict(model_cfg, seq, mbs, mbc, fp8=False, rc=0, overlay=None):
     import copy
 
     from simumax_amd import (PerfLLM, StrategyConfig, SystemConfig,
@@ -55,10 +57,14 @@ def predict(model_cfg, seq, mbs, mbc, fp8=False, rc=0):
         use_fp32_accum_grad=True,
         cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
         mem_factor=1.0)
+    sysc = SystemConfig.init_from_config_file(
+        get_simu_system_config("mi355x"))
+    if overlay is not None:
+        from simumax_amd.calib.insitu_overlay import apply_insitu_overlay
+
+        apply_insitu_overlay(sysc, overlay)
     p = PerfLLM()
-    p.configure(st, copy.deepcopy(model_cfg),
-                SystemConfig.init_from_config_file(
-                    get_simu_system_config("mi355x")))
+    p.configure(st, copy.deepcopy(model_cfg), sysc)
     p.run_estimate()
     return p.analysis_cost(), p.analysis_mem()
 
@@ -72,6 +78,25 @@ def run_case(name, model, seq, mbs, mbc, steps, fp8=False, rc=0, warmup=1):
     toks, labels = make_synthetic_batch(mc.vocab_size, mbc, mbs, seq, "cuda:0")
     for _ in range(warmup):
         train_step(m, opt, red, toks, labels, mbc)
+    # one untimed in-situ step: THIS box's per-shape efficiencies feed the
+    # prediction (same-machine calibrate-then-validate per case)
+    from simumax_amd.core.consts import OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+    from simumax_amd.kernels import insitu
+
+    insitu.enable()
+    train_step(m, opt, red, toks, labels, mbc)
+    torch.cuda.synchronize()
+    insitu.disable()
+    overlay = insitu.summarize()
+    s_ev = torch.cuda.Event(enable_timing=True)
+    e_ev = torch.cuda.Event(enable_timing=True)
+    s_ev.record()
+    opt.step()
+    e_ev.record()
+    torch.cuda.synchronize()
+    overlay.setdefault("bandwidth", {})["optimizer_eff"] = (
+        opt.flat_grad.numel() * OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+        / (s_ev.elapsed_time(e_ev) / 1e3) / (8000.0 * 1024**3))
     torch.cuda.reset_peak_memory_stats()
     torch.cuda.synchronize()
     t1 = time.time()
@@ -80,7 +105,7 @@ def run_case(name, model, seq, mbs, mbc, steps, fp8=False, rc=0, warmup=1):
     torch.cuda.synchronize()
     ms = (time.time() - t1) / steps * 1e3
     peak = torch.cuda.max_memory_allocated()
-    cost, mem = predict(mc, seq, mbs, mbc, fp8=fp8, rc=rc)
+    cost, mem = predict(mc, seq, mbs, mbc, fp8=fp8, rc=rc, overlay=overlay)
     row = dict(
         case=name, model=model, seq=seq, mbs=mbs, mbc=mbc,
         measured_ms=round(ms, 2), predicted_ms=round(cost["iter_time"], 2),
