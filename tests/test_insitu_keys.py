@@ -158,3 +158,24 @@ def test_mla_sdp_key_matches():
     ins_key = insitu.sdp_key(1, 4096, 128, 128, 192, 128,
                              contiguous=192 == 128)
     assert ins_key == sim_key
+
+
+def test_insitu_overlay_applies():
+    """apply_insitu_overlay writes per-shape efficiencies and bandwidth
+    factors onto a live SystemConfig (bench/validation self-calibration)."""
+    from simumax_amd import SystemConfig, get_simu_system_config
+    from simumax_amd.calib.insitu_overlay import apply_insitu_overlay
+
+    sysc = SystemConfig.init_from_config_file(get_simu_system_config("mi355x"))
+    key = next(iter(sysc.accelerator.op["matmul"].accurate_efficient_factor))
+    summary = {
+        "matmul": {key: {"eff": 0.1234, "t_ms": 1.0, "n": 1}},
+        "bandwidth": {"optimizer_eff": 0.4321, "rmsnorm_fwd_eff": 0.9,
+                      "moe_routing_ms": 0.5},
+    }
+    n = apply_insitu_overlay(sysc, summary)
+    assert n >= 4
+    assert sysc.accelerator.op["matmul"].accurate_efficient_factor[key] == 0.1234
+    assert sysc.accelerator.bandwidth["optimizer"].efficient_factor == 0.4321
+    assert sysc.accelerator.bandwidth["rmsnorm_fwd"].efficient_factor == 0.9
+    assert sysc.accelerator.bandwidth["moe_routing"].latency_us == 500.0
