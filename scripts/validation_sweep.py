@@ -34,7 +34,8 @@ CASES = [
     # full-block activation recompute (trainer torch.utils.checkpoint vs
     # simulator full_block/recompute_layer_num)
     ("70b_l12_rc_seq8192", "llama3-70b-l12", 8192, 1, 2, 3, False, 12),
-    ("8b_seq32768_mbc1", "llama3-8b", 32768, 1, 1, 2),
+    # 32k long context needs full recompute to fit (act ~80 GiB without)
+    ("8b_seq32768_rc_mbc1", "llama3-8b", 32768, 1, 1, 2, False, 32),
     ("llama2_7b_seq4096_mbc4", "llama2-7b", 4096, 1, 4, 3),
 ]
 
@@ -75,62 +76,76 @@ def run_case(name, model, seq, mbs, mbc, steps, fp8=False, rc=0, warmup=1):
                      fp8=fp8, recompute_layers=rc)
     t0 = time.time()
     m, opt, red = build_trainer(mc, tc, "cuda:0")
-    toks, labels = make_synthetic_batch(mc.vocab_size, mbc, mbs, seq, "cuda:0")
-    for _ in range(warmup):
-        train_step(m, opt, red, toks, labels, mbc)
-    # one untimed in-situ step: THIS box's per-shape efficiencies feed the
-    # prediction (same-machine calibrate-then-validate per case)
-    from simumax_amd.core.consts import OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
-    from simumax_amd.kernels import insitu
+    try:
+        toks, labels = make_synthetic_batch(mc.vocab_size, mbc, mbs, seq,
+                                            "cuda:0")
+        for _ in range(warmup):
+            train_step(m, opt, red, toks, labels, mbc)
+        # one untimed in-situ step: THIS box's per-shape efficiencies feed
+        # the prediction (same-machine calibrate-then-validate per case).
+        # Skipped for recompute rows: the checkpoint rerun's instances
+        # share the fwd shape keys and would double-count the measured
+        # recompute_factor.
+        overlay = None
+        if not rc:
+            from simumax_amd.core.consts import \
+                OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+            from simumax_amd.kernels import insitu
 
-    insitu.enable()
-    train_step(m, opt, red, toks, labels, mbc)
-    torch.cuda.synchronize()
-    insitu.disable()
-    overlay = insitu.summarize()
-    s_ev = torch.cuda.Event(enable_timing=True)
-    e_ev = torch.cuda.Event(enable_timing=True)
-    s_ev.record()
-    opt.step()
-    e_ev.record()
-    torch.cuda.synchronize()
-    overlay.setdefault("bandwidth", {})["optimizer_eff"] = (
-        opt.flat_grad.numel() * OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
-        / (s_ev.elapsed_time(e_ev) / 1e3) / (8000.0 * 1024**3))
-    torch.cuda.reset_peak_memory_stats()
-    torch.cuda.synchronize()
-    t1 = time.time()
-    for _ in range(steps):
-        train_step(m, opt, red, toks, labels, mbc)
-    torch.cuda.synchronize()
-    ms = (time.time() - t1) / steps * 1e3
-    peak = torch.cuda.max_memory_allocated()
-    cost, mem = predict(mc, seq, mbs, mbc, fp8=fp8, rc=rc, overlay=overlay)
-    row = dict(
-        case=name, model=model, seq=seq, mbs=mbs, mbc=mbc,
-        measured_ms=round(ms, 2), predicted_ms=round(cost["iter_time"], 2),
-        timing_err_pct=round((cost["iter_time"] - ms) / ms * 100, 2),
-        measured_gib=round(peak / 2**30, 2),
-        predicted_gib=round(mem["max_peak_mem"] / 2**30, 2),
-        mem_err_pct=round((mem["max_peak_mem"] - peak) / peak * 100, 2),
-        fp8=fp8,
-        measured_mfu=round(
-            mc.flops_per_token(seq) * mbs * mbc * seq / (ms / 1e3) / 2.5e15, 4),
-        predicted_mfu=round(cost["mfu"], 4),
-        build_s=round(t1 - t0, 1),
-    )
-    print(json.dumps(row), flush=True)
-    with open(OUT, "a") as f:
-        f.write(json.dumps(row) + "\n")
-    # free everything before the next case (hooks pin the model otherwise)
-    from simumax_amd.kernels.ops import clear_dummy_wgrads
+            insitu.enable()
+            train_step(m, opt, red, toks, labels, mbc)
+            torch.cuda.synchronize()
+            insitu.disable()
+            overlay = insitu.summarize()
+            s_ev = torch.cuda.Event(enable_timing=True)
+            e_ev = torch.cuda.Event(enable_timing=True)
+            s_ev.record()
+            opt.step()
+            e_ev.record()
+            torch.cuda.synchronize()
+            overlay.setdefault("bandwidth", {})["optimizer_eff"] = (
+                opt.flat_grad.numel() * OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+                / (s_ev.elapsed_time(e_ev) / 1e3) / (8000.0 * 1024**3))
+        torch.cuda.reset_peak_memory_stats()
+        torch.cuda.synchronize()
+        t1 = time.time()
+        for _ in range(steps):
+            train_step(m, opt, red, toks, labels, mbc)
+        torch.cuda.synchronize()
+        ms = (time.time() - t1) / steps * 1e3
+        peak = torch.cuda.max_memory_allocated()
+        cost, mem = predict(mc, seq, mbs, mbc, fp8=fp8, rc=rc,
+                            overlay=overlay)
+        row = dict(
+            case=name, model=model, seq=seq, mbs=mbs, mbc=mbc,
+            measured_ms=round(ms, 2),
+            predicted_ms=round(cost["iter_time"], 2),
+            timing_err_pct=round((cost["iter_time"] - ms) / ms * 100, 2),
+            measured_gib=round(peak / 2**30, 2),
+            predicted_gib=round(mem["max_peak_mem"] / 2**30, 2),
+            mem_err_pct=round((mem["max_peak_mem"] - peak) / peak * 100, 2),
+            fp8=fp8, recompute_layers=rc,
+            measured_mfu=round(
+                mc.flops_per_token(seq) * mbs * mbc * seq / (ms / 1e3)
+                / 2.5e15, 4),
+            predicted_mfu=round(cost["mfu"], 4),
+            build_s=round(t1 - t0, 1),
+        )
+        print(json.dumps(row), flush=True)
+        with open(OUT, "a") as f:
+            f.write(json.dumps(row) + "\n")
+        return row
+    finally:
+        # free everything EVEN ON FAILURE (the post-accumulate hooks pin
+        # the whole trainer from C++ storage; a leaked case once added
+        # 45 GiB to the next case's measured peak)
+        from simumax_amd.kernels.ops import clear_dummy_wgrads
 
-    red.remove_hooks()
-    clear_dummy_wgrads()
-    del m, opt, red, toks, labels
-    gc.collect()
-    torch.cuda.empty_cache()
-    return row
+        red.remove_hooks()
+        clear_dummy_wgrads()
+        del m, opt, red
+        gc.collect()
+        torch.cuda.empty_cache()
 
 
 def main():
