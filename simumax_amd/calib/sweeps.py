@@ -442,6 +442,10 @@ DEFAULT_CASES = [
     # broader family coverage: every registered model's headline shapes
     ("llama3-8b", dict(world_size=1, tp_size=1, pp_size=1, seq_len=16384,
                        enable_sequence_parallel=False, zero_state=0)),
+    ("llama3-8b", dict(world_size=1, tp_size=1, pp_size=1, seq_len=32768,
+                       enable_sequence_parallel=False, zero_state=0,
+                       enable_recompute=True,
+                       recompute_granularity="full_block")),
     ("qwen3-32b-l12", dict(world_size=1, tp_size=1, pp_size=1,
                            enable_sequence_parallel=False, zero_state=0)),
     ("qwen3-32b", dict(tp_size=4)),
