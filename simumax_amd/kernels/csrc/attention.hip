@@ -21,6 +21,7 @@
 
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(2))) int v2i;
 
 #define FA_BLOCK 256
 #define QTILE 32            // q rows per wave (two 16-row subtiles)
@@ -94,7 +95,6 @@ DEV bf16x8v tr_frag(const bf16raw *sub_base, int lane) {
 #define FWD_BLOCK 512
 #define FWD_WAVES 8
 #define KVT2 64             // kv tile (two 32-key tr-image groups)
-#define PS2 80              // padded P row stride (64 cols + 16)
 
 template <int DQK>
 __global__ __launch_bounds__(FWD_BLOCK, 2)
@@ -108,7 +108,6 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     constexpr int BUF_ELEMS = KVT2 * KS_T + 2 * 8 * VSUB;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     bf16raw *buf0 = reinterpret_cast<bf16raw *>(smem);
-    bf16raw *P_all = buf0 + 2 * BUF_ELEMS;
 
     const int qblk = blockIdx.x;            // 256-row q block
     const int h = blockIdx.y;
@@ -134,7 +133,6 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     const bf16raw *qp = q + ((long)b * S) * q_row + (long)h * DQK;
     const bf16raw *kp = k + ((long)b * S) * k_row + (long)hkv * DQK;
     const bf16raw *vp = v + ((long)b * S) * v_row + (long)hkv * DHEAD;
-    bf16raw *Pw = P_all + wave * 2 * 16 * PS2;
 
     bf16x8v a_q[NQS][KC];
 #pragma unroll
@@ -150,15 +148,16 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
         }
     }
 
-    float m[NQS][4], l[NQS][4];
-    f32x4 acc[NQS][8];
+    // swapped-QK^T layout (guide T12): S^T = mfma(A=K, B=Q) puts one q
+    // COLUMN per lane, so the softmax row stats are (nearly) lane-local
+    // and P never round-trips through LDS. m/l are per-lane scalars.
+    float m[NQS], l[NQS];
+    f32x4 acc[NQS][8];        // O^T: lane holds col q, rows d
 #pragma unroll
-    for (int qs = 0; qs < NQS; ++qs)
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-            m[qs][j] = -INFINITY;
-            l[qs][j] = 0.f;
-        }
+    for (int qs = 0; qs < NQS; ++qs) {
+        m[qs] = -INFINITY;
+        l[qs] = 0.f;
+    }
 #pragma unroll
     for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
@@ -238,14 +237,17 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
         for (int qs = 0; qs < NQS; ++qs)
 #pragma unroll
             for (int ks = 0; ks < 4; ++ks) sq[qs][ks] = f32x4{0, 0, 0, 0};
+        // swapped QK^T: A = K subtile fragments (same bytes as the old
+        // B-fragments), B = the pre-scaled Q fragments (same registers) ->
+        // S^T[kv][q]: lane holds col q = lane&15, rows kv = ks*16+kgrp*4+j
         if (act0) {
 #pragma unroll
             for (int kc = 0; kc < KC; ++kc)
 #pragma unroll
                 for (int ks = 0; ks < 4; ++ks) {
                     bf16x8v bk = ld_frag(K_lds + (ks * 16 + col) * KS_T + kc * 32 + kgrp * 8);
-                    sq[0][ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[0][kc], bk, sq[0][ks], 0, 0, 0);
-                    sq[1][ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[1][kc], bk, sq[1][ks], 0, 0, 0);
+                    sq[0][ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bk, a_q[0][kc], sq[0][ks], 0, 0, 0);
+                    sq[1][ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bk, a_q[1][kc], sq[1][ks], 0, 0, 0);
                 }
         } else {
 #pragma unroll
@@ -253,83 +255,97 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
 #pragma unroll
                 for (int ks = 0; ks < 4; ++ks) {
                     bf16x8v bk = ld_frag(K_lds + (ks * 16 + col) * KS_T + kc * 32 + kgrp * 8);
-                    sq[1][ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[1][kc], bk, sq[1][ks], 0, 0, 0);
+                    sq[1][ks] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bk, a_q[1][kc], sq[1][ks], 0, 0, 0);
                 }
         }
 
-        // online softmax + P->LDS per active subtile
+        // online softmax (lane-local over 16 values + 2 xor-shuffles
+        // across the 4 kgrp lanes of each q column) + in-register P via
+        // cvt_pk + the permlane32/16 butterfly -> PV B-fragments
+        bf16x8v b_p[NQS][2];
         auto softmax_p = [&](int qs) {
             const bool clean = (qsb[qs] + 16 <= S) && (kv + KVT2 <= S)
                 && (!causal || kv + KVT2 - 1 <= qsb[qs]);
-            float tile_max[4];
             if (!clean) {
+                const int qrow = qsb[qs] + col;
 #pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    const int row = qsb[qs] + kgrp * 4 + j;
+                for (int ks = 0; ks < 4; ++ks)
 #pragma unroll
-                    for (int ks = 0; ks < 4; ++ks) {
-                        const int c0 = kv + ks * 16 + col;
-                        if (row >= S || c0 >= S || (causal && c0 > row))
+                    for (int j = 0; j < 4; ++j) {
+                        const int c0 = kv + ks * 16 + kgrp * 4 + j;
+                        if (qrow >= S || c0 >= S || (causal && c0 > qrow))
                             sq[qs][ks][j] = -INFINITY;
                     }
-                }
             }
+            float mx = sq[qs][0][0];
 #pragma unroll
-            for (int j = 0; j < 4; ++j)
-                tile_max[j] = group16_max(
-                    fmaxf(fmaxf(sq[qs][0][j], sq[qs][1][j]),
-                          fmaxf(sq[qs][2][j], sq[qs][3][j])));
-            float alpha[4];
+            for (int ks = 0; ks < 4; ++ks)
 #pragma unroll
-            for (int j = 0; j < 4; ++j) {
-                float mn = fmaxf(m[qs][j], tile_max[j]);
-                // exp(-inf - mn) = 0, so the -INFINITY selects are free
-                alpha[j] = __expf(m[qs][j] - mn);
-                m[qs][j] = mn;
-                float ps = 0.f;
+                for (int j = 0; j < 4; ++j) mx = fmaxf(mx, sq[qs][ks][j]);
+            mx = fmaxf(mx, __shfl_xor(mx, 16, WAVE));
+            mx = fmaxf(mx, __shfl_xor(mx, 32, WAVE));
+            const float mn = fmaxf(m[qs], mx);
+            // exp(-inf - mn) = 0, so the masked entries cost nothing extra
+            const float alpha = __expf(m[qs] - mn);
+            m[qs] = mn;
+            float ps = 0.f;
 #pragma unroll
-                for (int ks = 0; ks < 4; ++ks) {
-                    float p = __expf(sq[qs][ks][j] - mn);
-                    sq[qs][ks][j] = p;
-                    ps += p;
+            for (int ks = 0; ks < 4; ++ks)
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    float pv = __expf(sq[qs][ks][j] - mn);
+                    sq[qs][ks][j] = pv;
+                    ps += pv;
                 }
-                // l kept per-lane; reduced once in the epilogue
-                l[qs][j] = l[qs][j] * alpha[j] + ps;
-            }
+            // l kept per-lane (per kgrp); reduced once in the epilogue
+            l[qs] = l[qs] * alpha + ps;
 #pragma unroll
             for (int dt = 0; dt < 8; ++dt)
 #pragma unroll
-                for (int j = 0; j < 4; ++j) acc[qs][dt][j] *= alpha[j];
-            bf16raw *pw = Pw + qs * 16 * PS2;
+                for (int j = 0; j < 4; ++j) acc[qs][dt][j] *= alpha;
+            // pack P to bf16 and butterfly into PV B-fragments: target
+            // lane kgrp k' holds kv 32t + 8k' + [0,8) of its q column
 #pragma unroll
-            for (int j = 0; j < 4; ++j)
-#pragma unroll
-                for (int ks = 0; ks < 4; ++ks)
-                    pw[(kgrp * 4 + j) * PS2 + ks * 16 + col] = f2bf(sq[qs][ks][j]);
+            for (int t2 = 0; t2 < 2; ++t2) {
+                int E0, E1, F0, F1;
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(E0)
+                    : "v"(sq[qs][2 * t2][0]), "v"(sq[qs][2 * t2][1]));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(E1)
+                    : "v"(sq[qs][2 * t2][2]), "v"(sq[qs][2 * t2][3]));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(F0)
+                    : "v"(sq[qs][2 * t2 + 1][0]), "v"(sq[qs][2 * t2 + 1][1]));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(F1)
+                    : "v"(sq[qs][2 * t2 + 1][2]), "v"(sq[qs][2 * t2 + 1][3]));
+                v2i u0 = __builtin_amdgcn_permlane32_swap(E0, F0, false, false);
+                v2i u1 = __builtin_amdgcn_permlane32_swap(E1, F1, false, false);
+                v2i d02 = __builtin_amdgcn_permlane16_swap(u0[0], u0[1], false, false);
+                v2i d13 = __builtin_amdgcn_permlane16_swap(u1[0], u1[1], false, false);
+                int frag[4] = {d02[0], d13[0], d02[1], d13[1]};
+                b_p[qs][t2] = *reinterpret_cast<bf16x8v *>(frag);
+            }
         };
         if (act0) softmax_p(0);
         softmax_p(1);
 
-        // PV over the two 32-key groups; V fragments shared across subtiles
+        // PV swapped: O^T += mfma(A = V^T tr-fragments (same bytes as the
+        // old B side), B = in-register P^T fragments)
 #pragma unroll
         for (int g = 0; g < 2; ++g) {
-            bf16x8v a_p0 = ld_frag(Pw + col * PS2 + g * 32 + kgrp * 8);
-            bf16x8v a_p1 = ld_frag(Pw + 16 * PS2 + col * PS2 + g * 32 + kgrp * 8);
             if (act0) {
 #pragma unroll
                 for (int dt = 0; dt < 8; ++dt) {
                     bf16x8v b_v = tr_frag(V_img + g * 8 * VSUB + dt * VSUB, lane);
                     acc[0][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a_p0, b_v, acc[0][dt], 0, 0, 0);
+                        b_v, b_p[0][g], acc[0][dt], 0, 0, 0);
                     acc[1][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a_p1, b_v, acc[1][dt], 0, 0, 0);
+                        b_v, b_p[1][g], acc[1][dt], 0, 0, 0);
                 }
             } else {
 #pragma unroll
                 for (int dt = 0; dt < 8; ++dt) {
                     bf16x8v b_v = tr_frag(V_img + g * 8 * VSUB + dt * VSUB, lane);
                     acc[1][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a_p1, b_v, acc[1][dt], 0, 0, 0);
+                        b_v, b_p[1][g], acc[1][dt], 0, 0, 0);
                 }
             }
         }
@@ -339,25 +355,31 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
     }
 
     bf16raw *op = o + ((long)b * S) * o_row + (long)h * DHEAD;
-    // reduce the per-lane l partials across the 16-lane column group once
+    // O^T epilogue: lane owns q column `col`; its 32 acc values are d =
+    // dt*16 + kgrp*4 + [0,4) — contiguous, so each dt is ONE 8-byte store
 #pragma unroll
-    for (int qs = 0; qs < NQS; ++qs)
+    for (int qs = 0; qs < NQS; ++qs) {
+        float lt = l[qs];
+        lt += __shfl_xor(lt, 16, WAVE);
+        lt += __shfl_xor(lt, 32, WAVE);
+        const int row = qsb[qs] + col;
+        if (row >= S) continue;
+        const float inv = (lt > 0.f) ? 1.f / lt : 0.f;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) l[qs][j] = group16_sum(l[qs][j]);
-#pragma unroll
-    for (int qs = 0; qs < NQS; ++qs)
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-            const int row = qsb[qs] + kgrp * 4 + j;
-            if (row >= S) continue;
-            const float inv = (l[qs][j] > 0.f) ? 1.f / l[qs][j] : 0.f;
-#pragma unroll
-            for (int dt = 0; dt < 8; ++dt)
-                op[(long)row * o_row + dt * 16 + col] = f2bf(acc[qs][dt][j] * inv);
-            if (col == 0)
-                lse[((long)b * Hq + h) * S + row] =
-                    (l[qs][j] > 0.f) ? m[qs][j] + __logf(l[qs][j]) : -INFINITY;
+        for (int dt = 0; dt < 8; ++dt) {
+            const unsigned w0 =
+                (unsigned)f2bf(acc[qs][dt][0] * inv) |
+                ((unsigned)f2bf(acc[qs][dt][1] * inv) << 16);
+            const unsigned w1 =
+                (unsigned)f2bf(acc[qs][dt][2] * inv) |
+                ((unsigned)f2bf(acc[qs][dt][3] * inv) << 16);
+            *reinterpret_cast<uint2 *>(
+                op + (long)row * o_row + dt * 16 + kgrp * 4) = uint2{w0, w1};
         }
+        if (kgrp == 0)
+            lse[((long)b * Hq + h) * S + row] =
+                (lt > 0.f) ? m[qs] + __logf(lt) : -INFINITY;
+    }
 }
 
 // ==========================================================================
@@ -824,16 +846,14 @@ extern "C" void fa_fwd_launch(const void *q, const void *k, const void *v,
                               hipStream_t stream) {
     if (dqk == 128) {
         dim3 grid(CDIV(S, 256), Hq, B);
-        size_t smem = (2 * (KVT2 * (128 + 16) + 2 * 8 * VSUB)
-                       + FWD_WAVES * 2 * 16 * PS2) * sizeof(bf16raw);
+        size_t smem = 2 * (KVT2 * (128 + 16) + 2 * 8 * VSUB) * sizeof(bf16raw);
         hipLaunchKernelGGL((fa_fwd_kernel<128>), grid, dim3(FWD_BLOCK),
                            smem, stream, (const bf16raw *)q,
                            (const bf16raw *)k, (const bf16raw *)v,
                            (bf16raw *)o, (float *)lse, B, S, Hq, Hkv, causal);
     } else if (dqk == 192) {
         dim3 grid(CDIV(S, 256), Hq, B);
-        size_t smem = (2 * (KVT2 * (192 + 16) + 2 * 8 * VSUB)
-                       + FWD_WAVES * 2 * 16 * PS2) * sizeof(bf16raw);
+        size_t smem = 2 * (KVT2 * (192 + 16) + 2 * 8 * VSUB) * sizeof(bf16raw);
         hipLaunchKernelGGL((fa_fwd_kernel<192>), grid, dim3(FWD_BLOCK),
                            smem, stream, (const bf16raw *)q,
                            (const bf16raw *)k, (const bf16raw *)v,
