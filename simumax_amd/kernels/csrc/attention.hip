@@ -32,26 +32,14 @@ typedef __attribute__((ext_vector_type(2))) int v2i;
 // makes (col*stride + kgrp*8) distinct banks within every ds_read_b128
 // 16-lane group (a +8 pad leaves a 2-way conflict: stride dwords % 16 == 4)
 #define VTS 48              // padded V^T row stride
-#define PS 48               // padded P row stride
 
 DEV bf16x8v ld_frag(const bf16raw *p) {
     uint4 r = *reinterpret_cast<const uint4 *>(p);
     return *reinterpret_cast<bf16x8v *>(&r);
 }
 
-DEV float group16_max(float v) {
-#pragma unroll
-    for (int off = 1; off < 16; off <<= 1) v = fmaxf(v, __shfl_xor(v, off, WAVE));
-    return v;
-}
-
-DEV float group16_sum(float v) {
-#pragma unroll
-    for (int off = 1; off < 16; off <<= 1) v += __shfl_xor(v, off, WAVE);
-    return v;
-}
-
-// LDS (fwd): 2 x { K[32][KS], Vt[128][VTS] } + per-wave P[32][PS]
+// LDS (fwd): 2 x double-buffered { K rows, tr-readable V image }; P stays
+// in registers (swapped-QK^T layout)
 // V is staged as a tr-read image: 8 d-subtiles of [32 permuted key rows][16],
 // each padded to VSUB elems; B-fragments come from two ds_read_b64_tr_b16
 // per MFMA (guide T10: lane l elem j reads lds[(l&15) + j*16 + (l>>4)*64]).
@@ -392,7 +380,6 @@ void fa_fwd_kernel(const bf16raw *__restrict__ q, const bf16raw *__restrict__ k,
 // ==========================================================================
 #define BW_QT 32
 #define QS2 144   // padded [32][128] row stride
-#define TS2 48    // padded [..][32] row stride
 #define BKV 16    // keys per wave in dkv kernel
 
 extern "C" __global__ __launch_bounds__(FA_BLOCK)
@@ -447,10 +434,8 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
     const int lane = threadIdx.x % WAVE;
     const int col = lane & 15;
     const int kgrp = lane >> 4;
-    const int WSZ = BKV * KS_T + 2 * BKV * TS2;
+    const int WSZ = BKV * KS_T;
     bf16raw *K_l = wbase + wave * WSZ;         // [16][KS_T] scaled
-    bf16raw *Pt_l = K_l + BKV * KS_T;          // [16k][TS2]
-    bf16raw *dSt_l = Pt_l + BKV * TS2;         // [16k][TS2]
 
     const int kvblk = blockIdx.x;
     const int h = blockIdx.y;
@@ -559,6 +544,7 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
     q_stage_load(q_start + BW_QT);
     for (int qt = q_start; qt < S; qt += BW_QT) {
 
+        f32x4 p_r[2], ds_r[2];
 #pragma unroll
         for (int qs = 0; qs < 2; ++qs) {
             f32x4 s0 = f32x4{0, 0, 0, 0};
@@ -590,16 +576,34 @@ void fa_bwd_dkv_kernel(const bf16raw *__restrict__ dout,
                     p0 = (row < S && c0 < S && (!causal || c0 <= row)
                           && ls != -INFINITY) ? __expf(s0[j] - ls) : 0.f;
                 }
-                float ds0 = p0 * (dp0[j] - Dv) * scale;
-                const int qrow = qs * 16 + kgrp * 4 + j;
-                Pt_l[col * TS2 + qrow] = f2bf(p0);
-                dSt_l[col * TS2 + qrow] = f2bf(ds0);
+                p_r[qs][j] = p0;
+                ds_r[qs][j] = p0 * (dp0[j] - Dv) * scale;
             }
         }
-        // dV += P^T @ dO ; dK += dS^T @ Q(unscaled via Q_img)
+        // dV += P^T @ dO ; dK += dS^T @ Q(unscaled via Q_img). P^T and
+        // dS^T come straight from the C tiles via the cvt_pk + permlane
+        // butterfly (the output registers read as an A-fragment ARE the
+        // transpose) — no element-wise LDS transpose round trip.
         {
-            bf16x8v a_pt = ld_frag(Pt_l + col * TS2 + kgrp * 8);
-            bf16x8v a_dst = ld_frag(dSt_l + col * TS2 + kgrp * 8);
+            auto butterfly = [&](const f32x4 &e, const f32x4 &f) {
+                int E0, E1, F0, F1;
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(E0)
+                    : "v"(e[0]), "v"(e[1]));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(E1)
+                    : "v"(e[2]), "v"(e[3]));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(F0)
+                    : "v"(f[0]), "v"(f[1]));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(F1)
+                    : "v"(f[2]), "v"(f[3]));
+                v2i u0 = __builtin_amdgcn_permlane32_swap(E0, F0, false, false);
+                v2i u1 = __builtin_amdgcn_permlane32_swap(E1, F1, false, false);
+                v2i d02 = __builtin_amdgcn_permlane16_swap(u0[0], u0[1], false, false);
+                v2i d13 = __builtin_amdgcn_permlane16_swap(u1[0], u1[1], false, false);
+                int frag[4] = {d02[0], d13[0], d02[1], d13[1]};
+                return *reinterpret_cast<bf16x8v *>(frag);
+            };
+            bf16x8v a_pt = butterfly(p_r[0], p_r[1]);
+            bf16x8v a_dst = butterfly(ds_r[0], ds_r[1]);
 #pragma unroll
             for (int dt = 0; dt < 8; ++dt) {
                 bf16x8v b_do = tr_frag(dO_img + dt * VSUB, lane);
@@ -652,7 +656,6 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
     constexpr int BUF = KVTILE * KS_T + KVTILE * KS + QSUB * VSUB;
     extern __shared__ __attribute__((aligned(16))) char smem[];
     bf16raw *buf0 = reinterpret_cast<bf16raw *>(smem);
-    bf16raw *dS_all = buf0 + 2 * BUF;
 
     const int qblk = blockIdx.x;
     const int h = blockIdx.y;
@@ -673,13 +676,14 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
     const bf16raw *kp = k + ((long)b * S) * k_row + (long)hkv * DQK;
     const bf16raw *vp = v + ((long)b * S) * v_row + (long)hkv * DHEAD;
     const bf16raw *dop = dout + ((long)b * S) * o_row + (long)h * DHEAD;
-    bf16raw *dSw = dS_all + wave * QT * PS;
 
     // per-wave q-side registers: scaled Q frags, dO frags, lse, D — all
     // kv-tile-invariant, so they are loaded ONCE (r1 PMC: re-reading dO
-    // from global inside the kv loop parked the wave 51.5% of its cycles)
+    // from global inside the kv loop parked the wave 51.5% of its cycles).
+    // Swapped layout (T12, like the forward): the wave's q rows become
+    // per-lane COLUMNS, so lse/D are per-lane scalars.
     bf16x8v a_q[NQS][KC], a_do[NQS][4];
-    float lse_r[NQS][4], D_r[NQS][4];
+    float lse_r[NQS], D_r[NQS];
 #pragma unroll
     for (int qs = 0; qs < NQS; ++qs) {
         const int qrow = qbase + qs * 16 + col;
@@ -695,13 +699,9 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
         for (int kc = 0; kc < 4; ++kc)
             a_do[qs][kc] = ld_frag(
                 dop + (long)min(qrow, S - 1) * o_row + kc * 32 + kgrp * 8);
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-            const int row = qbase + qs * 16 + kgrp * 4 + j;
-            const long idx = ((long)b * Hq + h) * S + min(row, S - 1);
-            lse_r[qs][j] = lse[idx];
-            D_r[qs][j] = Dsum[idx];
-        }
+        const long idx = ((long)b * Hq + h) * S + min(qrow, S - 1);
+        lse_r[qs] = lse[idx];
+        D_r[qs] = Dsum[idx];
     }
 
     f32x4 dq_acc[NQS][QSUB];
@@ -772,53 +772,68 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
         if (!(causal && kv >= qbase + QT)) {
 #pragma unroll
             for (int qs = 0; qs < NQS; ++qs) {
-                f32x4 s0 = f32x4{0, 0, 0, 0}, s1 = f32x4{0, 0, 0, 0};
-                f32x4 dp0 = f32x4{0, 0, 0, 0}, dp1 = f32x4{0, 0, 0, 0};
+                // swapped: S^T[kv][q], dP^T[kv][q] — lane col = q, rows kv
+                f32x4 s_t[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
+                f32x4 dp_t[2] = {f32x4{0, 0, 0, 0}, f32x4{0, 0, 0, 0}};
 #pragma unroll
                 for (int kc = 0; kc < KC; ++kc) {
                     bf16x8v bk0 = ld_frag(K_lds + col * KS_T + kc * 32 + kgrp * 8);
                     bf16x8v bk1 = ld_frag(K_lds + (16 + col) * KS_T + kc * 32 + kgrp * 8);
-                    s0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], bk0, s0, 0, 0, 0);
-                    s1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_q[qs][kc], bk1, s1, 0, 0, 0);
+                    s_t[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bk0, a_q[qs][kc], s_t[0], 0, 0, 0);
+                    s_t[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bk1, a_q[qs][kc], s_t[1], 0, 0, 0);
                 }
 #pragma unroll
                 for (int kc = 0; kc < 4; ++kc) {
                     bf16x8v bv0 = ld_frag(V_lds + col * KS + kc * 32 + kgrp * 8);
                     bf16x8v bv1 = ld_frag(V_lds + (16 + col) * KS + kc * 32 + kgrp * 8);
-                    dp0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do[qs][kc], bv0, dp0, 0, 0, 0);
-                    dp1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_do[qs][kc], bv1, dp1, 0, 0, 0);
+                    dp_t[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bv0, a_do[qs][kc], dp_t[0], 0, 0, 0);
+                    dp_t[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(bv1, a_do[qs][kc], dp_t[1], 0, 0, 0);
                 }
                 const bool clean = (qbase + qs * 16 + 16 <= S)
                     && (kv + KVTILE <= S)
                     && (!causal || kv + KVTILE - 1 <= qbase + qs * 16);
+                const float ls = lse_r[qs];
+                const float Dv = D_r[qs];
+                const int qcol = qbase + qs * 16 + col;
 #pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    const float ls = lse_r[qs][j];
-                    const float Dv = D_r[qs][j];
-                    float p0, p1;
-                    if (clean) {
-                        p0 = __expf(s0[j] - ls);
-                        p1 = __expf(s1[j] - ls);
-                    } else {
-                        const int row = qbase + qs * 16 + kgrp * 4 + j;
-                        const int c0 = kv + col, c1 = kv + 16 + col;
-                        p0 = (row < S && c0 < S && (!causal || c0 <= row)
-                              && ls != -INFINITY) ? __expf(s0[j] - ls) : 0.f;
-                        p1 = (row < S && c1 < S && (!causal || c1 <= row)
-                              && ls != -INFINITY) ? __expf(s1[j] - ls) : 0.f;
+                for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+                    for (int j = 0; j < 4; ++j) {
+                        float pv;
+                        if (clean) {
+                            pv = __expf(s_t[ks][j] - ls);
+                        } else {
+                            const int c0 = kv + ks * 16 + kgrp * 4 + j;
+                            pv = (qcol < S && c0 < S
+                                  && (!causal || c0 <= qcol)
+                                  && ls != -INFINITY)
+                                     ? __expf(s_t[ks][j] - ls) : 0.f;
+                        }
+                        s_t[ks][j] = pv * (dp_t[ks][j] - Dv) * scale;
                     }
-                    float ds0 = p0 * (dp0[j] - Dv) * scale;
-                    float ds1 = p1 * (dp1[j] - Dv) * scale;
-                    bf16raw *pw = dSw + qs * 16 * PS;
-                    pw[(kgrp * 4 + j) * PS + col] = f2bf(ds0);
-                    pw[(kgrp * 4 + j) * PS + 16 + col] = f2bf(ds1);
-                }
-                bf16x8v a_ds = ld_frag(dSw + qs * 16 * PS + col * PS + kgrp * 8);
+                // dS^T -> B-fragment in registers (cvt_pk + permlane
+                // butterfly, same mapping as the forward's P^T)
+                int E0, E1, F0, F1;
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(E0)
+                    : "v"(s_t[0][0]), "v"(s_t[0][1]));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(E1)
+                    : "v"(s_t[0][2]), "v"(s_t[0][3]));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(F0)
+                    : "v"(s_t[1][0]), "v"(s_t[1][1]));
+                asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(F1)
+                    : "v"(s_t[1][2]), "v"(s_t[1][3]));
+                v2i u0 = __builtin_amdgcn_permlane32_swap(E0, F0, false, false);
+                v2i u1 = __builtin_amdgcn_permlane32_swap(E1, F1, false, false);
+                v2i d02 = __builtin_amdgcn_permlane16_swap(u0[0], u0[1], false, false);
+                v2i d13 = __builtin_amdgcn_permlane16_swap(u1[0], u1[1], false, false);
+                int frag[4] = {d02[0], d13[0], d02[1], d13[1]};
+                bf16x8v b_ds = *reinterpret_cast<bf16x8v *>(frag);
+                // dq^T = mfma(A = K^T tr-fragments, B = dS^T)
 #pragma unroll
                 for (int dt = 0; dt < QSUB; ++dt) {
-                    bf16x8v b_k = tr_frag(K_img + dt * VSUB, lane);
+                    bf16x8v a_k = tr_frag(K_img + dt * VSUB, lane);
                     dq_acc[qs][dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a_ds, b_k, dq_acc[qs][dt], 0, 0, 0);
+                        a_k, b_ds, dq_acc[qs][dt], 0, 0, 0);
                 }
             }
         }
@@ -828,16 +843,24 @@ void fa_bwd_dq_kernel(const bf16raw *__restrict__ dout,
     }
 
     bf16raw *dqp = dq + ((long)b * S) * q_row + (long)h * DQK;
+    // dq^T epilogue: lane owns q column; 4 contiguous d per (dt) -> one
+    // 8-byte store each (same shape as the forward's O^T epilogue)
 #pragma unroll
-    for (int qs = 0; qs < NQS; ++qs)
+    for (int qs = 0; qs < NQS; ++qs) {
+        const int row = qbase + qs * 16 + col;
+        if (row >= S) continue;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-            const int row = qbase + qs * 16 + kgrp * 4 + j;
-            if (row >= S) continue;
-#pragma unroll
-            for (int dt = 0; dt < QSUB; ++dt)
-                dqp[(long)row * q_row + dt * 16 + col] = f2bf(dq_acc[qs][dt][j]);
+        for (int dt = 0; dt < QSUB; ++dt) {
+            const unsigned w0 =
+                (unsigned)f2bf(dq_acc[qs][dt][0]) |
+                ((unsigned)f2bf(dq_acc[qs][dt][1]) << 16);
+            const unsigned w1 =
+                (unsigned)f2bf(dq_acc[qs][dt][2]) |
+                ((unsigned)f2bf(dq_acc[qs][dt][3]) << 16);
+            *reinterpret_cast<uint2 *>(
+                dqp + (long)row * q_row + dt * 16 + kgrp * 4) = uint2{w0, w1};
         }
+    }
 }
 
 extern "C" void fa_fwd_launch(const void *q, const void *k, const void *v,
@@ -885,7 +908,7 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
             smem = (BW_QT * (128 + 16) + BW_QT * QS2 + (128 / 16 + 8) * VSUB)
                        * sizeof(bf16raw)
                    + 2 * BW_QT * sizeof(float)
-                   + WAVES * (BKV * (128 + 16) + 2 * BKV * TS2) * sizeof(bf16raw);
+                   + WAVES * BKV * (128 + 16) * sizeof(bf16raw);
             hipLaunchKernelGGL((fa_bwd_dkv_kernel<128>), grid, dim3(FA_BLOCK),
                                smem, stream, (const bf16raw *)dout,
                                (const bf16raw *)q, (const bf16raw *)k,
@@ -896,7 +919,7 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
             smem = (BW_QT * (192 + 16) + BW_QT * QS2 + (192 / 16 + 8) * VSUB)
                        * sizeof(bf16raw)
                    + 2 * BW_QT * sizeof(float)
-                   + WAVES * (BKV * (192 + 16) + 2 * BKV * TS2) * sizeof(bf16raw);
+                   + WAVES * BKV * (192 + 16) * sizeof(bf16raw);
             hipLaunchKernelGGL((fa_bwd_dkv_kernel<192>), grid, dim3(FA_BLOCK),
                                smem, stream, (const bf16raw *)dout,
                                (const bf16raw *)q, (const bf16raw *)k,
@@ -911,8 +934,8 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
     if (dqk == 128) {
         constexpr int QT = 16;   // NQS=1: occupancy over per-wave work
         dim3 grid(CDIV(S, WAVES * QT), Hq, B);
-        size_t smem = (2 * (KVTILE * (128 + 16) + KVTILE * KS + 8 * VSUB)
-                       + WAVES * QT * PS) * sizeof(bf16raw);
+        size_t smem = 2 * (KVTILE * (128 + 16) + KVTILE * KS + 8 * VSUB)
+                      * sizeof(bf16raw);
         hipLaunchKernelGGL((fa_bwd_dq_kernel<128, 1>), grid, dim3(FA_BLOCK),
                            smem, stream, (const bf16raw *)dout,
                            (const bf16raw *)q, (const bf16raw *)k,
@@ -922,8 +945,8 @@ extern "C" void fa_bwd_launch(const void *dout, const void *q, const void *k,
     } else {
         constexpr int QT = 16;
         dim3 grid(CDIV(S, WAVES * QT), Hq, B);
-        size_t smem = (2 * (KVTILE * (192 + 16) + KVTILE * KS + 12 * VSUB)
-                       + WAVES * QT * PS) * sizeof(bf16raw);
+        size_t smem = 2 * (KVTILE * (192 + 16) + KVTILE * KS + 12 * VSUB)
+                      * sizeof(bf16raw);
         hipLaunchKernelGGL((fa_bwd_dq_kernel<192, 1>), grid, dim3(FA_BLOCK),
                            smem, stream, (const bf16raw *)dout,
                            (const bf16raw *)q, (const bf16raw *)k,
