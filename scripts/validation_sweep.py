@@ -83,11 +83,11 @@ def run_case(name, model, seq, mbs, mbc, steps, fp8=False, rc=0, warmup=1):
             train_step(m, opt, red, toks, labels, mbc)
         # one untimed in-situ step: THIS box's per-shape efficiencies feed
         # the prediction (same-machine calibrate-then-validate per case).
-        # Skipped for recompute rows: the checkpoint rerun's instances
-        # share the fwd shape keys and would double-count the measured
-        # recompute_factor.
+        # Recompute rows participate too: insitu tags instances recorded
+        # inside the backward graph task, so checkpoint reruns cannot
+        # shift the fwd keys (recompute_factor prices the rerun).
         overlay = None
-        if not rc:
+        if True:
             from simumax_amd.core.consts import \
                 OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
             from simumax_amd.kernels import insitu
@@ -125,6 +125,8 @@ def run_case(name, model, seq, mbs, mbc, steps, fp8=False, rc=0, warmup=1):
             predicted_gib=round(mem["max_peak_mem"] / 2**30, 2),
             mem_err_pct=round((mem["max_peak_mem"] - peak) / peak * 100, 2),
             fp8=fp8, recompute_layers=rc,
+            recompute_factor_meas=(overlay or {}).get("meta", {}).get(
+                "recompute_factor"),
             measured_mfu=round(
                 mc.flops_per_token(seq) * mbs * mbc * seq / (ms / 1e3)
                 / 2.5e15, 4),

@@ -14,6 +14,11 @@ def apply_insitu_overlay(system_config, summary) -> int:
     acc = system_config.accelerator
     n = 0
     for table, rows in summary.items():
+        if table == "meta":
+            if "recompute_factor" in rows:
+                acc.recompute_factor = float(rows["recompute_factor"])
+                n += 1
+            continue
         if table == "bandwidth":
             for k, v in rows.items():
                 if k.endswith("_eff"):

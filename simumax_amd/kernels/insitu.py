@@ -43,14 +43,21 @@ def disable():
 
 def start(table: str, key: str):
     """Record an event pair around the op the caller is about to launch.
-    Returns a closure to call right after the launch (same stream)."""
+    Returns a closure to call right after the launch (same stream).
+
+    Each instance is tagged with whether it ran inside a backward graph
+    task: activation-recompute reruns (torch.utils.checkpoint) execute
+    the SAME fwd-shaped ops during .backward(), and summarize() must not
+    let those duplicates shift the fwd keys (the simulator prices the
+    rerun separately via recompute_factor)."""
     s = torch.cuda.Event(enable_timing=True)
     e = torch.cuda.Event(enable_timing=True)
+    in_bwd = torch._C._current_graph_task_id() != -1
     s.record()
 
     def stop():
         e.record()
-        _RECORDS[(table, key)].append((s, e))
+        _RECORDS[(table, key)].append((s, e, in_bwd))
 
     return stop
 
@@ -99,15 +106,49 @@ def summarize():
     torch.cuda.synchronize()
     out = defaultdict(dict)
     bw_acc = defaultdict(lambda: [0.0, 0.0, 0])  # op -> [bytes, ms, n]
+    # measured checkpoint-rerun cost ratio: for every key seen both
+    # outside and inside backward, the in-backward instances are the
+    # recompute reruns — their per-instance time over the fwd instance
+    # time, weighted by rerun count, IS the accelerator recompute_factor
+    rc_wsum = fwd_wsum = 0.0
     for (table, key), pairs in _RECORDS.items():
         if table.startswith("bw_"):
             acc = bw_acc[table[3:]]
-            for s, e in pairs:
-                acc[0] += float(key)
-                acc[1] += s.elapsed_time(e)
-                acc[2] += 1
+            f_ms = f_n = r_ms = r_n = 0
+            for s, e, inb in pairs:
+                t = s.elapsed_time(e)
+                if inb:
+                    r_ms += t
+                    r_n += 1
+                else:
+                    f_ms += t
+                    f_n += 1
+            if f_n and r_n:
+                # mixed context = fwd op rerun under recompute: price the
+                # bandwidth from true-fwd instances only, feed the ratio
+                # into the measured recompute factor
+                fwd_wsum += (f_ms / f_n) * r_n
+                rc_wsum += r_ms
+                acc[0] += float(key) * f_n
+                acc[1] += f_ms
+                acc[2] += f_n
+            else:
+                acc[0] += float(key) * len(pairs)
+                acc[1] += f_ms + r_ms
+                acc[2] += len(pairs)
             continue
-        ts = sorted(s.elapsed_time(e) for s, e in pairs)
+        # a key seen both outside and inside backward is a fwd op whose
+        # checkpoint-recompute rerun shares the shape key: keep only the
+        # true-fwd instances (pure-bwd keys keep everything)
+        fwd_p = [p for p in pairs if not p[2]]
+        rc_p = [p for p in pairs if p[2]]
+        if fwd_p and rc_p:
+            m_f = sorted(s.elapsed_time(e) for s, e, _ in fwd_p)
+            m_r = sorted(s.elapsed_time(e) for s, e, _ in rc_p)
+            fwd_wsum += m_f[len(m_f) // 2] * len(rc_p)
+            rc_wsum += m_r[len(m_r) // 2] * len(rc_p)
+            pairs = fwd_p
+        ts = sorted(s.elapsed_time(e) for s, e, _ in pairs)
         t = ts[len(ts) // 2]  # median instance
         row = dict(t_ms=round(t, 5), n=len(ts))
         if table == "matmul":
@@ -130,6 +171,8 @@ def summarize():
         ms = max(ms - n * 0.004, 1e-6)
         out["bandwidth"][f"{op}_eff"] = round(
             byt / (ms / 1e3) / (HBM_PEAK_GBPS * 1024**3), 4)
+    if fwd_wsum > 0 and rc_wsum > 0:
+        out["meta"]["recompute_factor"] = round(rc_wsum / fwd_wsum, 4)
     return out
 
 

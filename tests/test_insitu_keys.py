@@ -117,7 +117,7 @@ def test_bandwidth_summarize_units():
     I._RECORDS.clear()
     # 4 GiB moved in 1 ms -> 4000 GiB/s -> eff 0.5 of 8000 GiB/s
     I._RECORDS[("bw_rmsnorm_fwd", str(4 * 1024**3))] = [
-        (FakeEvt(0.0), FakeEvt(1.0 + 0.004))]
+        (FakeEvt(0.0), FakeEvt(1.0 + 0.004), False)]
     import torch
 
     sync = torch.cuda.synchronize
