@@ -111,6 +111,8 @@ def summarize():
     # recompute reruns — their per-instance time over the fwd instance
     # time, weighted by rerun count, IS the accelerator recompute_factor
     rc_wsum = fwd_wsum = 0.0
+    saw_compute_rerun = False  # bw_ ops reuse keys across fwd/bwd (e.g.
+    # rope), so only compute-table mixed keys prove a recompute ran
     for (table, key), pairs in _RECORDS.items():
         if table.startswith("bw_"):
             acc = bw_acc[table[3:]]
@@ -147,6 +149,7 @@ def summarize():
             m_r = sorted(s.elapsed_time(e) for s, e, _ in rc_p)
             fwd_wsum += m_f[len(m_f) // 2] * len(rc_p)
             rc_wsum += m_r[len(m_r) // 2] * len(rc_p)
+            saw_compute_rerun = True
             pairs = fwd_p
         ts = sorted(s.elapsed_time(e) for s, e, _ in pairs)
         t = ts[len(ts) // 2]  # median instance
@@ -171,7 +174,7 @@ def summarize():
         ms = max(ms - n * 0.004, 1e-6)
         out["bandwidth"][f"{op}_eff"] = round(
             byt / (ms / 1e3) / (HBM_PEAK_GBPS * 1024**3), 4)
-    if fwd_wsum > 0 and rc_wsum > 0:
+    if saw_compute_rerun and fwd_wsum > 0 and rc_wsum > 0:
         out["meta"]["recompute_factor"] = round(rc_wsum / fwd_wsum, 4)
     return out
 
