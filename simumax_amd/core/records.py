@@ -126,10 +126,14 @@ class ModuleMemoryInfo:
     moe_state_bytes: float = 0.0
     # analog of TE dummy-wgrad workspace shapes (kept for schema parity)
     dummy_wgrad_bytes: float = 0.0
+    # persistent non-parameter caches (fp8 weight-quant copies): counted
+    # in peak memory but NOT in the optimizer/DP param-count derivations
+    # (dense/moe_weight_bytes / 2 = numel must stay true)
+    cache_bytes: float = 0.0
 
     @property
     def weight_bytes(self):
-        return self.dense_weight_bytes + self.moe_weight_bytes
+        return self.dense_weight_bytes + self.moe_weight_bytes + self.cache_bytes
 
     @property
     def grad_bytes(self):
@@ -152,6 +156,7 @@ class ModuleMemoryInfo:
             self.moe_grad_bytes + other.moe_grad_bytes,
             self.moe_state_bytes + other.moe_state_bytes,
             max(self.dummy_wgrad_bytes, other.dummy_wgrad_bytes),
+            self.cache_bytes + other.cache_bytes,
         )
 
 

@@ -15,6 +15,7 @@ void fp8_cast_kernel(const bf16raw *__restrict__ x,
                      float *__restrict__ amax,
                      const float *__restrict__ scale_p,
                      long n, int e5m2, float fmax) {
+    __shared__ float red[CAST_BLOCK / WAVE];
     const float scale = *scale_p;
     float local = 0.f;
     const long stride = (long)gridDim.x * CAST_BLOCK * 8;
@@ -41,10 +42,21 @@ void fp8_cast_kernel(const bf16raw *__restrict__ x,
         *reinterpret_cast<uint2 *>(out + i) =
             *reinterpret_cast<uint2 *>(o);
     }
-    // wave-reduce then one atomic per wave (positive floats order as ints)
+    // block-reduce, then ONE pre-checked atomic per block: thousands of
+    // waves hammering a single dword serialize in L2 and dominated the
+    // whole kernel (measured 0.26 TB/s for a pure streaming cast)
     local = wave_max(local);
-    if ((threadIdx.x % WAVE) == 0 && local > 0.f)
-        atomicMax(reinterpret_cast<int *>(amax), __float_as_int(local));
+    if ((threadIdx.x % WAVE) == 0) red[threadIdx.x / WAVE] = local;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float m = red[0];
+#pragma unroll
+        for (int w = 1; w < CAST_BLOCK / WAVE; ++w) m = fmaxf(m, red[w]);
+        // positive floats order as ints; racy pre-read only skips work
+        if (m > 0.f &&
+            __float_as_int(m) > *reinterpret_cast<volatile int *>(amax))
+            atomicMax(reinterpret_cast<int *>(amax), __float_as_int(m));
+    }
 }
 
 extern "C" void fp8_cast_launch(const void *x, void *out, void *amax,
@@ -62,9 +74,36 @@ extern "C" void fp8_cast_launch(const void *x, void *out, void *amax,
 // fp8 image AND the [N,M] transposed image (wgrad's A / dgrad's B want
 // the other layout; a torch .t().contiguous() on fp8 bytes runs at
 // ~0.5 TB/s uncoalesced and costs more than the GEMM it feeds).
-// 64x64 tiles staged through LDS at byte granularity.
+//
+// The transpose rides the gfx950 16-bit LDS transpose reads: each 64x64
+// bf16 tile is staged as 8 tr-readable sub-images ([32 permuted rows] x
+// [16 cols], the attention V-image layout), so BOTH LDS directions are
+// 8-byte ops — ds_write_b128 staging, ds_read_b64_tr_b16 fragments —
+// and each lane then owns 8 consecutive m-values of one n-column, i.e.
+// one packed 8-byte store into out_t. The byte-granular LDS version of
+// this kernel measured eff 0.08-0.23; this one ~0.6+.
 #define CT_TILE 64
-#define CT_PAD 72
+#define CT_SUB 520            // padded sub-image elems ([32][16] + 8)
+
+DEV int ct_img_row(int key) {
+    const int kg = key >> 3, j = key & 7;
+    return kg * 4 + (j & 3) + ((j >> 2) << 4);
+}
+
+typedef __attribute__((ext_vector_type(4))) __bf16 ct_bf16x4v;
+typedef __attribute__((ext_vector_type(8))) __bf16 ct_bf16x8v;
+typedef __attribute__((address_space(3))) ct_bf16x4v ct_lds_b64_t;
+
+DEV ct_bf16x8v ct_tr_frag(const bf16raw *sub_base, int lane) {
+    ct_bf16x4v lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (ct_lds_b64_t *)(sub_base + lane * 4));
+    ct_bf16x4v hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (ct_lds_b64_t *)(sub_base + 256 + lane * 4));
+    ct_bf16x8v r;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) { r[i] = lo[i]; r[i + 4] = hi[i]; }
+    return r;
+}
 
 __global__ __launch_bounds__(CAST_BLOCK)
 void fp8_cast_t_kernel(const bf16raw *__restrict__ x,
@@ -73,14 +112,19 @@ void fp8_cast_t_kernel(const bf16raw *__restrict__ x,
                        float *__restrict__ amax,
                        const float *__restrict__ scale_p,
                        int M, int N, int e5m2, float fmax) {
-    __shared__ unsigned char tile[CT_TILE * CT_PAD];
+    __shared__ __attribute__((aligned(16))) bf16raw img[8 * CT_SUB];
+    __shared__ float red[CAST_BLOCK / WAVE];
     const float scale = *scale_p;
     const int tiles_n = N / CT_TILE;
+    const int lane = threadIdx.x % WAVE;
+    const int wave = threadIdx.x / WAVE;
     float local = 0.f;
     for (int t = blockIdx.x; t < (M / CT_TILE) * tiles_n; t += gridDim.x) {
         const int tm = (t / tiles_n) * CT_TILE;
         const int tn = (t % tiles_n) * CT_TILE;
         __syncthreads();   // previous tile fully read
+        // phase 1: coalesced load, cast+write the row-major image, stage
+        // the bf16 tile into tr sub-images (sub = n-subtile*2 + m-group)
 #pragma unroll
         for (int it = 0; it < 2; ++it) {
             const int e = (threadIdx.x + it * CAST_BLOCK) * 8;
@@ -103,25 +147,49 @@ void fp8_cast_t_kernel(const bf16raw *__restrict__ x,
             }
             *reinterpret_cast<uint2 *>(out + (long)(tm + r) * N + tn + c) =
                 *reinterpret_cast<uint2 *>(o);
-#pragma unroll
-            for (int j = 0; j < 8; ++j) tile[r * CT_PAD + c + j] = o[j];
+            store8(img + ((c >> 4) * 2 + (r >> 5)) * CT_SUB
+                       + ct_img_row(r & 31) * 16 + (c & 15), v);
         }
         __syncthreads();
-        // transposed write: thread covers out_t[tn + r'][tm + c'..+8]
+        // phase 2: tr fragments hand each lane 8 consecutive m of one
+        // n-column -> one packed 8-byte out_t store per lane per sub
 #pragma unroll
-        for (int it = 0; it < 2; ++it) {
-            const int e = (threadIdx.x + it * CAST_BLOCK) * 8;
-            const int r2 = e / CT_TILE, c2 = e % CT_TILE;
+        for (int i = 0; i < 2; ++i) {
+            const int sub = wave * 2 + i;
+            const int sn = sub >> 1, g = sub & 1;
+            ct_bf16x8v f = ct_tr_frag(img + sub * CT_SUB, lane);
+            bf16x8 vv = *reinterpret_cast<bf16x8 *>(&f);
             unsigned char o[8];
 #pragma unroll
-            for (int j = 0; j < 8; ++j) o[j] = tile[(c2 + j) * CT_PAD + r2];
-            *reinterpret_cast<uint2 *>(out_t + (long)(tn + r2) * M + tm + c2)
-                = *reinterpret_cast<uint2 *>(o);
+            for (int p = 0; p < 4; ++p) {
+                float as = fminf(fmaxf(vv.get(2 * p) * scale, -fmax), fmax);
+                float bs = fminf(fmaxf(vv.get(2 * p + 1) * scale, -fmax), fmax);
+                int packed;
+                if (e5m2)
+                    packed = __builtin_amdgcn_cvt_pk_bf8_f32(as, bs, 0, false);
+                else
+                    packed = __builtin_amdgcn_cvt_pk_fp8_f32(as, bs, 0, false);
+                o[2 * p] = packed & 0xff;
+                o[2 * p + 1] = (packed >> 8) & 0xff;
+            }
+            const int n_l = sn * 16 + (lane & 15);
+            const int m_l = g * 32 + (lane >> 4) * 8;
+            *reinterpret_cast<uint2 *>(
+                out_t + (long)(tn + n_l) * M + tm + m_l) =
+                *reinterpret_cast<uint2 *>(o);
         }
     }
     local = wave_max(local);
-    if ((threadIdx.x % WAVE) == 0 && local > 0.f)
-        atomicMax(reinterpret_cast<int *>(amax), __float_as_int(local));
+    if ((threadIdx.x % WAVE) == 0) red[threadIdx.x / WAVE] = local;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float m = red[0];
+#pragma unroll
+        for (int w = 1; w < CAST_BLOCK / WAVE; ++w) m = fmaxf(m, red[w]);
+        if (m > 0.f &&
+            __float_as_int(m) > *reinterpret_cast<volatile int *>(amax))
+            atomicMax(reinterpret_cast<int *>(amax), __float_as_int(m));
+    }
 }
 
 extern "C" void fp8_cast_t_launch(const void *x, void *out, void *out_t,

@@ -170,10 +170,7 @@ class LinearCol(LinearBase, ParamMixin):
             # per-step weight-quant cache: wq + column-major copy, 1 B/elem
             # each (kernels/fp8.py Fp8Linear._weight_quant)
             nw = self.input_size * self.output_size
-            if self.is_expert:
-                info.moe_weight_bytes += 2 * nw
-            else:
-                info.dense_weight_bytes += 2 * nw
+            info.cache_bytes += 2 * nw
 
     def _leaf_act_info(self, info):
         # caches the (sharded under SP) input; the gathered copy is transient
@@ -269,10 +266,7 @@ class LinearRow(LinearBase, ParamMixin):
         self.add_param(info, self.input_size * self.output_size, self.is_expert)
         if self.strategy.fp8:
             nw = self.input_size * self.output_size
-            if self.is_expert:
-                info.moe_weight_bytes += 2 * nw
-            else:
-                info.dense_weight_bytes += 2 * nw
+            info.cache_bytes += 2 * nw
 
     def _leaf_act_info(self, info):
         info.activation_mem_cache = self.input_info.first.mem_bytes()
