@@ -102,7 +102,8 @@ def main():
             acc["bandwidth"]["ce"]["efficient_factor"] = acc["bandwidth"][
                 "ce_fusion"]["efficient_factor"]
         for key in ("permute_fwd", "permute_bwd", "rmsnorm_fwd",
-                    "rmsnorm_bwd", "rope", "swiglu", "swiglu_bwd"):
+                    "rmsnorm_bwd", "rope", "swiglu", "swiglu_bwd",
+                    "fp8_quant"):
             if f"{key}_eff" in bw:
                 acc["bandwidth"].setdefault(key, {
                     "gbps": 8000.0, "efficient_factor": 0.55,
