@@ -179,3 +179,103 @@ def test_insitu_overlay_applies():
     assert sysc.accelerator.bandwidth["optimizer"].efficient_factor == 0.4321
     assert sysc.accelerator.bandwidth["rmsnorm_fwd"].efficient_factor == 0.9
     assert sysc.accelerator.bandwidth["moe_routing"].latency_us == 500.0
+
+
+def _fake_records(I):
+    class FakeEvt:
+        def __init__(self, t):
+            self.t = t
+
+        def elapsed_time(self, other):
+            return other.t - self.t
+
+    return FakeEvt
+
+
+def _summarize_nosync(I):
+    import torch
+
+    sync = torch.cuda.synchronize
+    torch.cuda.synchronize = lambda: None
+    try:
+        return I.summarize()
+    finally:
+        torch.cuda.synchronize = sync
+        I._RECORDS.clear()
+
+
+def test_recompute_rerun_filtering_and_factor():
+    """A compute key seen both outside and inside backward is a
+    checkpoint rerun: the fwd efficiency median must come from the
+    true-fwd instances only, and the rerun/fwd time ratio becomes the
+    measured recompute_factor in the meta table."""
+    import simumax_amd.kernels.insitu as I
+
+    FakeEvt = _fake_records(I)
+    I._RECORDS.clear()
+    key = I.gemm_key(1, 4096, 4096, 4096, "TN", False, "bf16")
+    # 2 fwd instances at 1.0 ms, 2 rerun instances (in backward) at 0.8
+    I._RECORDS[("matmul", key)] = [
+        (FakeEvt(0.0), FakeEvt(1.0), False),
+        (FakeEvt(0.0), FakeEvt(1.0), False),
+        (FakeEvt(0.0), FakeEvt(0.8), True),
+        (FakeEvt(0.0), FakeEvt(0.8), True),
+    ]
+    out = _summarize_nosync(I)
+    row = out["matmul"][key]
+    assert abs(row["t_ms"] - 1.0) < 1e-9      # fwd median, reruns excluded
+    assert row["n"] == 2
+    assert abs(out["meta"]["recompute_factor"] - 0.8) < 1e-3
+
+
+def test_no_meta_factor_from_bw_only_mix():
+    """bw_ ops legitimately reuse one key across fwd and bwd (e.g. rope);
+    without a compute-table mixed key no recompute_factor is emitted and
+    the bandwidth efficiency uses the fwd instances only."""
+    import simumax_amd.kernels.insitu as I
+
+    FakeEvt = _fake_records(I)
+    I._RECORDS.clear()
+    byt = str(4 * 1024**3)
+    I._RECORDS[("bw_rope", byt)] = [
+        (FakeEvt(0.0), FakeEvt(1.0 + 0.004), False),   # fwd: eff 0.5
+        (FakeEvt(0.0), FakeEvt(2.0 + 0.004), True),    # bwd: slower
+    ]
+    out = _summarize_nosync(I)
+    assert "meta" not in out
+    assert abs(out["bandwidth"]["rope_eff"] - 0.5) < 1e-3
+
+
+def test_fp8_cache_bytes_not_in_param_count():
+    """fp8 weight-quant caches count toward peak memory but must not
+    inflate the optimizer/DP param-count derivations (which read
+    dense/moe_weight_bytes / 2 as numel)."""
+    import copy
+
+    from simumax_amd import (ModelConfig, PerfLLM, StrategyConfig,
+                             SystemConfig, get_simu_model_config,
+                             get_simu_system_config)
+
+    mc = ModelConfig.init_from_config_file(get_simu_model_config("llama3-8b"))
+    res = {}
+    for fp8 in (False, True):
+        st = StrategyConfig(
+            seq_len=2048, micro_batch_size=1, micro_batch_num=1,
+            world_size=1, tp_size=1, pp_size=1, fp8=fp8,
+            enable_recompute=False, enable_sequence_parallel=False,
+            zero_state=0, use_fp32_accum_grad=True,
+            cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
+            mem_factor=1.0)
+        sysc = SystemConfig.init_from_config_file(
+            get_simu_system_config("mi355x"))
+        p = PerfLLM()
+        p.configure(st, copy.deepcopy(mc), sysc)
+        p.run_estimate()
+        mi = p.chunks[0].get_model_info()
+        res[fp8] = (p.analysis_cost()["optim_time"], mi.dense_weight_bytes,
+                    mi.cache_bytes, mi.weight_bytes)
+    # optimizer traffic identical; caches only in cache_bytes/weight_bytes
+    assert abs(res[True][0] - res[False][0]) < 1e-6
+    assert res[True][1] == res[False][1]
+    assert res[True][2] > 0 and res[False][2] == 0
+    assert res[True][3] > res[False][3]
