@@ -19,3 +19,18 @@ from .registry import (  # noqa: F401
     show_simu_strategy_configs,
     show_simu_system_configs,
 )
+
+
+def config_from_names(model: str, strategy: str, system: str):
+    """Resolve registry names to the (strategy, model, system) config
+    triple in `PerfLLM.configure` argument order:
+
+        perf = PerfLLM()
+        perf.configure(*config_from_names("llama3-8b",
+                                          "tp1_pp2_dp4_mbs1", "mi355x"))
+    """
+    return (
+        StrategyConfig.init_from_config_file(get_simu_strategy_config(strategy)),
+        ModelConfig.init_from_config_file(get_simu_model_config(model)),
+        SystemConfig.init_from_config_file(get_simu_system_config(system)),
+    )
