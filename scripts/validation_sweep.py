@@ -86,26 +86,24 @@ def run_case(name, model, seq, mbs, mbc, steps, fp8=False, rc=0, warmup=1):
         # Recompute rows participate too: insitu tags instances recorded
         # inside the backward graph task, so checkpoint reruns cannot
         # shift the fwd keys (recompute_factor prices the rerun).
-        overlay = None
-        if True:
-            from simumax_amd.core.consts import \
-                OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
-            from simumax_amd.kernels import insitu
+        from simumax_amd.core.consts import \
+            OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+        from simumax_amd.kernels import insitu
 
-            insitu.enable()
-            train_step(m, opt, red, toks, labels, mbc)
-            torch.cuda.synchronize()
-            insitu.disable()
-            overlay = insitu.summarize()
-            s_ev = torch.cuda.Event(enable_timing=True)
-            e_ev = torch.cuda.Event(enable_timing=True)
-            s_ev.record()
-            opt.step()
-            e_ev.record()
-            torch.cuda.synchronize()
-            overlay.setdefault("bandwidth", {})["optimizer_eff"] = (
-                opt.flat_grad.numel() * OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
-                / (s_ev.elapsed_time(e_ev) / 1e3) / (8000.0 * 1024**3))
+        insitu.enable()
+        train_step(m, opt, red, toks, labels, mbc)
+        torch.cuda.synchronize()
+        insitu.disable()
+        overlay = insitu.summarize()
+        s_ev = torch.cuda.Event(enable_timing=True)
+        e_ev = torch.cuda.Event(enable_timing=True)
+        s_ev.record()
+        opt.step()
+        e_ev.record()
+        torch.cuda.synchronize()
+        overlay.setdefault("bandwidth", {})["optimizer_eff"] = (
+            opt.flat_grad.numel() * OPTIMIZER_TRAFFIC_BYTES_PER_PARAM
+            / (s_ev.elapsed_time(e_ev) / 1e3) / (8000.0 * 1024**3))
         torch.cuda.reset_peak_memory_stats()
         torch.cuda.synchronize()
         t1 = time.time()
