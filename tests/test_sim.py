@@ -238,3 +238,27 @@ def test_trace_flow_arrows(tmp_path):
     for e in flows:
         by_id.setdefault(e["id"], []).append(e["pid"])
     assert all(len(set(v)) == 2 for v in by_id.values())
+
+
+def test_sim_vs_perf_agreement_moe_ep(tmp_path):
+    """EP all-to-all dispatch/combine events replay through the simulator
+    to the analytic iteration time (MoE path: router, permutation, grouped
+    GEMM, unpermutation)."""
+    p = build(strategy="ep8_pp1_dp8_mbs1", model="mixtral-8x7b-l8")
+    cost = p.analysis_cost()
+    res = p.simulate(str(tmp_path))
+    analytic = (cost["pipeline_time"] / cost["straggler_ratio"]
+                + cost["dp_time"] + cost["optim_time"])
+    assert res["total_time"] == pytest.approx(analytic, rel=0.10)
+
+
+def test_sim_vs_perf_agreement_cp(tmp_path):
+    """Ulysses CP a2a events (q/k/v pre + o post, mirrored in bwd) replay
+    consistently."""
+    p = build(strategy="tp1_pp1_dp8_mbs1", model="llama2-tiny",
+              cp_size=2)
+    cost = p.analysis_cost()
+    res = p.simulate(str(tmp_path))
+    analytic = (cost["pipeline_time"] / cost["straggler_ratio"]
+                + cost["dp_time"] + cost["optim_time"])
+    assert res["total_time"] == pytest.approx(analytic, rel=0.10)
