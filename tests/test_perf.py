@@ -360,3 +360,41 @@ def test_search_best_parallel_strategy_with_recompute_probe():
     assert best is not None
     assert best["mfu"] > 0
     assert best["tp"] * best["pp"] * best["dp"] == 8
+
+
+def test_multi_node_tier_selection_and_nic_model():
+    """Beyond one 8-GPU node the DP/EP groups price on the inter_node
+    tier (NIC bandwidth, shared per node), and a 2-node run is strictly
+    slower per step than the single-node run of the same per-GPU work
+    (weak scaling with a slower fabric)."""
+    import copy
+
+    from simumax_amd import (ModelConfig, PerfLLM, StrategyConfig,
+                             SystemConfig, get_simu_model_config,
+                             get_simu_system_config)
+
+    mc = ModelConfig.init_from_config_file(get_simu_model_config("llama2-tiny"))
+
+    def run(world):
+        st = StrategyConfig(
+            seq_len=512, micro_batch_size=1, micro_batch_num=1,
+            world_size=world, tp_size=1, pp_size=1,
+            enable_sequence_parallel=False, zero_state=1,
+            use_fp32_accum_grad=True, overlap_grad_reduce=False,
+            cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
+            mem_factor=1.0)
+        p = PerfLLM()
+        p.configure(st, copy.deepcopy(mc), SystemConfig.init_from_config_file(
+            get_simu_system_config("mi355x")))
+        p.run_estimate()
+        return p
+
+    p8 = run(8)     # one node: dp on high_intra_node (xGMI)
+    p16 = run(16)   # two nodes: dp crosses the NIC
+    c8, c16 = p8.analysis_cost(), p16.analysis_cost()
+    assert c16["dp_time"] > c8["dp_time"], (c8["dp_time"], c16["dp_time"])
+    # the dp group's tier must actually be inter_node at world 16
+    nets = p16.analysis_net()
+    dp_tier = next((v for k, v in nets.items() if k.startswith("dp")), None)
+    if isinstance(dp_tier, str):
+        assert "inter" in dp_tier
