@@ -58,7 +58,13 @@ class HumanReadableSize:
             return val
         if unit.upper().endswith("B") and unit.upper() in HumanReadableSize.BYTE_UNITS:
             return val * 1024.0 ** HumanReadableSize.BYTE_UNITS.index(unit.upper())
-        metric = unit.rstrip("FLOPSflops")
+        # strip a whole FLOPS/TGS-style suffix — a char-set rstrip would
+        # also eat the P of PFLOPS and mis-scale peta quantities by 1e15
+        metric = unit
+        for suf in ("FLOPS", "FLOPs", "Flops", "flops", "TGS", "tgs"):
+            if metric.endswith(suf):
+                metric = metric[: -len(suf)]
+                break
         if metric.upper() in ("", "K", "M", "G", "T", "P", "E"):
             return val * 1000.0 ** (
                 HumanReadableSize.METRIC_UNITS.index(metric.upper())
