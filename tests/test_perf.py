@@ -398,3 +398,46 @@ def test_multi_node_tier_selection_and_nic_model():
     dp_tier = next((v for k, v in nets.items() if k.startswith("dp")), None)
     if isinstance(dp_tier, str):
         assert "inter" in dp_tier
+
+
+def test_mem_model_invariants():
+    """Peak >= static (weights+grads+state); fp8 adds resident quant
+    caches to static memory while HALVING the fwd activation caches; the
+    analysis_mem strings parse back through HumanReadableSize (the
+    search path depends on that quirk)."""
+    import copy
+
+    from simumax_amd import (ModelConfig, PerfLLM, StrategyConfig,
+                             SystemConfig, get_simu_model_config,
+                             get_simu_system_config)
+    from simumax_amd.core.utils import HumanReadableSize as H
+
+    mc = ModelConfig.init_from_config_file(get_simu_model_config("llama2-tiny"))
+
+    def run(fp8):
+        st = StrategyConfig(
+            seq_len=1024, micro_batch_size=1, micro_batch_num=2,
+            world_size=1, tp_size=1, pp_size=1, fp8=fp8,
+            enable_sequence_parallel=False, zero_state=0,
+            use_fp32_accum_grad=True, cross_entropy_loss_fusion=True,
+            attention_sparse_ratio=0.5, mem_factor=1.0)
+        p = PerfLLM()
+        p.configure(st, copy.deepcopy(mc), SystemConfig.init_from_config_file(
+            get_simu_system_config("mi355x")))
+        p.run_estimate()
+        return p
+
+    p_bf, p_f8 = run(False), run(True)
+    m_bf, m_f8 = p_bf.analysis_mem(), p_f8.analysis_mem()
+    mi_bf = p_bf.chunks[0].get_model_info()
+    mi_f8 = p_f8.chunks[0].get_model_info()
+
+    static_bf = mi_bf.weight_bytes + mi_bf.grad_bytes + mi_bf.state_bytes
+    assert m_bf["max_peak_mem"] >= static_bf
+    # fp8 static grows by exactly the resident quant caches
+    assert mi_f8.weight_bytes - mi_bf.weight_bytes == mi_f8.cache_bytes > 0
+    # per-stage mem report strings parse back to bytes
+    for row in m_bf.get("stage_mem", [m_bf]):
+        for key, val in (row.items() if isinstance(row, dict) else []):
+            if isinstance(val, str) and val.endswith("B"):
+                assert H.from_string(val) >= 0
