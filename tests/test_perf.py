@@ -441,3 +441,41 @@ def test_mem_model_invariants():
         for key, val in (row.items() if isinstance(row, dict) else []):
             if isinstance(val, str) and val.endswith("B"):
                 assert H.from_string(val) >= 0
+
+
+def test_dp_overlap_exposure_model():
+    """Bucketed DP all-reduce: raw cost DECREASES with world size on the
+    FC8 xGMI mesh (more participating links), and only the part that
+    does not fit under the last microbatch's backward is exposed
+    (Megatron no_sync semantics)."""
+    import copy
+
+    from simumax_amd import (ModelConfig, PerfLLM, StrategyConfig,
+                             SystemConfig, get_simu_model_config,
+                             get_simu_system_config)
+
+    mc = ModelConfig.init_from_config_file(get_simu_model_config("llama3-8b"))
+
+    def run(world, overlap=True):
+        st = StrategyConfig(
+            seq_len=4096, micro_batch_size=1, micro_batch_num=4,
+            world_size=world, tp_size=1, pp_size=1,
+            enable_sequence_parallel=False, zero_state=0,
+            use_fp32_accum_grad=True, overlap_grad_reduce=overlap,
+            cross_entropy_loss_fusion=True, attention_sparse_ratio=0.5,
+            mem_factor=1.0)
+        p = PerfLLM()
+        p.configure(st, copy.deepcopy(mc), SystemConfig.init_from_config_file(
+            get_simu_system_config("mi355x")))
+        p.run_estimate()
+        return p.analysis_cost()
+
+    c2, c4, c8 = run(2), run(4), run(8)
+    assert c2["dp_time_raw"] > c4["dp_time_raw"] > c8["dp_time_raw"] > 0
+    for c in (c2, c4, c8):
+        assert c["dp_time"] <= c["dp_time_raw"]
+        assert c["dp_time"] >= 0
+    # without overlap the whole reduce is exposed
+    c8_no = run(8, overlap=False)
+    assert c8_no["dp_time"] == pytest.approx(c8_no["dp_time_raw"])
+    assert c8_no["iter_time"] > c8["iter_time"]
