@@ -120,3 +120,31 @@ def test_rccl_fit_bw_latency():
     bw, lat = fit_bw_latency(rows, scale, offset, n)
     assert bw == pytest.approx(900, rel=0.01)
     assert lat == pytest.approx(lat_true_ms, rel=0.05)
+
+
+def test_sanity_check_rejects_bad_configs():
+    """Invalid parallel layouts must fail loudly at config time, not as
+    a shape error deep inside estimate()."""
+    import pytest
+
+    # world not divisible by tp*pp
+    s = StrategyConfig(seq_len=4096, micro_batch_size=1, micro_batch_num=1,
+                       world_size=8, tp_size=3, pp_size=1)
+    with pytest.raises((AssertionError, ValueError)):
+        s.sanity_check()
+    # unknown dtype
+    s = StrategyConfig(seq_len=4096, micro_batch_size=1, micro_batch_num=1,
+                       world_size=8, tp_size=1, pp_size=1, dtype="int7")
+    with pytest.raises((AssertionError, ValueError)):
+        s.sanity_check()
+    # SP sequence shard must divide seq_len
+    s = StrategyConfig(seq_len=4098, micro_batch_size=1, micro_batch_num=1,
+                       world_size=8, tp_size=4, pp_size=1,
+                       enable_sequence_parallel=True)
+    with pytest.raises((AssertionError, ValueError)):
+        s.sanity_check()
+    # ep cannot exceed world/(tp*pp)
+    s = StrategyConfig(seq_len=4096, micro_batch_size=1, micro_batch_num=1,
+                       world_size=4, tp_size=2, pp_size=2, ep_size=4)
+    with pytest.raises((AssertionError, ValueError, ZeroDivisionError)):
+        s.sanity_check()
