@@ -262,3 +262,16 @@ def test_sim_vs_perf_agreement_cp(tmp_path):
     analytic = (cost["pipeline_time"] / cost["straggler_ratio"]
                 + cost["dp_time"] + cost["optim_time"])
     assert res["total_time"] == pytest.approx(analytic, rel=0.10)
+
+
+def test_sim_vs_perf_agreement_pp4_uneven(tmp_path):
+    """Deeper pipeline with uneven first/last stage layer counts replays
+    to the analytic estimate."""
+    p = build(strategy="tp1_pp2_dp4_mbs1", model="llama2-tiny",
+              pp_size=4, num_layers_in_first_pipeline_stage=1,
+              num_layers_in_last_pipeline_stage=1)
+    cost = p.analysis_cost()
+    res = p.simulate(str(tmp_path))
+    analytic = (cost["pipeline_time"] / cost["straggler_ratio"]
+                + cost["dp_time"] + cost["optim_time"])
+    assert res["total_time"] == pytest.approx(analytic, rel=0.10)
