@@ -268,8 +268,7 @@ def test_sim_vs_perf_agreement_pp4_uneven(tmp_path):
     """Deeper pipeline with uneven first/last stage layer counts replays
     to the analytic estimate."""
     p = build(strategy="tp1_pp2_dp4_mbs1", model="llama2-tiny",
-              pp_size=4, num_layers_in_first_pipeline_stage=1,
-              num_layers_in_last_pipeline_stage=1)
+              pp_size=2, num_layers_in_first_pipeline_stage=1)  # 1/3 split
     cost = p.analysis_cost()
     res = p.simulate(str(tmp_path))
     analytic = (cost["pipeline_time"] / cost["straggler_ratio"]
