@@ -42,6 +42,9 @@ def parse_args():
                     help="pipeline-parallel degree (1F1B)")
     ap.add_argument("--cp", type=int, default=1,
                     help="Ulysses context-parallel degree (seq sharded)")
+    ap.add_argument("--cp-comm-type", default="a2a",
+                    choices=["a2a", "all_gather"],
+                    help="CP mode: Ulysses a2a (flash path) or kv all_gather")
     ap.add_argument("--fp8", action="store_true",
                     help="fp8 (e4m3/e5m2) decoder linears via _scaled_mm")
     ap.add_argument("--no-self-calibrate", action="store_true",
@@ -61,7 +64,7 @@ def predict(model_cfg, world, args, overlay=None):
         micro_batch_num=args.mbc,
         world_size=world,
         tp_size=args.tp, pp_size=args.pp, ep_size=1, cp_size=args.cp,
-        fp8=args.fp8,
+        cp_comm_type=args.cp_comm_type, fp8=args.fp8,
         enable_sequence_parallel=False,
         zero_state=0,                # trainer replicates optimizer state
         use_fp32_accum_grad=True,
@@ -118,7 +121,8 @@ def main():
 
     tc = TrainConfig(seq_len=args.seq_len, micro_batch_size=args.mbs,
                      micro_batch_num=args.mbc, tp_size=args.tp,
-                     pp_size=args.pp, cp_size=args.cp, fp8=args.fp8)
+                     pp_size=args.pp, cp_size=args.cp,
+                     cp_comm_type=args.cp_comm_type, fp8=args.fp8)
     device = f"cuda:{local_rank}"
     t0 = time.time()
     ps = None

@@ -117,3 +117,62 @@ def cp_post_attention(x, group):
     if group is None:
         return x
     return _SeqScatterHeadGather.apply(x, group)
+
+
+# ---- kv all-gather mode (cp_comm_type="all_gather") ----------------------
+# The simulator prices this mode (ops/dense.py CoreAttention else-branch:
+# ag(k+v) fwd, ag + rs in bwd); the reference models the comm but raises
+# NotImplementedError in its flops path (dense_module.py:1521-1524) — here
+# it is executable end to end. Each rank keeps its q shard and attends to
+# the full gathered K/V with an offset-causal mask.
+
+
+class _AllGatherSeq(torch.autograd.Function):
+    """[B, s, Hkv, d] -> [B, s*cp, Hkv, d] along seq; backward sums the
+    full-length gradient over cp and returns this rank's slice (the
+    reduce_scatter of the cost model, expressed backend-portably)."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        cp = dist.get_world_size(group)
+        gathered = [torch.empty_like(x) for _ in range(cp)]
+        dist.all_gather(gathered, x.contiguous(), group=group)
+        return torch.cat(gathered, dim=1)
+
+    @staticmethod
+    def backward(ctx, dy):
+        group = ctx.group
+        cp = dist.get_world_size(group)
+        me = dist.get_rank(group)
+        dy = dy.contiguous()
+        dist.all_reduce(dy, group=group)
+        s = dy.shape[1] // cp
+        return dy[:, me * s:(me + 1) * s].contiguous(), None
+
+
+def cp_allgather_kv(x, group):
+    if group is None:
+        return x
+    return _AllGatherSeq.apply(x, group)
+
+
+def offset_causal_sdp(q, k, v, q_offset):
+    """Math SDP with a GLOBAL causal mask for a seq-sharded q block:
+    query i (global position q_offset+i) attends keys 0..q_offset+i.
+    fp32 softmax; GQA via head repetition. q [B,s,H,d], k/v [B,S,Hkv,d]."""
+    B, s, H, d = q.shape
+    S, Hkv = k.shape[1], k.shape[2]
+    rep = H // Hkv
+    kx = k.repeat_interleave(rep, dim=2) if rep > 1 else k
+    vx = v.repeat_interleave(rep, dim=2) if rep > 1 else v
+    scores = torch.einsum("bqhd,bkhd->bhqk", q.float(), kx.float())
+    scores *= d ** -0.5
+    qpos = torch.arange(q_offset, q_offset + s, device=q.device)
+    kpos = torch.arange(S, device=q.device)
+    scores = scores.masked_fill(kpos[None, None, None, :]
+                                > qpos[None, None, :, None],
+                                float("-inf"))
+    p = torch.softmax(scores, dim=-1)
+    out = torch.einsum("bhqk,bkhd->bqhd", p, vx.float())
+    return out.to(q.dtype)

@@ -26,18 +26,21 @@ class GQAAttention(nn.Module):
 
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
                  tp_group=None, tp_size=1, sp=False, cp_group=None,
-                 cp_size=1, fp8=False):
+                 cp_size=1, cp_rank=0, cp_comm_type="a2a", fp8=False):
         super().__init__()
         h = cfg.hidden_size
         assert cfg.head_num % tp_size == 0 and cfg.kv_head_num % tp_size == 0
         self.heads = cfg.head_num // tp_size
         self.kv_heads = cfg.kv_head_num // tp_size
-        if cp_size > 1:
+        if cp_size > 1 and cp_comm_type == "a2a":
             # Ulysses CP: heads scattered over cp inside attention
+            # (all_gather mode keeps heads whole — no divisibility rule)
             assert self.heads % cp_size == 0 and self.kv_heads % cp_size == 0
         self.head_size = cfg.head_size
         self.tp_group = tp_group
         self.cp_group = cp_group
+        self.cp_rank = cp_rank
+        self.cp_comm_type = cp_comm_type
         self.sp = sp
         qkv_out = (self.heads + 2 * self.kv_heads) * cfg.head_size
         Lin = K.FusedLinear
@@ -67,7 +70,16 @@ class GQAAttention(nn.Module):
         q = q.view(B, S, self.heads, d)
         k = k.view(B, S, self.kv_heads, d)
         v = v.reshape(B, S, self.kv_heads, d)
-        if self.cp_group is not None:
+        if self.cp_group is not None and self.cp_comm_type == "all_gather":
+            # kv all_gather: q stays seq-sharded; K/V are gathered to the
+            # full sequence and attention runs with an offset-causal mask
+            # (math SDP — the a2a mode is the flash-kernel path)
+            from .cp import cp_allgather_kv, offset_causal_sdp
+
+            k_full = cp_allgather_kv(k, self.cp_group)
+            v_full = cp_allgather_kv(v, self.cp_group)
+            ctx = offset_causal_sdp(q, k_full, v_full, self.cp_rank * S)
+        elif self.cp_group is not None:
             # a2a: scatter heads / gather sequence (Ulysses), flash on the
             # full sequence, inverse a2a on the context
             from .cp import cp_post_attention, cp_pre_attention
@@ -146,7 +158,8 @@ class MLAAttention(nn.Module):
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
                  layer_idx=0, ep_group=None, ep_size=1, tp_group=None,
-                 tp_size=1, sp=False, cp_group=None, cp_size=1, fp8=False):
+                 tp_size=1, sp=False, cp_group=None, cp_size=1, cp_rank=0,
+                 cp_comm_type="a2a", fp8=False):
         super().__init__()
         h = cfg.hidden_size
         self.tp_group = tp_group
@@ -160,7 +173,8 @@ class LlamaDecoderLayer(nn.Module):
             self.attention = GQAAttention(cfg, dtype=dtype, device=device,
                                           tp_group=tp_group, tp_size=tp_size,
                                           sp=sp, cp_group=cp_group,
-                                          cp_size=cp_size, fp8=fp8)
+                                          cp_size=cp_size, cp_rank=cp_rank,
+                                          cp_comm_type=cp_comm_type, fp8=fp8)
         if sp:
             # SP norms see only the local seq shard: their weight grads
             # are partial sums and need a tp all_reduce (reducer handles
@@ -225,8 +239,8 @@ class LlamaForTraining(nn.Module):
     def __init__(self, cfg: ModelConfig, seq_len: int, dtype=torch.bfloat16,
                  rope_base=500000.0, device=None, ep_group=None, ep_size=1,
                  tp_group=None, tp_size=1, tp_rank=0, sp=False,
-                 cp_group=None, cp_rank=0, cp_size=1, fp8=False,
-                 recompute_layers=0):
+                 cp_group=None, cp_rank=0, cp_size=1, cp_comm_type="a2a",
+                 fp8=False, recompute_layers=0):
         super().__init__()
         self.cfg = cfg
         self.seq_len = seq_len          # FULL sequence (rope cache size)
@@ -246,7 +260,8 @@ class LlamaForTraining(nn.Module):
                                ep_group=ep_group, ep_size=ep_size,
                                tp_group=tp_group, tp_size=tp_size,
                                sp=sp and tp_size > 1, cp_group=cp_group,
-                               cp_size=cp_size, fp8=fp8)
+                               cp_size=cp_size, cp_rank=cp_rank,
+                               cp_comm_type=cp_comm_type, fp8=fp8)
              for i in range(cfg.layer_num)])
         self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
         if self.sp:

@@ -37,7 +37,8 @@ class TrainConfig:
     ep_size: int = 1
     tp_size: int = 1
     pp_size: int = 1
-    cp_size: int = 1   # Ulysses context parallel (seq_len = FULL sequence)
+    cp_size: int = 1   # context parallel (seq_len = FULL sequence)
+    cp_comm_type: str = "a2a"   # "a2a" (Ulysses, flash path) | "all_gather"
     fp8: bool = False  # e4m3/e5m2 GEMMs via _scaled_mm (decoder linears)
     recompute_layers: int = 0  # full-block activation recompute for the
                                # first N layers (torch.utils.checkpoint;
@@ -409,7 +410,8 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                              tp_group=tp_group, tp_size=tp_size,
                              tp_rank=tp_rank, sp=cfg.sequence_parallel,
                              cp_group=cp_group, cp_rank=cp_rank,
-                             cp_size=cfg.cp_size, fp8=cfg.fp8,
+                             cp_size=cfg.cp_size,
+                             cp_comm_type=cfg.cp_comm_type, fp8=cfg.fp8,
                              recompute_layers=cfg.recompute_layers)
     # ZeRO-1 shards the fp32 optimizer state over the DATA-parallel group
     # (Megatron distributed optimizer); with tp > 1 that is dp_group, not

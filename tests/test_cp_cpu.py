@@ -22,7 +22,7 @@ def _tiny_cfg():
                        vocab_size=512, use_swiglu=True)
 
 
-def _worker(rank, world, port, q):
+def _worker(rank, world, port, q, cp_comm_type="a2a"):
     import torch.distributed as dist
 
     sys.path.insert(0, REPO)
@@ -39,7 +39,8 @@ def _worker(rank, world, port, q):
         cfg = _tiny_cfg()
         S = 64
         tc = TrainConfig(seq_len=S, micro_batch_size=2, micro_batch_num=1,
-                         overlap_grad_reduce=False, cp_size=2)
+                         overlap_grad_reduce=False, cp_size=2,
+                         cp_comm_type=cp_comm_type)
         model, opt, red = build_trainer(cfg, tc, "cpu")
 
         # single-process full-sequence reference (identical init seed)
@@ -82,12 +83,11 @@ def _worker(rank, world, port, q):
         dist.destroy_process_group()
 
 
-@pytest.mark.timeout(300)
-def test_cp2_matches_single_process():
+def _run_cp(port, cp_comm_type):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29631
-    ps = [ctx.Process(target=_worker, args=(r, 2, port, q)) for r in range(2)]
+    ps = [ctx.Process(target=_worker, args=(r, 2, port, q, cp_comm_type))
+          for r in range(2)]
     for p in ps:
         p.start()
     results = [q.get(timeout=280) for _ in range(2)]
@@ -96,3 +96,15 @@ def test_cp2_matches_single_process():
     for rank, lerr, gerr in results:
         assert lerr < 2e-3, f"rank {rank} loss err {lerr}"
         assert gerr < 0.06, f"rank {rank} grad rel err {gerr}"
+
+
+@pytest.mark.timeout(300)
+def test_cp2_matches_single_process():
+    _run_cp(29631, "a2a")
+
+
+@pytest.mark.timeout(300)
+def test_cp2_all_gather_matches_single_process():
+    """kv all_gather mode: q stays seq-sharded, K/V gathered, offset-
+    causal mask — the mode the reference prices but cannot execute."""
+    _run_cp(29634, "all_gather")
