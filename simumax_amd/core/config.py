@@ -371,7 +371,7 @@ class StrategyConfig(Config):
             )
         for m in self.megatron_recompute_module_set:
             assert m in VALID_MEGATRON_RECOMPUTE_MODULES, m
-        assert self.cp_comm_type in ("a2a", "all_gather")
+        assert self.cp_comm_type in ("a2a", "all_gather", "ring")
         assert self.cp_a2a_mode in VALID_CP_A2A_MODES
         if self.enable_sequence_parallel:
             assert self.seq_len % (self.tp_size * self.cp_size) == 0

@@ -404,6 +404,7 @@ class CoreAttention(MetaModule):
         self.use_flash = strategy.use_flash_sdp
         self.cp = strategy.cp_size
         self.cp_a2a = strategy.cp_comm_type == "a2a"
+        self.cp_mode = strategy.cp_comm_type if self.cp > 1 else None
 
     # ---- geometry ------------------------------------------------------
     @property
@@ -415,7 +416,14 @@ class CoreAttention(MetaModule):
     @property
     def full_seq(self):
         b, s = self._bsd
-        return s * self.cp if self.cp_a2a else s * self.cp
+        return s * self.cp
+
+    @property
+    def q_seq(self):
+        """query rows per rank: a2a gathers the full sequence (with H/cp
+        heads); all_gather/ring keep q seq-sharded with full heads."""
+        b, s = self._bsd
+        return s * self.cp if self.cp_a2a else s
 
     @property
     def sdp_head_num(self):
@@ -448,14 +456,16 @@ class CoreAttention(MetaModule):
         )
 
     def _sdp_bytes(self):
+        """Kernel-visible tensor bytes per rank: q/o/lse span q_seq rows,
+        k/v span the full (gathered or circulated) sequence."""
         b, _ = self._bsd
         s = self.full_seq
         e = self.element_size
-        q = b * s * self.sdp_head_num * self.qk_head_dim * e
+        q = b * self.q_seq * self.sdp_head_num * self.qk_head_dim * e
         k = b * s * self.sdp_kv_head_num * self.qk_head_dim * e
         v = b * s * self.sdp_kv_head_num * self.v_head_dim * e
-        o = b * s * self.sdp_head_num * self.v_head_dim * e
-        lse = b * s * self.sdp_head_num * LSE_BYTES
+        o = b * self.q_seq * self.sdp_head_num * self.v_head_dim * e
+        lse = b * self.q_seq * self.sdp_head_num * LSE_BYTES
         return q, k, v, o, lse
 
     def _leaf_act_info(self, info):
@@ -482,16 +492,30 @@ class CoreAttention(MetaModule):
         else:
             b, _ = self._bsd
             s = self.full_seq
-            scores = b * self.sdp_head_num * s * s * self.element_size
+            scores = b * self.sdp_head_num * self.q_seq * s * self.element_size
             info.activation_mem_cache = q + k + v + scores
             info.fwd_peak_mem_no_cache = scores
+        if self.cp_mode == "ring":
+            # blockwise ring (flash assumption): each rank caches only its
+            # OWN K/V shard — blocks re-circulate in backward (the 2x bwd
+            # p2p is priced) — and holds at most two in-flight blocks
+            info.activation_mem_cache -= (k + v) * (self.cp - 1) / self.cp
+            info.fwd_peak_mem_no_cache = max(
+                info.fwd_peak_mem_no_cache, 2 * (k + v) / self.cp)
 
     def _leaf_compute_info(self, info):
         b, _ = self._bsd
         s = self.full_seq
-        sparse = 1.0 - self.strategy.attention_sparse_ratio
-        qk = 2 * b * self.sdp_head_num * s * s * self.qk_head_dim
-        pv = 2 * b * self.sdp_head_num * s * s * self.v_head_dim
+        r = self.strategy.attention_sparse_ratio
+        if self.cp_mode in ("all_gather", "ring"):
+            # seq-sharded q against the full K/V: the WORST rank (last
+            # contiguous shard) sees a causal share of 1 - r/cp of the
+            # keys (reduces to 1-r at cp=1); the step is bounded by it
+            sparse = 1.0 - r / self.cp
+        else:
+            sparse = 1.0 - r
+        qk = 2 * b * self.sdp_head_num * self.q_seq * s * self.qk_head_dim
+        pv = 2 * b * self.sdp_head_num * self.q_seq * s * self.v_head_dim
         info.fwd_flops = (qk + pv) * sparse
         extra = 1 if self.use_flash else 0  # flash bwd recomputes QK^T
         info.bwd_grad_act_flops = (2 * qk + 2 * pv + extra * qk) * sparse
@@ -507,7 +531,7 @@ class CoreAttention(MetaModule):
                 info.fwd_extra_mem = 2 * (q + k + v)
                 info.bwd_grad_act_extra_mem = 2 * (q + k + v)
         else:
-            scores = b * self.sdp_head_num * s * s * self.element_size
+            scores = b * self.sdp_head_num * self.q_seq * s * self.element_size
             info.fwd_accessed_mem = q + k + v + o + 4 * scores
             info.bwd_grad_act_accessed_mem = 2 * (q + k + v + o) + 6 * scores
 
@@ -522,6 +546,16 @@ class CoreAttention(MetaModule):
                 self.add_comm("bwd_act", "all2all", t, self.cp, "cp")
             self.add_comm("fwd", "all2all", o, self.cp, "cp")
             self.add_comm("bwd_act", "all2all", o, self.cp, "cp")
+        elif self.strategy.cp_comm_type == "ring":
+            # ring attention (extension — absent in the reference): cp-1
+            # p2p hops of one K/V block each way; backward re-circulates
+            # the blocks and accumulates dK/dV over the reverse ring. On
+            # xGMI each hop is an independent point-to-point link, so the
+            # hops are priced individually (per-hop latency counts).
+            blk = (k + v) / self.cp
+            for _ in range(self.cp - 1):
+                self.add_comm("fwd", "p2p", blk, 2, "cp")
+                self.add_comm("bwd_act", "p2p", 2 * blk, 2, "cp")
         else:
             # kv all_gather mode
             self.add_comm("fwd", "all_gather", k + v, self.cp, "cp")

@@ -176,3 +176,97 @@ def offset_causal_sdp(q, k, v, q_offset):
     p = torch.softmax(scores, dim=-1)
     out = torch.einsum("bhqk,bkhd->bqhd", p, vx.float())
     return out.to(q.dtype)
+
+
+# ---- ring mode (cp_comm_type="ring") -------------------------------------
+# Ring attention (blockwise SDP with K/V blocks circulating over p2p and
+# online log-sum-exp accumulation) — ABSENT in the reference (SURVEY §2.2
+# lists it as not implemented). q stays seq-sharded; each of the cp-1
+# ring steps passes the K/V block to the next rank, so peak memory holds
+# ONE remote block instead of the all_gather mode's full sequence, and on
+# xGMI every hop is an independent point-to-point link.
+
+
+class _RingPass(torch.autograd.Function):
+    """Send x to the next cp rank, receive the previous rank's block.
+    Backward reverses the flow (the received block's grad belongs to the
+    sender). Deadlock-free via even/odd send/recv ordering (gloo-safe)."""
+
+    @staticmethod
+    def _exchange(x, group, to_next):
+        cp = dist.get_world_size(group)
+        me = dist.get_rank(group)
+        ranks = dist.get_process_group_ranks(group)
+        nxt = ranks[(me + 1) % cp]
+        prv = ranks[(me - 1) % cp]
+        dst, src = (nxt, prv) if to_next else (prv, nxt)
+        out = torch.empty_like(x)
+        x = x.contiguous()
+        if me % 2 == 0:
+            dist.send(x, dst, group=group)
+            dist.recv(out, src, group=group)
+        else:
+            dist.recv(out, src, group=group)
+            dist.send(x, dst, group=group)
+        return out
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        return _RingPass._exchange(x, group, to_next=True)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return _RingPass._exchange(dy.contiguous(), ctx.group,
+                                   to_next=False), None
+
+
+def ring_attention(q, k, v, group, cp_rank):
+    """Blockwise causal attention over the ring. q/k/v [B, s, H(kv), d]
+    seq-sharded (contiguous shards, global positions cp_rank*s + i).
+    fp32 accumulation with online LSE merging across blocks.
+
+    Autograd-over-p2p rules: K and V travel as ONE stacked tensor per
+    step (the chain blk(t) <- blk(t-1) gives backward a strict, rank-
+    independent exchange order), and the FINAL block is folded into the
+    output with zero weight — a rank whose last blocks are causally
+    skipped would otherwise prune their _RingPass.backward nodes while
+    its peer still runs them (observed gloo deadlock)."""
+    cp = dist.get_world_size(group)
+    B, s, H, d = q.shape
+    Hkv = k.shape[2]
+    rep = H // Hkv
+    qf = q.float()
+    qpos = torch.arange(cp_rank * s, (cp_rank + 1) * s, device=q.device)
+    num = torch.zeros(B, H, s, d, device=q.device)
+    den = torch.zeros(B, H, s, device=q.device)
+    m_run = torch.full((B, H, s), float("-inf"), device=q.device)
+    blk = torch.cat([k, v], dim=-1)
+    for t in range(cp):
+        src = (cp_rank - t) % cp
+        if t > 0:
+            blk = _RingPass.apply(blk, group)
+        if src > cp_rank:
+            continue       # block entirely in the causal future: weight 0
+        blk_k, blk_v = blk[..., :d], blk[..., d:]
+        kx = (blk_k.repeat_interleave(rep, dim=2) if rep > 1 else blk_k)
+        vx = (blk_v.repeat_interleave(rep, dim=2) if rep > 1 else blk_v)
+        scores = torch.einsum("bqhd,bkhd->bhqk", qf, kx.float()) * d ** -0.5
+        if src == cp_rank:  # diagonal block: triangular mask
+            kpos = torch.arange(src * s, (src + 1) * s, device=q.device)
+            scores = scores.masked_fill(
+                kpos[None, None, None, :] > qpos[None, None, :, None],
+                float("-inf"))
+        m_blk = scores.amax(dim=-1)
+        m_new = torch.maximum(m_run, m_blk)
+        alpha = torch.exp(m_run - m_new)
+        p = torch.exp(scores - m_new[..., None])
+        num = num * alpha[..., None] + torch.einsum(
+            "bhqk,bkhd->bhqd", p, vx.float())
+        den = den * alpha + p.sum(dim=-1)
+        m_run = m_new
+    out = (num / den[..., None]).permute(0, 2, 1, 3)
+    # zero-weight use of the last block: keeps the backward ring walk
+    # symmetric across ranks (grads through it are exactly zero)
+    out = out + 0.0 * blk.float().sum()
+    return out.to(q.dtype)

@@ -70,7 +70,13 @@ class GQAAttention(nn.Module):
         q = q.view(B, S, self.heads, d)
         k = k.view(B, S, self.kv_heads, d)
         v = v.reshape(B, S, self.kv_heads, d)
-        if self.cp_group is not None and self.cp_comm_type == "all_gather":
+        if self.cp_group is not None and self.cp_comm_type == "ring":
+            # ring attention: K/V blocks circulate over p2p, online-LSE
+            # accumulation — one remote block resident at a time
+            from .cp import ring_attention
+
+            ctx = ring_attention(q, k, v, self.cp_group, self.cp_rank)
+        elif self.cp_group is not None and self.cp_comm_type == "all_gather":
             # kv all_gather: q stays seq-sharded; K/V are gathered to the
             # full sequence and attention runs with an offset-causal mask
             # (math SDP — the a2a mode is the flash-kernel path)

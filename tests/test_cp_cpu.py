@@ -108,3 +108,10 @@ def test_cp2_all_gather_matches_single_process():
     """kv all_gather mode: q stays seq-sharded, K/V gathered, offset-
     causal mask — the mode the reference prices but cannot execute."""
     _run_cp(29634, "all_gather")
+
+
+@pytest.mark.timeout(300)
+def test_cp2_ring_matches_single_process():
+    """ring attention: K/V blocks circulate over p2p with online-LSE
+    block accumulation — absent in the reference entirely."""
+    _run_cp(29637, "ring")

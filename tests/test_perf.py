@@ -479,3 +479,44 @@ def test_dp_overlap_exposure_model():
     c8_no = run(8, overlap=False)
     assert c8_no["dp_time"] == pytest.approx(c8_no["dp_time_raw"])
     assert c8_no["iter_time"] > c8["iter_time"]
+
+
+def test_cp_ring_mode():
+    """Ring CP (extension — the reference has no ring attention at all):
+    per-hop p2p events, lower activation cache than kv-all_gather (only
+    the local K/V block is cached), and worst-rank causal flops."""
+    import copy
+
+    from simumax_amd import (ModelConfig, PerfLLM, StrategyConfig,
+                             SystemConfig, get_simu_model_config,
+                             get_simu_system_config)
+
+    mc = ModelConfig.init_from_config_file(get_simu_model_config("llama3-8b"))
+
+    def run(mode):
+        st = StrategyConfig(
+            seq_len=16384, micro_batch_size=1, micro_batch_num=1,
+            world_size=4, tp_size=1, pp_size=1, cp_size=4,
+            cp_comm_type=mode, enable_sequence_parallel=False, zero_state=0,
+            use_fp32_accum_grad=True, cross_entropy_loss_fusion=True,
+            attention_sparse_ratio=0.5, mem_factor=1.0)
+        p = PerfLLM()
+        p.configure(st, copy.deepcopy(mc), SystemConfig.init_from_config_file(
+            get_simu_system_config("mi355x")))
+        p.run_estimate()
+        return p
+
+    ring, ag, a2a = run("ring"), run("all_gather"), run("a2a")
+    # ring comm = per-hop p2p events
+    sdp = next(lf for lf in ring.chunks[0].leaf_modules()
+               if type(lf).__name__ == "CoreAttention")
+    kinds = [(e.stage, e.op_name) for e in sdp.comm_ops]
+    assert ("fwd", "p2p") in kinds and ("bwd_act", "p2p") in kinds
+    # one event per ring hop: cp-1 = 3 fwd hops
+    assert kinds.count(("fwd", "p2p")) == 3
+    # ring caches only the local kv block -> lower peak than all_gather
+    assert (ring.analysis_mem()["max_peak_mem"]
+            < ag.analysis_mem()["max_peak_mem"])
+    # all modes end up within a sane band of each other on time
+    ts = [m.analysis_cost()["iter_time"] for m in (ring, ag, a2a)]
+    assert max(ts) / min(ts) < 1.8, ts
