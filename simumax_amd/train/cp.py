@@ -235,10 +235,11 @@ def ring_attention(q, k, v, group, cp_rank):
     cp = dist.get_world_size(group)
     B, s, H, d = q.shape
     Hkv = k.shape[2]
+    dv = v.shape[-1]          # may differ from d (MLA: 192 qk / 128 v)
     rep = H // Hkv
     qf = q.float()
     qpos = torch.arange(cp_rank * s, (cp_rank + 1) * s, device=q.device)
-    num = torch.zeros(B, H, s, d, device=q.device)
+    num = torch.zeros(B, H, s, dv, device=q.device)
     den = torch.zeros(B, H, s, device=q.device)
     m_run = torch.full((B, H, s), float("-inf"), device=q.device)
     blk = torch.cat([k, v], dim=-1)

@@ -109,10 +109,14 @@ class MLAAttention(nn.Module):
     (Dqk = qk_head_dim + qk_pos_emb_head_dim, Dv = v_head_dim). Mirrors
     the simulator's MLAAttention op graph (ops/dense.py MLAAttention)."""
 
-    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None):
+    def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
+                 cp_group=None, cp_rank=0, cp_comm_type="a2a"):
         super().__init__()
         h = cfg.hidden_size
         self.heads = cfg.head_num
+        self.cp_group = cp_group
+        self.cp_rank = cp_rank
+        self.cp_comm_type = cp_comm_type
         self.dn = cfg.qk_head_dim            # nope part (128)
         self.dp = cfg.qk_pos_emb_head_dim    # rope part (64)
         self.dv = cfg.v_head_dim
@@ -157,7 +161,20 @@ class MLAAttention(nn.Module):
         qf = torch.cat([q[..., :dn], q_pe], dim=-1).view(B, S, H, dn + dp)
         kf = torch.cat([kvu[..., :dn], k_pe.expand(B * S, H, dp)], dim=-1)             .view(B, S, H, dn + dp)
         v = kvu[..., dn:].reshape(B, S, H, dv).contiguous()
-        ctx = K.flash_attention(qf, kf, v, causal=True)
+        if self.cp_group is not None and self.cp_comm_type == "ring":
+            from .cp import ring_attention
+
+            ctx = ring_attention(qf, kf, v, self.cp_group, self.cp_rank)
+        elif self.cp_group is not None:
+            # kv all_gather (a2a head-scatter is incompatible with MLA's
+            # per-token k_pe shared across heads)
+            from .cp import cp_allgather_kv, offset_causal_sdp
+
+            kf_full = cp_allgather_kv(kf, self.cp_group)
+            v_full = cp_allgather_kv(v, self.cp_group)
+            ctx = offset_causal_sdp(qf, kf_full, v_full, self.cp_rank * S)
+        else:
+            ctx = K.flash_attention(qf, kf, v, causal=True)
         return self.out_proj(ctx.reshape(B, S, H * dv))
 
 
@@ -173,8 +190,12 @@ class LlamaDecoderLayer(nn.Module):
         self.input_norm = K.RMSNorm(h, dtype=dtype, device=device)
         if getattr(cfg, "attention_type", "gqa") == "mla":
             assert tp_size == 1, "MLA requires tp_size == 1 (simulator parity)"
-            assert cp_size == 1, "trainer CP supports GQA attention"
-            self.attention = MLAAttention(cfg, dtype=dtype, device=device)
+            assert cp_size == 1 or cp_comm_type in ("all_gather", "ring"), \
+                "MLA CP needs all_gather or ring (a2a head-scatter clashes " \
+                "with the shared k_pe)"
+            self.attention = MLAAttention(cfg, dtype=dtype, device=device,
+                                          cp_group=cp_group, cp_rank=cp_rank,
+                                          cp_comm_type=cp_comm_type)
         else:
             self.attention = GQAAttention(cfg, dtype=dtype, device=device,
                                           tp_group=tp_group, tp_size=tp_size,
