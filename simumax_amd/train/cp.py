@@ -23,23 +23,29 @@ import torch.distributed as dist
 _CP_GROUPS = {}
 
 
-def get_cp_groups(cp_size):
-    """cp group = cp_size consecutive ranks. Returns (cp_group, cp_rank)."""
+def get_cp_groups(cp_size, tp_size=1):
+    """cp group = cp_size ranks strided by tp within each tp*cp block
+    (Megatron tp-cp-dp rank order, core/utils.get_rank_group). With
+    tp_size=1 this is cp_size consecutive ranks.
+    Returns (cp_group, cp_rank)."""
     if cp_size <= 1 or not dist.is_initialized():
         return None, 0
-    key = (cp_size, dist.get_world_size())
+    key = (cp_size, tp_size, dist.get_world_size())
     if key not in _CP_GROUPS:
         world = dist.get_world_size()
-        assert world % cp_size == 0
+        block = cp_size * tp_size
+        assert world % block == 0
         groups = {}
-        for start in range(0, world, cp_size):
-            g = dist.new_group(list(range(start, start + cp_size)))
-            for r in range(start, start + cp_size):
-                groups[r] = g
+        for start in range(0, world, block):
+            for off in range(tp_size):
+                ranks = [start + off + i * tp_size for i in range(cp_size)]
+                g = dist.new_group(ranks)
+                for r in ranks:
+                    groups[r] = g
         _CP_GROUPS[key] = groups
     groups = _CP_GROUPS[key]
     r = dist.get_rank()
-    return groups[r], r % cp_size
+    return groups[r], (r // tp_size) % cp_size
 
 
 def _a2a_exchange(chunks, group):

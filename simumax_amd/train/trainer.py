@@ -382,8 +382,8 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
         # EP a2a carries no duplicates (Megatron etp=1 semantics)
         assert cfg.sequence_parallel, "tp x ep requires sequence_parallel"
     if cfg.cp_size > 1:
-        assert cfg.tp_size == 1 and cfg.ep_size == 1 and cfg.pp_size == 1, \
-            "trainer CP composes with pure DP for now"
+        assert cfg.ep_size == 1 and cfg.pp_size == 1, \
+            "trainer CP composes with DP and TP (not EP/PP yet)"
     if cfg.zero_state == 1:
         assert cfg.ep_size == 1 and cfg.pp_size == 1, \
             "trainer ZeRO-1 composes with DP and TP (not EP/PP yet)"
@@ -399,9 +399,12 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
     from .tp import get_tp_groups
 
     tp_group, dp_group, tp_rank = get_tp_groups(tp_size)
-    cp_group, cp_rank = get_cp_groups(cfg.cp_size)
+    cp_group, cp_rank = get_cp_groups(cfg.cp_size, tp_size)
     dp_size = None
     if tp_group is not None:
+        # the "dp" reduce group from get_tp_groups spans cp too (dp_cp):
+        # cp ranks hold different seq shards of the SAME batch, so their
+        # grads average exactly like data parallelism
         dp_size = dist.get_world_size() // tp_size
     # cp ranks average grads with the dp group (dp_cp): the default
     # world-spanning reducer already does that when tp = ep = 1
