@@ -22,7 +22,7 @@ def _tiny_cfg():
                        vocab_size=512, use_swiglu=True)
 
 
-def _worker(rank, world, port, q, cp_comm_type="a2a"):
+def _worker(rank, world, port, q, cp_comm_type="a2a", cp_size=2):
     import torch.distributed as dist
 
     sys.path.insert(0, REPO)
@@ -38,10 +38,13 @@ def _worker(rank, world, port, q, cp_comm_type="a2a"):
 
         cfg = _tiny_cfg()
         S = 64
+        dp = world // cp_size
         tc = TrainConfig(seq_len=S, micro_batch_size=2, micro_batch_num=1,
-                         overlap_grad_reduce=False, cp_size=2,
+                         overlap_grad_reduce=False, cp_size=cp_size,
                          cp_comm_type=cp_comm_type)
         model, opt, red = build_trainer(cfg, tc, "cpu")
+        c = rank % cp_size          # cp consecutive (tp=1)
+        d = rank // cp_size         # dp replica
 
         # single-process full-sequence reference (identical init seed)
         torch.manual_seed(1234)
@@ -52,17 +55,21 @@ def _worker(rank, world, port, q, cp_comm_type="a2a"):
             for name, p in model.named_parameters():
                 p.copy_(rd[name])
 
-        toks, labels = make_synthetic_batch(cfg.vocab_size, 1, 2, S,
+        toks, labels = make_synthetic_batch(cfg.vocab_size, dp, 2, S,
                                             "cpu", seed=77)
-        s_loc = S // world
-        sl = slice(rank * s_loc, (rank + 1) * s_loc)
-        loss = model(toks[0][:, sl], labels[0][:, sl])
+        s_loc = S // cp_size
+        sl = slice(c * s_loc, (c + 1) * s_loc)
+        loss = model(toks[d][:, sl], labels[d][:, sl])
         loss.backward()
         accumulate_main_grads([p for p in model.parameters()])
         red.finalize()
 
-        ref_loss = ref(toks[0], labels[0])
-        ref_loss.backward()
+        # reference: mean loss/summed grads over the dp batches
+        ref_loss = 0.0
+        for mb in range(dp):
+            lo = ref(toks[mb], labels[mb])
+            (lo / dp).backward()
+            ref_loss = ref_loss + lo.detach() / dp
         accumulate_main_grads([p for p in ref.parameters()])
 
         # loss: mean over cp shards == full-seq mean
@@ -83,14 +90,15 @@ def _worker(rank, world, port, q, cp_comm_type="a2a"):
         dist.destroy_process_group()
 
 
-def _run_cp(port, cp_comm_type):
+def _run_cp(port, cp_comm_type, world=2, cp_size=2):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    ps = [ctx.Process(target=_worker, args=(r, 2, port, q, cp_comm_type))
-          for r in range(2)]
+    ps = [ctx.Process(target=_worker,
+                      args=(r, world, port, q, cp_comm_type, cp_size))
+          for r in range(world)]
     for p in ps:
         p.start()
-    results = [q.get(timeout=280) for _ in range(2)]
+    results = [q.get(timeout=280) for _ in range(world)]
     for p in ps:
         p.join(timeout=60)
     for rank, lerr, gerr in results:
@@ -115,3 +123,10 @@ def test_cp2_ring_matches_single_process():
     """ring attention: K/V blocks circulate over p2p with online-LSE
     block accumulation — absent in the reference entirely."""
     _run_cp(29637, "ring")
+
+
+@pytest.mark.timeout(420)
+def test_cp2_dp2_composition():
+    """world 4 = cp2 x dp2: seq shards within the cp pair, distinct
+    batches across dp, one world-spanning dp_cp grad average."""
+    _run_cp(29640, "a2a", world=4, cp_size=2)
