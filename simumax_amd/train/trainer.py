@@ -243,7 +243,7 @@ class DataParallelGradReducer:
 
     def __init__(self, opt, overlap: bool, bucket_bytes: int,
                  edp_group=None, edp_size=1, dp_group=None, dp_size=None,
-                 tp_group=None):
+                 tp_group=None, cp_size=1):
         self.params = opt.params
         self.flat_grad = opt.flat_grad
         self.dense_numel = opt.dense_numel
@@ -251,6 +251,11 @@ class DataParallelGradReducer:
         # expert grads are replicated only across edp and reduce there
         self.edp_group = edp_group
         self.edp_size = edp_size
+        # cp x ep: each rank's loss is a mean over its SEQ SHARD, so raw
+        # expert grads carry a cp x token weight; the dense path absorbs
+        # it in the world-spanning average, the expert path must divide
+        # by edp*cp to keep both on the dp-mean convention
+        self.exp_div = edp_size * cp_size
         self.dp_group = dp_group
         self.tp_group = tp_group
         # sequence-parallel norms produce PARTIAL weight grads (each tp
@@ -325,8 +330,9 @@ class DataParallelGradReducer:
                 if self.overlap and self.reduce_this_pass:
                     sl = self.flat_grad[span[0]:span[1]]
                     if is_expert:
+                        if self.exp_div > 1:
+                            sl.div_(self.exp_div)
                         if self.edp_size > 1:
-                            sl.div_(self.edp_size)
                             self.handles.append(dist.all_reduce(
                                 sl, group=self.edp_group, async_op=True))
                     else:
@@ -359,10 +365,12 @@ class DataParallelGradReducer:
             dense = self.flat_grad[:self.dense_numel]
             dense.div_(self.dp_size)
             dist.all_reduce(dense, group=self.dp_group)
-            if self.dense_numel < self.flat_grad.numel() and self.edp_size > 1:
+            if self.dense_numel < self.flat_grad.numel():
                 exp = self.flat_grad[self.dense_numel:]
-                exp.div_(self.edp_size)
-                dist.all_reduce(exp, group=self.edp_group)
+                if self.exp_div > 1:
+                    exp.div_(self.exp_div)
+                if self.edp_size > 1:
+                    dist.all_reduce(exp, group=self.edp_group)
 
 
 def accumulate_main_grads(params):
@@ -382,8 +390,13 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
         # EP a2a carries no duplicates (Megatron etp=1 semantics)
         assert cfg.sequence_parallel, "tp x ep requires sequence_parallel"
     if cfg.cp_size > 1:
-        assert cfg.ep_size == 1 and cfg.pp_size == 1, \
-            "trainer CP composes with DP and TP (not EP/PP yet)"
+        assert cfg.pp_size == 1, "trainer CP does not compose with PP yet"
+        # cp x ep: MoE treats cp ranks like dp members with distinct
+        # (seq-shard) tokens — the consecutive ep groups span the cp pair
+        # and edp averaging covers the dp replicas (Megatron dp_cp-hosted
+        # expert parallelism)
+        assert cfg.ep_size == 1 or cfg.tp_size == 1, \
+            "cp x ep x tp not supported"
     if cfg.zero_state == 1:
         assert cfg.ep_size == 1 and cfg.pp_size == 1, \
             "trainer ZeRO-1 composes with DP and TP (not EP/PP yet)"
@@ -434,7 +447,8 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                                       cfg.bucket_bytes,
                                       edp_group=edp_group, edp_size=edp_size,
                                       dp_group=dp_group, dp_size=dp_size,
-                                      tp_group=tp_group)
+                                      tp_group=tp_group,
+                                      cp_size=cfg.cp_size)
     return model, opt, reducer
 
 
