@@ -38,20 +38,24 @@ class PipelineStageModel(nn.Module):
 
     def __init__(self, cfg: ModelConfig, seq_len: int, stage: int, pp: int,
                  dtype=torch.bfloat16, rope_base=500000.0, device=None,
-                 tp_group=None, tp_size=1, tp_rank=0):
+                 tp_group=None, tp_size=1, tp_rank=0, cp_group=None,
+                 cp_rank=0, cp_size=1, cp_comm_type="a2a"):
         super().__init__()
         self.cfg = cfg
         self.stage = stage
         self.pp = pp
         self.tp_group = tp_group
         self.tp_size = tp_size
+        self.cp_rank = cp_rank
         lo, hi = stage_layer_range(cfg.layer_num, pp, stage)
         if stage == 0:
             self.embedding = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
                                           dtype=dtype, device=device)
         self.layers = nn.ModuleList(
             [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i,
-                               tp_group=tp_group, tp_size=tp_size)
+                               tp_group=tp_group, tp_size=tp_size,
+                               cp_group=cp_group, cp_size=cp_size,
+                               cp_rank=cp_rank, cp_comm_type=cp_comm_type)
              for i in range(lo, hi)])
         if stage == pp - 1:
             assert cfg.vocab_size % tp_size == 0
@@ -71,13 +75,12 @@ class PipelineStageModel(nn.Module):
         # x: tokens [B, S] on stage 0, hidden [B, S, H] elsewhere
         if self.stage == 0:
             B, S = x.shape
-            pos = (torch.arange(S, device=x.device, dtype=torch.int32)
-                   .repeat(B))
             x = self.embedding(x)
         else:
             B, S, _ = x.shape
-            pos = (torch.arange(S, device=x.device, dtype=torch.int32)
-                   .repeat(B))
+        # global positions of the local (CP) seq shard
+        pos = (torch.arange(self.cp_rank * S, (self.cp_rank + 1) * S,
+                            device=x.device, dtype=torch.int32).repeat(B))
         for layer in self.layers:
             x = layer(x, self.rope_cs, pos)
         if self.stage == self.pp - 1:
@@ -184,9 +187,22 @@ def build_pp_trainer(model_cfg: ModelConfig, cfg, device="cpu"):
     torch.manual_seed(1234)
     model_cfg.maybe_pad_vocab_size(cfg.tp_size)
     ps = init_parallel_state(tp_size=cfg.tp_size, pp_size=cfg.pp_size)
+    cp_group, cp_rank = None, 0
+    if getattr(cfg, "cp_size", 1) > 1:
+        # pp x cp: cp rides inside each stage's dp block (consecutive
+        # ranks, tp must be 1); the stage-peer pairing at rank +- dp
+        # already preserves the cp coordinate across stages
+        assert cfg.tp_size == 1, "pp x cp x tp not supported"
+        from .cp import get_cp_groups
+
+        cp_group, cp_rank = get_cp_groups(cfg.cp_size)
     model = PipelineStageModel(model_cfg, cfg.seq_len, ps.stage, cfg.pp_size,
                                device=device, tp_group=ps.tp_group,
-                               tp_size=cfg.tp_size, tp_rank=ps.tp_rank)
+                               tp_size=cfg.tp_size, tp_rank=ps.tp_rank,
+                               cp_group=cp_group, cp_rank=cp_rank,
+                               cp_size=getattr(cfg, "cp_size", 1),
+                               cp_comm_type=getattr(cfg, "cp_comm_type",
+                                                    "a2a"))
     opt = MixedPrecisionAdam(model.parameters(), cfg)
     if ps.tp_group is not None or ps.pp_norm_group is not None:
         for p in opt.params:
