@@ -134,7 +134,8 @@ def main():
         def step_fn(m, o, r, t, l, mbc):
             return pp_train_step(
                 m, o, r, t, l, mbc, ps.pp_prev, ps.pp_next,
-                (args.mbs, args.seq_len, model_cfg.hidden_size),
+                (args.mbs, args.seq_len // args.cp,
+                 model_cfg.hidden_size),
                 __import__("torch").bfloat16)
     else:
         model, opt, reducer = build_trainer(model_cfg, tc, device)
@@ -152,9 +153,10 @@ def main():
                                         args.mbs, args.seq_len, device,
                                         seed=1000 + dp_rank)
     if args.cp > 1:
-        # cp ranks train on their seq slice of the SAME batch
+        # cp ranks train on their seq slice of the SAME batch (cp strides
+        # by tp in the rank order; consecutive when tp=1)
         s_loc = args.seq_len // args.cp
-        cp_rank = rank % args.cp
+        cp_rank = (rank // args.tp) % args.cp
         sl = slice(cp_rank * s_loc, (cp_rank + 1) * s_loc)
         toks, labels = toks[:, :, sl].contiguous(), labels[:, :, sl].contiguous()
 
