@@ -145,9 +145,11 @@ def main():
         print(f"[bench] built {args.model} ({n_params/1e9:.2f}B params"
               f"{' on this rank' if args.tp * args.pp > 1 else ''}) "
               f"in {time.time()-t0:.1f}s", file=sys.stderr)
-    # each DATA-parallel column gets distinct data (tp/pp/cp peers share it)
+    # each DATA-parallel column gets distinct data (tp/pp/cp peers share
+    # it); with pp, ps.dp_rank is the within-stage index whose low bits
+    # are the cp coordinate
     mp_deg = args.tp * args.cp
-    dp_rank = ps.dp_rank if ps is not None else (
+    dp_rank = (ps.dp_rank // args.cp) if ps is not None else (
         rank // mp_deg if mp_deg > 1 else rank)
     toks, labels = make_synthetic_batch(model_cfg.vocab_size, args.mbc,
                                         args.mbs, args.seq_len, device,
