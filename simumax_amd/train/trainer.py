@@ -86,48 +86,83 @@ class MixedPrecisionAdam:
     step is ~8 full-size kernels instead of ~3 small kernels per param —
     matching the calibrated optimizer bandwidth."""
 
-    def __init__(self, params, cfg: TrainConfig, zero_group=None):
+    def __init__(self, params, cfg: TrainConfig, zero_group=None,
+                 expert_zero_group=None):
         params = [p for p in params if p.requires_grad]
         # dense params first, expert params last: the flat grad buffer then
-        # splits into one contiguous slice per reduction group
+        # splits into one contiguous SEGMENT per reduction group; under
+        # ZeRO-1 each segment shards over ITS group (dense over dp[_cp],
+        # experts over edp — Megatron distributed-optimizer semantics)
         self.params = ([p for p in params if not getattr(p, "_is_expert", False)]
                        + [p for p in params if getattr(p, "_is_expert", False)])
-        self.dense_numel = sum(p.numel() for p in self.params
-                               if not getattr(p, "_is_expert", False))
+        dense_total = sum(p.numel() for p in self.params
+                          if not getattr(p, "_is_expert", False))
+        expert_total = sum(p.numel() for p in self.params
+                           if getattr(p, "_is_expert", False))
         self.cfg = cfg
-        total = sum(p.numel() for p in self.params)
         dev = self.params[0].device
-        # ZeRO-1 (Megatron distributed optimizer): fp32 master/m/v exist
-        # only for this rank's flat shard; params are all_gathered after
-        # the step. Buffers are padded so the shards are equal-sized.
         self.zero = (cfg.zero_state == 1 and dist.is_initialized()
-                     and dist.get_world_size(zero_group) > 1)
+                     and (dist.get_world_size(zero_group) > 1
+                          or (expert_total and expert_zero_group is not None
+                              and dist.get_world_size(expert_zero_group)
+                              > 1)))
         self.zero_group = zero_group if self.zero else None
-        if self.zero:
-            zr = dist.get_rank(self.zero_group)
-            zw = dist.get_world_size(self.zero_group)
+        self.expert_zero_group = expert_zero_group if self.zero else None
+
+        def seg(total, group, world_default):
+            """(padded_len, this rank's local slice) for one segment.
+            group=None means the WORLD for the dense segment (Megatron
+            default) but UNSHARDED for experts (edp may be trivial)."""
+            if (not self.zero or total == 0
+                    or (group is None and not world_default)):
+                return total, slice(0, total)
+            zw = dist.get_world_size(group)
+            zr = dist.get_rank(group)
             pad = (total + zw - 1) // zw * zw
-            self.shard = slice(zr * pad // zw, (zr + 1) * pad // zw)
-            self.zero_world = zw
-        else:
-            pad = total
-            self.shard = slice(0, total)
+            return pad, slice(zr * pad // zw, (zr + 1) * pad // zw)
+
+        d_pad, d_sh = seg(dense_total, zero_group, True)
+        e_pad, e_sh = seg(expert_total, expert_zero_group, False)
+        self.dense_numel = dense_total
+        self.expert_offset = d_pad          # experts start after the pad
+        self.expert_numel = expert_total
+        pad = d_pad + e_pad
         self.flat_param = torch.empty(pad, dtype=self.params[0].dtype,
                                       device=dev)
         self.flat_grad = torch.zeros(pad, dtype=torch.float32, device=dev)
+        self.offsets = {}
         off = 0
         for p in self.params:
+            if getattr(p, "_is_expert", False) and off < d_pad:
+                off = d_pad                 # jump over the dense padding
             n = p.numel()
             self.flat_param[off:off + n].view_as(p).copy_(p.data)
             p.data = self.flat_param[off:off + n].view_as(p)
             p.main_grad = self.flat_grad[off:off + n].view(p.shape)
+            self.offsets[id(p)] = (off, off + n)
             off += n
         if self.zero:
-            self.flat_param[total:].zero_()
-        self.master = self.flat_param[self.shard].float()
-        self.m = torch.zeros_like(self.master)
-        self.v = torch.zeros_like(self.master)
+            self.flat_param[dense_total:d_pad].zero_()
+            self.flat_param[d_pad + expert_total:].zero_()
+        # per-segment shard state: (flat slice, master, m, v, group, zw)
+        self.segments = []
+        for base, (seg_pad, sh), grp in (
+            (0, (d_pad, d_sh), self.zero_group),
+            (d_pad, (e_pad, e_sh), self.expert_zero_group),
+        ):
+            if seg_pad == 0:
+                continue
+            fsl = slice(base + sh.start, base + sh.stop)
+            master = self.flat_param[fsl].float()
+            sharded = self.zero and (sh.stop - sh.start) < seg_pad
+            self.segments.append(dict(
+                fsl=fsl, base=base, pad=seg_pad,
+                master=master, m=torch.zeros_like(master),
+                v=torch.zeros_like(master), group=grp,
+                zw=(dist.get_world_size(grp) if sharded else 1)))
         self.t = 0
+        self.master_numel = sum(seg["master"].numel()
+                                for seg in self.segments)
         # model-parallel grad-norm clipping (Megatron semantics: the clip
         # uses the GLOBAL grad norm): sum shard-unique ||g||^2 over the
         # model-parallel group, count replicated params once
@@ -144,12 +179,9 @@ class MixedPrecisionAdam:
         self._norm_group = group
         self._pp_norm_group = pp_group
         self._replicated_slices = []
-        off = 0
         for p in self.params:
-            n = p.numel()
             if getattr(p, replicated_flag, False):
-                self._replicated_slices.append((off, off + n))
-            off += n
+                self._replicated_slices.append(self.offsets[id(p)])
 
     @staticmethod
     def _sq_sum(t):
@@ -196,15 +228,10 @@ class MixedPrecisionAdam:
         allocate 2x the fp32 state = +64 GiB on an 8B model)."""
         self.t += 1
         b1, b2 = self.cfg.adam_betas
-        g = self.flat_grad[self.shard] if self.zero else self.flat_grad
         # global grad-norm clip (Megatron clip_grad; global across the
         # model-parallel group when configured)
         norm = self._global_grad_norm()
         scale = self.cfg.grad_clip / (norm + 1e-6)
-        if scale < 1.0:
-            g.mul_(scale)
-        self.m.mul_(b1).add_(g, alpha=1 - b1)
-        self.v.mul_(b2).addcmul_(g, g, value=1 - b2)
         bc1 = 1 - b1 ** self.t
         bc2 = 1 - b2 ** self.t
         # fold bias corrections: m/(sqrt(v)/sqrt(bc2)+eps)/bc1
@@ -212,28 +239,35 @@ class MixedPrecisionAdam:
         sqrt_bc2 = bc2 ** 0.5
         eps2 = self.cfg.adam_eps * sqrt_bc2
         step_size = -self.cfg.lr * sqrt_bc2 / bc1
-        # chunked denom: bounds the fp32 sqrt temporary to CHUNK elements
-        # (a flat v.sqrt() would transiently allocate the full 4B/param)
         CHUNK = 1 << 29  # 512M elems = 2 GiB fp32
-        for off in range(0, self.v.numel(), CHUNK):
-            sl = slice(off, min(off + CHUNK, self.v.numel()))
-            denom = self.v[sl].sqrt().add_(eps2)
-            self.master[sl].addcdiv_(self.m[sl], denom, value=step_size)
-        if self.zero:
-            self.flat_param[self.shard].copy_(self.master)
-            if dist.get_backend(self.zero_group) == "nccl":
-                dist.all_gather_into_tensor(
-                    self.flat_param, self.flat_param[self.shard].contiguous(),
-                    group=self.zero_group)
+        for seg in self.segments:
+            g = self.flat_grad[seg["fsl"]]
+            if scale < 1.0:
+                g.mul_(scale)
+            m, v, master = seg["m"], seg["v"], seg["master"]
+            m.mul_(b1).add_(g, alpha=1 - b1)
+            v.mul_(b2).addcmul_(g, g, value=1 - b2)
+            # chunked denom: bounds the fp32 sqrt temporary to CHUNK
+            # elements (a flat v.sqrt() would transiently allocate the
+            # full 4B/param)
+            for off in range(0, v.numel(), CHUNK):
+                sl = slice(off, min(off + CHUNK, v.numel()))
+                denom = v[sl].sqrt().add_(eps2)
+                master[sl].addcdiv_(m[sl], denom, value=step_size)
+            seg_flat = self.flat_param[seg["base"]:seg["base"] + seg["pad"]]
+            if self.zero and seg["zw"] > 1:
+                self.flat_param[seg["fsl"]].copy_(master)
+                local = self.flat_param[seg["fsl"]].contiguous()
+                if dist.get_backend(seg["group"]) == "nccl":
+                    dist.all_gather_into_tensor(seg_flat, local,
+                                                group=seg["group"])
+                else:
+                    shards = [torch.empty_like(local)
+                              for _ in range(seg["zw"])]
+                    dist.all_gather(shards, local, group=seg["group"])
+                    seg_flat.copy_(torch.cat(shards))
             else:
-                shards = [torch.empty_like(self.flat_param[self.shard])
-                          for _ in range(self.zero_world)]
-                dist.all_gather(shards,
-                                self.flat_param[self.shard].contiguous(),
-                                group=self.zero_group)
-                self.flat_param.copy_(torch.cat(shards))
-        else:
-            self.flat_param.copy_(self.master)
+                self.flat_param[seg["fsl"]].copy_(master)
 
 
 class DataParallelGradReducer:
@@ -272,11 +306,12 @@ class DataParallelGradReducer:
         # Megatron no_sync semantics: only the LAST microbatch's backward
         # triggers the bucketed all_reduce
         self.reduce_this_pass = True
-        # param offsets in the flat buffer (optimizer order: dense first)
-        offs, off = {}, 0
-        for p in self.params:
-            offs[id(p)] = (off, off + p.numel())
-            off += p.numel()
+        # param offsets in the flat buffer (the optimizer's layout — the
+        # dense segment may be padded under ZeRO, so offsets cannot be
+        # recomputed by cumulative numel)
+        offs = opt.offsets
+        self.offsets = offs
+        self.expert_offset = opt.expert_offset
         # Hooks are registered UNCONDITIONALLY: releasing p.grad right after
         # each accumulation is a memory-correctness requirement — without
         # the hook the weight-sized placeholder wgrads pile up in p.grad
@@ -345,12 +380,10 @@ class DataParallelGradReducer:
 
     def _bucket_ids(self, span):
         ids = set()
-        off = 0
         for p in self.params:
-            n = p.numel()
-            if off >= span[0] and off + n <= span[1]:
+            lo, hi = self.offsets[id(p)]
+            if lo >= span[0] and hi <= span[1]:
                 ids.add(id(p))
-            off += n
         return ids
 
     def finalize(self):
@@ -365,8 +398,8 @@ class DataParallelGradReducer:
             dense = self.flat_grad[:self.dense_numel]
             dense.div_(self.dp_size)
             dist.all_reduce(dense, group=self.dp_group)
-            if self.dense_numel < self.flat_grad.numel():
-                exp = self.flat_grad[self.dense_numel:]
+            if self.expert_offset < self.flat_grad.numel():
+                exp = self.flat_grad[self.expert_offset:]
                 if self.exp_div > 1:
                     exp.div_(self.exp_div)
                 if self.edp_size > 1:
@@ -398,8 +431,8 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
         assert cfg.ep_size == 1 or cfg.tp_size == 1, \
             "cp x ep x tp not supported"
     if cfg.zero_state == 1:
-        assert cfg.ep_size == 1 and cfg.pp_size == 1, \
-            "trainer ZeRO-1 composes with DP and TP (not EP/PP yet)"
+        assert cfg.pp_size == 1, \
+            "trainer ZeRO-1 composes with DP/TP/EP (not PP yet)"
     tp_size = max(tp_size, cfg.tp_size)
     # Megatron-style vocab padding (keeps CE vocab a GPU-friendly multiple
     # and makes GEMM shape keys match the calibration tables)
@@ -432,8 +465,12 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
     # ZeRO-1 shards the fp32 optimizer state over the DATA-parallel group
     # (Megatron distributed optimizer); with tp > 1 that is dp_group, not
     # the world
+    # ZeRO-1: dense state shards over the dp(_cp) group; expert state
+    # over edp (each ep rank's experts are unique — only their dp
+    # replicas share state)
     opt = MixedPrecisionAdam(model.parameters(), cfg,
-                             zero_group=dp_group if tp_size > 1 else None)
+                             zero_group=dp_group if tp_size > 1 else None,
+                             expert_zero_group=edp_group)
     if tp_group is not None:
         # tp shards are unique; norms/embedding are replicated across tp
         for p in opt.params:

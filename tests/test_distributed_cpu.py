@@ -95,8 +95,8 @@ def _worker_zero(rank, world, port, q):
             model, opt, red = build_trainer(cfg, tc, "cpu")
             if zero == 1:
                 # state really is sharded
-                assert opt.master.numel() * 2 >= opt.flat_param.numel()
-                assert opt.master.numel() < opt.flat_param.numel()
+                assert opt.master_numel * 2 >= opt.flat_param.numel()
+                assert opt.master_numel < opt.flat_param.numel()
             toks, labels = make_synthetic_batch(cfg.vocab_size, 2, 2, 32,
                                                 "cpu", seed=rank * 7 + 1)
             for _ in range(2):

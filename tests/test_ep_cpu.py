@@ -111,3 +111,58 @@ def test_ep2_matches_ep1_gradients():
         assert loss == loss  # finite
         bad = {n: e for n, e in errs.items() if e > 3e-2}
         assert not bad, f"rank {rank} grad mismatches: {bad}"
+
+
+def _zero_ep_worker(rank, world, port, q):
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from simumax_amd.train.trainer import (TrainConfig, build_trainer,
+                                               make_synthetic_batch,
+                                               train_step)
+
+        cfg = _tiny_cfg()
+        results = {}
+        for zero in (0, 1):
+            tc = TrainConfig(seq_len=32, micro_batch_size=2,
+                             micro_batch_num=2, overlap_grad_reduce=False,
+                             ep_size=2, zero_state=zero, grad_clip=1e9)
+            model, opt, red = build_trainer(cfg, tc, "cpu")
+            if zero == 1:
+                # dense shards over world(4), expert over edp(2)
+                assert opt.master_numel < opt.flat_param.numel()
+            toks, labels = make_synthetic_batch(cfg.vocab_size, 2, 2, 32,
+                                                "cpu",
+                                                seed=(rank // 2) * 7 + 1)
+            for _ in range(2):
+                train_step(model, opt, red, toks, labels, 2)
+            results[zero] = opt.flat_param.detach().float().clone()
+            red.remove_hooks()
+            del model, opt, red
+        err = (results[0] - results[1]).abs().max().item()
+        q.put((rank, err))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(420)
+def test_zero1_with_ep2_matches_zero0():
+    """ZeRO-1 x EP on world 4 (ep2 x edp2): segmented optimizer state
+    (dense sharded over the world, experts over edp) must reproduce the
+    replicated-optimizer parameters after 2 steps."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29667
+    ps = [ctx.Process(target=_zero_ep_worker, args=(r, 4, port, q))
+          for r in range(4)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=400) for _ in range(4)]
+    for p in ps:
+        p.join(timeout=60)
+    for rank, err in results:
+        assert err < 2e-3, f"rank {rank} param drift {err}"
