@@ -203,7 +203,10 @@ def build_pp_trainer(model_cfg: ModelConfig, cfg, device="cpu"):
                                cp_size=getattr(cfg, "cp_size", 1),
                                cp_comm_type=getattr(cfg, "cp_comm_type",
                                                     "a2a"))
-    opt = MixedPrecisionAdam(model.parameters(), cfg)
+    # ZeRO-1 under PP: each stage's fp32 state shards over the stage's
+    # own dp replicas (never across stages — they hold different params)
+    opt = MixedPrecisionAdam(model.parameters(), cfg,
+                             zero_group=ps.dp_group)
     if ps.tp_group is not None or ps.pp_norm_group is not None:
         for p in opt.params:
             p._replicated_tp = not getattr(p, "_is_tp_shard", False)

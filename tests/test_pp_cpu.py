@@ -421,3 +421,63 @@ def test_vpp2_matches_single_process_gradients(mbc, port):
     for rank, errs in results.items():
         bad = {n: e for n, e in errs.items() if e > 1e-3}
         assert not bad, f"rank {rank} mismatches: {bad}"
+
+
+def _worker_pp_zero(rank, world, port, q):
+    import torch.distributed as dist
+
+    sys.path.insert(0, REPO)
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+                      RANK=str(rank), WORLD_SIZE=str(world))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from simumax_amd.train.pp import build_pp_trainer, pp_train_step
+        from simumax_amd.train.trainer import (TrainConfig,
+                                               make_synthetic_batch)
+
+        cfg = _tiny_cfg()
+        mbc = 2
+        results = {}
+        for zero in (0, 1):
+            tc = TrainConfig(seq_len=32, micro_batch_size=2,
+                             micro_batch_num=mbc, overlap_grad_reduce=False,
+                             pp_size=2, zero_state=zero, grad_clip=1e9)
+            model, opt, red, ps = build_pp_trainer(cfg, tc, "cpu")
+            if zero == 1:
+                # sharded over the stage's dp pair, never across stages
+                assert opt.master_numel * 2 >= opt.flat_param.numel()
+                assert opt.master_numel < opt.flat_param.numel()
+            toks, labels = make_synthetic_batch(cfg.vocab_size, 2 * mbc, 2,
+                                                32, "cpu", seed=13)
+            my_toks = toks[ps.dp_rank * mbc:(ps.dp_rank + 1) * mbc]
+            my_labels = labels[ps.dp_rank * mbc:(ps.dp_rank + 1) * mbc]
+            for _ in range(2):
+                pp_train_step(model, opt, red, my_toks, my_labels, mbc,
+                              ps.pp_prev, ps.pp_next,
+                              (2, 32, cfg.hidden_size), torch.bfloat16)
+            results[zero] = opt.flat_param.detach().float().clone()
+            red.remove_hooks()
+            del model, opt, red
+        err = (results[0] - results[1]).abs().max().item()
+        q.put((rank, err))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(420)
+def test_zero1_with_pp2_dp2_matches_zero0():
+    """ZeRO-1 x PP on world 4 (pp2 x dp2): per-stage optimizer state
+    shards over the stage's dp pair and must reproduce the replicated
+    parameters after 2 steps."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29671
+    ps = [ctx.Process(target=_worker_pp_zero, args=(r, 4, port, q))
+          for r in range(4)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=400) for _ in range(4)]
+    for p in ps:
+        p.join(timeout=60)
+    for rank, err in results:
+        assert err < 2e-3, f"rank {rank} param drift {err}"
