@@ -189,13 +189,13 @@ def build_pp_trainer(model_cfg: ModelConfig, cfg, device="cpu"):
     ps = init_parallel_state(tp_size=cfg.tp_size, pp_size=cfg.pp_size)
     cp_group, cp_rank = None, 0
     if getattr(cfg, "cp_size", 1) > 1:
-        # pp x cp: cp rides inside each stage's dp block (consecutive
-        # ranks, tp must be 1); the stage-peer pairing at rank +- dp
-        # already preserves the cp coordinate across stages
-        assert cfg.tp_size == 1, "pp x cp x tp not supported"
+        # pp x (tp x) cp: cp rides inside each stage's dp block, strided
+        # by tp (Megatron tp-cp-dp order); stage blocks are multiples of
+        # tp*cp so the groups never straddle a stage, and the stage-peer
+        # pairing at rank +- tp*cp*dp preserves the (tp, cp) coordinate
         from .cp import get_cp_groups
 
-        cp_group, cp_rank = get_cp_groups(cfg.cp_size)
+        cp_group, cp_rank = get_cp_groups(cfg.cp_size, cfg.tp_size)
     model = PipelineStageModel(model_cfg, cfg.seq_len, ps.stage, cfg.pp_size,
                                device=device, tp_group=ps.tp_group,
                                tp_size=cfg.tp_size, tp_rank=ps.tp_rank,
