@@ -43,8 +43,13 @@ def parse_args():
     ap.add_argument("--cp", type=int, default=1,
                     help="Ulysses context-parallel degree (seq sharded)")
     ap.add_argument("--cp-comm-type", default="a2a",
-                    choices=["a2a", "all_gather"],
-                    help="CP mode: Ulysses a2a (flash path) or kv all_gather")
+                    choices=["a2a", "all_gather", "ring"],
+                    help="CP mode: Ulysses a2a (flash path), kv "
+                         "all_gather, or ring attention")
+    ap.add_argument("--cp-sharding", default="contiguous",
+                    choices=["contiguous", "zigzag"],
+                    help="CP shard assignment (zigzag balances the "
+                         "causal load; all_gather/ring only)")
     ap.add_argument("--fp8", action="store_true",
                     help="fp8 (e4m3/e5m2) decoder linears via _scaled_mm")
     ap.add_argument("--no-self-calibrate", action="store_true",
@@ -64,7 +69,8 @@ def predict(model_cfg, world, args, overlay=None):
         micro_batch_num=args.mbc,
         world_size=world,
         tp_size=args.tp, pp_size=args.pp, ep_size=1, cp_size=args.cp,
-        cp_comm_type=args.cp_comm_type, fp8=args.fp8,
+        cp_comm_type=args.cp_comm_type, cp_sharding=args.cp_sharding,
+        fp8=args.fp8,
         enable_sequence_parallel=False,
         zero_state=0,                # trainer replicates optimizer state
         use_fp32_accum_grad=True,
@@ -122,7 +128,8 @@ def main():
     tc = TrainConfig(seq_len=args.seq_len, micro_batch_size=args.mbs,
                      micro_batch_num=args.mbc, tp_size=args.tp,
                      pp_size=args.pp, cp_size=args.cp,
-                     cp_comm_type=args.cp_comm_type, fp8=args.fp8)
+                     cp_comm_type=args.cp_comm_type,
+                     cp_sharding=args.cp_sharding, fp8=args.fp8)
     device = f"cuda:{local_rank}"
     t0 = time.time()
     ps = None
@@ -157,10 +164,12 @@ def main():
     if args.cp > 1:
         # cp ranks train on their seq slice of the SAME batch (cp strides
         # by tp in the rank order; consecutive when tp=1)
-        s_loc = args.seq_len // args.cp
+        from simumax_amd.train.cp import cp_slice_batch
+
         cp_rank = (rank // args.tp) % args.cp
-        sl = slice(cp_rank * s_loc, (cp_rank + 1) * s_loc)
-        toks, labels = toks[:, :, sl].contiguous(), labels[:, :, sl].contiguous()
+        zig = args.cp_sharding == "zigzag"
+        toks = cp_slice_batch(toks, args.seq_len, args.cp, cp_rank, zig)
+        labels = cp_slice_batch(labels, args.seq_len, args.cp, cp_rank, zig)
 
     for _ in range(args.warmup):
         step_fn(model, opt, reducer, toks, labels, args.mbc)

@@ -129,6 +129,7 @@ class StrategyConfig(Config):
     etp_size: int = 1
     cp_comm_type: str = "a2a"
     cp_a2a_mode: str = "async_cp"
+    cp_sharding: str = "contiguous"   # | "zigzag" (balanced causal load)
     order_of_paralielism: str = "tp-cp-ep-dp-pp"  # (sic) reference spelling
     moe_dispatcher_policy: str = "all2all"
     num_layers_in_first_pipeline_stage: Optional[int] = None
@@ -373,6 +374,10 @@ class StrategyConfig(Config):
             assert m in VALID_MEGATRON_RECOMPUTE_MODULES, m
         assert self.cp_comm_type in ("a2a", "all_gather", "ring")
         assert self.cp_a2a_mode in VALID_CP_A2A_MODES
+        assert self.cp_sharding in ("contiguous", "zigzag")
+        if self.cp_sharding == "zigzag":
+            assert self.cp_comm_type != "a2a", \
+                "a2a reassembles shards in rank order (contiguous only)"
         if self.enable_sequence_parallel:
             assert self.seq_len % (self.tp_size * self.cp_size) == 0
         if self.interleaving_size > 1:

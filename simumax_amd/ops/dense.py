@@ -508,10 +508,14 @@ class CoreAttention(MetaModule):
         s = self.full_seq
         r = self.strategy.attention_sparse_ratio
         if self.cp_mode in ("all_gather", "ring"):
-            # seq-sharded q against the full K/V: the WORST rank (last
-            # contiguous shard) sees a causal share of 1 - r/cp of the
-            # keys (reduces to 1-r at cp=1); the step is bounded by it
-            sparse = 1.0 - r / self.cp
+            if self.strategy.cp_sharding == "zigzag":
+                # zigzag chunk pairing balances the causal load exactly:
+                # every rank computes full_causal/cp
+                sparse = 1.0 - r
+            else:
+                # contiguous shards: the WORST (last) rank sees a causal
+                # share of 1 - r/cp of the keys; the step is bounded by it
+                sparse = 1.0 - r / self.cp
         else:
             sparse = 1.0 - r
         qk = 2 * b * self.sdp_head_num * self.q_seq * s * self.qk_head_dim

@@ -163,27 +163,6 @@ def cp_allgather_kv(x, group):
     return _AllGatherSeq.apply(x, group)
 
 
-def offset_causal_sdp(q, k, v, q_offset):
-    """Math SDP with a GLOBAL causal mask for a seq-sharded q block:
-    query i (global position q_offset+i) attends keys 0..q_offset+i.
-    fp32 softmax; GQA via head repetition. q [B,s,H,d], k/v [B,S,Hkv,d]."""
-    B, s, H, d = q.shape
-    S, Hkv = k.shape[1], k.shape[2]
-    rep = H // Hkv
-    kx = k.repeat_interleave(rep, dim=2) if rep > 1 else k
-    vx = v.repeat_interleave(rep, dim=2) if rep > 1 else v
-    scores = torch.einsum("bqhd,bkhd->bhqk", q.float(), kx.float())
-    scores *= d ** -0.5
-    qpos = torch.arange(q_offset, q_offset + s, device=q.device)
-    kpos = torch.arange(S, device=q.device)
-    scores = scores.masked_fill(kpos[None, None, None, :]
-                                > qpos[None, None, :, None],
-                                float("-inf"))
-    p = torch.softmax(scores, dim=-1)
-    out = torch.einsum("bhqk,bkhd->bqhd", p, vx.float())
-    return out.to(q.dtype)
-
-
 # ---- ring mode (cp_comm_type="ring") -------------------------------------
 # Ring attention (blockwise SDP with K/V blocks circulating over p2p and
 # online log-sum-exp accumulation) — ABSENT in the reference (SURVEY §2.2
@@ -227,53 +206,99 @@ class _RingPass(torch.autograd.Function):
                                    to_next=False), None
 
 
-def ring_attention(q, k, v, group, cp_rank):
-    """Blockwise causal attention over the ring. q/k/v [B, s, H(kv), d]
-    seq-sharded (contiguous shards, global positions cp_rank*s + i).
-    fp32 accumulation with online LSE merging across blocks.
+# ---- shard assignment ----------------------------------------------------
+# "contiguous": rank c owns tokens [c*S/cp, (c+1)*S/cp) — simple, but the
+# causal mask loads the last rank ~2x the first. "zigzag": 2*cp chunks,
+# rank c owns chunks {c, 2cp-1-c} (Megatron CP load balancing) — every
+# rank sees the same causal work. a2a (Ulysses) requires contiguous
+# (its seq-gather reassembles in rank order for the causal flash kernel).
 
-    Autograd-over-p2p rules: K and V travel as ONE stacked tensor per
-    step (the chain blk(t) <- blk(t-1) gives backward a strict, rank-
-    independent exchange order), and the FINAL block is folded into the
-    output with zero weight — a rank whose last blocks are causally
-    skipped would otherwise prune their _RingPass.backward nodes while
-    its peer still runs them (observed gloo deadlock)."""
+
+def cp_shard_slices(S, cp, rank, zigzag):
+    if not zigzag:
+        s = S // cp
+        return [slice(rank * s, (rank + 1) * s)]
+    h = S // (2 * cp)
+    a, b = rank, 2 * cp - 1 - rank
+    return [slice(a * h, (a + 1) * h), slice(b * h, (b + 1) * h)]
+
+
+def cp_positions(S, cp, rank, zigzag, device):
+    """Global positions of this rank's shard, concatenated in shard
+    order (int32, length S/cp)."""
+    parts = [torch.arange(sl.start, sl.stop, device=device,
+                          dtype=torch.int32)
+             for sl in cp_shard_slices(S, cp, rank, zigzag)]
+    return torch.cat(parts) if len(parts) > 1 else parts[0]
+
+
+def cp_slice_batch(t, S, cp, rank, zigzag, dim=-1):
+    """Slice a [..., S] token/label tensor to this rank's shard."""
+    parts = [t.index_select(dim, torch.arange(sl.start, sl.stop,
+                                              device=t.device))
+             for sl in cp_shard_slices(S, cp, rank, zigzag)]
+    return parts[0].contiguous() if len(parts) == 1 else torch.cat(
+        parts, dim=dim).contiguous()
+
+
+def masked_sdp(q, k, v, qpos, kpos):
+    """Math SDP with an arbitrary-position causal mask (key kpos[j]
+    visible to query qpos[i] iff kpos[j] <= qpos[i]). fp32 softmax; GQA
+    via head repetition. Rows with no visible key return 0."""
+    B, s, H, d = q.shape
+    Hkv = k.shape[2]
+    rep = H // Hkv
+    kx = k.repeat_interleave(rep, dim=2) if rep > 1 else k
+    vx = v.repeat_interleave(rep, dim=2) if rep > 1 else v
+    scores = torch.einsum("bqhd,bkhd->bhqk", q.float(), kx.float())
+    scores *= d ** -0.5
+    scores = scores.masked_fill(kpos[None, None, None, :]
+                                > qpos[None, None, :, None],
+                                float("-inf"))
+    p = torch.softmax(scores, dim=-1)
+    out = torch.einsum("bhqk,bkhd->bqhd", p, vx.float())
+    return out.to(q.dtype)
+
+
+def ring_attention_pos(q, k, v, group, cp_rank, qpos, zigzag):
+    """Position-aware ring attention (zigzag or contiguous shards):
+    blocks circulate with deterministically reconstructed key positions;
+    online-LSE accumulation in fp32."""
     cp = dist.get_world_size(group)
     B, s, H, d = q.shape
     Hkv = k.shape[2]
-    dv = v.shape[-1]          # may differ from d (MLA: 192 qk / 128 v)
+    dv = v.shape[-1]
     rep = H // Hkv
+    S_full = s * cp
     qf = q.float()
-    qpos = torch.arange(cp_rank * s, (cp_rank + 1) * s, device=q.device)
     num = torch.zeros(B, H, s, dv, device=q.device)
     den = torch.zeros(B, H, s, device=q.device)
     m_run = torch.full((B, H, s), float("-inf"), device=q.device)
     blk = torch.cat([k, v], dim=-1)
+    qmax = int(qpos.max())
     for t in range(cp):
         src = (cp_rank - t) % cp
         if t > 0:
             blk = _RingPass.apply(blk, group)
-        if src > cp_rank:
-            continue       # block entirely in the causal future: weight 0
+        kpos = cp_positions(S_full, cp, src, zigzag, q.device)
+        if int(kpos.min()) > qmax:
+            continue       # block entirely in the causal future
         blk_k, blk_v = blk[..., :d], blk[..., d:]
         kx = (blk_k.repeat_interleave(rep, dim=2) if rep > 1 else blk_k)
         vx = (blk_v.repeat_interleave(rep, dim=2) if rep > 1 else blk_v)
         scores = torch.einsum("bqhd,bkhd->bhqk", qf, kx.float()) * d ** -0.5
-        if src == cp_rank:  # diagonal block: triangular mask
-            kpos = torch.arange(src * s, (src + 1) * s, device=q.device)
-            scores = scores.masked_fill(
-                kpos[None, None, None, :] > qpos[None, None, :, None],
-                float("-inf"))
+        scores = scores.masked_fill(kpos[None, None, None, :]
+                                    > qpos[None, None, :, None],
+                                    float("-inf"))
         m_blk = scores.amax(dim=-1)
         m_new = torch.maximum(m_run, m_blk)
-        alpha = torch.exp(m_run - m_new)
-        p = torch.exp(scores - m_new[..., None])
+        # fully-masked rows keep m=-inf until a visible block arrives
+        alpha = torch.exp((m_run - m_new).nan_to_num(0.0))
+        p = torch.exp((scores - m_new[..., None]).nan_to_num(float("-inf")))
         num = num * alpha[..., None] + torch.einsum(
             "bhqk,bkhd->bhqd", p, vx.float())
         den = den * alpha + p.sum(dim=-1)
         m_run = m_new
     out = (num / den[..., None]).permute(0, 2, 1, 3)
-    # zero-weight use of the last block: keeps the backward ring walk
-    # symmetric across ranks (grads through it are exactly zero)
     out = out + 0.0 * blk.float().sum()
     return out.to(q.dtype)

@@ -26,7 +26,8 @@ class GQAAttention(nn.Module):
 
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
                  tp_group=None, tp_size=1, sp=False, cp_group=None,
-                 cp_size=1, cp_rank=0, cp_comm_type="a2a", fp8=False):
+                 cp_size=1, cp_rank=0, cp_comm_type="a2a",
+                 cp_sharding="contiguous", fp8=False):
         super().__init__()
         h = cfg.hidden_size
         assert cfg.head_num % tp_size == 0 and cfg.kv_head_num % tp_size == 0
@@ -36,11 +37,15 @@ class GQAAttention(nn.Module):
             # Ulysses CP: heads scattered over cp inside attention
             # (all_gather mode keeps heads whole — no divisibility rule)
             assert self.heads % cp_size == 0 and self.kv_heads % cp_size == 0
+            assert cp_sharding == "contiguous", \
+                "a2a reassembles shards in rank order (contiguous only)"
         self.head_size = cfg.head_size
         self.tp_group = tp_group
         self.cp_group = cp_group
         self.cp_rank = cp_rank
         self.cp_comm_type = cp_comm_type
+        self.cp_size_ = cp_size
+        self.cp_sharding = cp_sharding
         self.sp = sp
         qkv_out = (self.heads + 2 * self.kv_heads) * cfg.head_size
         Lin = K.FusedLinear
@@ -73,18 +78,25 @@ class GQAAttention(nn.Module):
         if self.cp_group is not None and self.cp_comm_type == "ring":
             # ring attention: K/V blocks circulate over p2p, online-LSE
             # accumulation — one remote block resident at a time
-            from .cp import ring_attention
+            from .cp import ring_attention_pos
 
-            ctx = ring_attention(q, k, v, self.cp_group, self.cp_rank)
+            ctx = ring_attention_pos(q, k, v, self.cp_group, self.cp_rank,
+                                     pos[:S].long(),
+                                     self.cp_sharding == "zigzag")
         elif self.cp_group is not None and self.cp_comm_type == "all_gather":
             # kv all_gather: q stays seq-sharded; K/V are gathered to the
-            # full sequence and attention runs with an offset-causal mask
-            # (math SDP — the a2a mode is the flash-kernel path)
-            from .cp import cp_allgather_kv, offset_causal_sdp
+            # full sequence and attention runs with a positional causal
+            # mask (math SDP — the a2a mode is the flash-kernel path)
+            from .cp import cp_allgather_kv, cp_positions, masked_sdp
 
             k_full = cp_allgather_kv(k, self.cp_group)
             v_full = cp_allgather_kv(v, self.cp_group)
-            ctx = offset_causal_sdp(q, k_full, v_full, self.cp_rank * S)
+            cp = self.cp_size_
+            kpos = torch.cat([cp_positions(S * cp, cp, r,
+                                           self.cp_sharding == "zigzag",
+                                           q.device)
+                              for r in range(cp)]).long()
+            ctx = masked_sdp(q, k_full, v_full, pos[:S].long(), kpos)
         elif self.cp_group is not None:
             # a2a: scatter heads / gather sequence (Ulysses), flash on the
             # full sequence, inverse a2a on the context
@@ -110,13 +122,16 @@ class MLAAttention(nn.Module):
     the simulator's MLAAttention op graph (ops/dense.py MLAAttention)."""
 
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
-                 cp_group=None, cp_rank=0, cp_comm_type="a2a"):
+                 cp_group=None, cp_rank=0, cp_comm_type="a2a", cp_size=1,
+                 cp_sharding="contiguous"):
         super().__init__()
         h = cfg.hidden_size
         self.heads = cfg.head_num
         self.cp_group = cp_group
         self.cp_rank = cp_rank
         self.cp_comm_type = cp_comm_type
+        self.cp_size_ = cp_size
+        self.cp_sharding = cp_sharding
         self.dn = cfg.qk_head_dim            # nope part (128)
         self.dp = cfg.qk_pos_emb_head_dim    # rope part (64)
         self.dv = cfg.v_head_dim
@@ -162,17 +177,24 @@ class MLAAttention(nn.Module):
         kf = torch.cat([kvu[..., :dn], k_pe.expand(B * S, H, dp)], dim=-1)             .view(B, S, H, dn + dp)
         v = kvu[..., dn:].reshape(B, S, H, dv).contiguous()
         if self.cp_group is not None and self.cp_comm_type == "ring":
-            from .cp import ring_attention
+            from .cp import ring_attention_pos
 
-            ctx = ring_attention(qf, kf, v, self.cp_group, self.cp_rank)
+            ctx = ring_attention_pos(qf, kf, v, self.cp_group, self.cp_rank,
+                                     pos[:S].long(),
+                                     self.cp_sharding == "zigzag")
         elif self.cp_group is not None:
             # kv all_gather (a2a head-scatter is incompatible with MLA's
             # per-token k_pe shared across heads)
-            from .cp import cp_allgather_kv, offset_causal_sdp
+            from .cp import cp_allgather_kv, cp_positions, masked_sdp
 
             kf_full = cp_allgather_kv(kf, self.cp_group)
             v_full = cp_allgather_kv(v, self.cp_group)
-            ctx = offset_causal_sdp(qf, kf_full, v_full, self.cp_rank * S)
+            cp = self.cp_size_
+            kpos = torch.cat([cp_positions(S * cp, cp, r,
+                                           self.cp_sharding == "zigzag",
+                                           qf.device)
+                              for r in range(cp)]).long()
+            ctx = masked_sdp(qf, kf_full, v_full, pos[:S].long(), kpos)
         else:
             ctx = K.flash_attention(qf, kf, v, causal=True)
         return self.out_proj(ctx.reshape(B, S, H * dv))
@@ -182,7 +204,7 @@ class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: ModelConfig, dtype=torch.bfloat16, device=None,
                  layer_idx=0, ep_group=None, ep_size=1, tp_group=None,
                  tp_size=1, sp=False, cp_group=None, cp_size=1, cp_rank=0,
-                 cp_comm_type="a2a", fp8=False):
+                 cp_comm_type="a2a", cp_sharding="contiguous", fp8=False):
         super().__init__()
         h = cfg.hidden_size
         self.tp_group = tp_group
@@ -195,13 +217,16 @@ class LlamaDecoderLayer(nn.Module):
                 "with the shared k_pe)"
             self.attention = MLAAttention(cfg, dtype=dtype, device=device,
                                           cp_group=cp_group, cp_rank=cp_rank,
-                                          cp_comm_type=cp_comm_type)
+                                          cp_comm_type=cp_comm_type,
+                                          cp_size=cp_size,
+                                          cp_sharding=cp_sharding)
         else:
             self.attention = GQAAttention(cfg, dtype=dtype, device=device,
                                           tp_group=tp_group, tp_size=tp_size,
                                           sp=sp, cp_group=cp_group,
                                           cp_size=cp_size, cp_rank=cp_rank,
-                                          cp_comm_type=cp_comm_type, fp8=fp8)
+                                          cp_comm_type=cp_comm_type,
+                                          cp_sharding=cp_sharding, fp8=fp8)
         if sp:
             # SP norms see only the local seq shard: their weight grads
             # are partial sums and need a tp all_reduce (reducer handles
@@ -267,12 +292,13 @@ class LlamaForTraining(nn.Module):
                  rope_base=500000.0, device=None, ep_group=None, ep_size=1,
                  tp_group=None, tp_size=1, tp_rank=0, sp=False,
                  cp_group=None, cp_rank=0, cp_size=1, cp_comm_type="a2a",
-                 fp8=False, recompute_layers=0):
+                 cp_sharding="contiguous", fp8=False, recompute_layers=0):
         super().__init__()
         self.cfg = cfg
         self.seq_len = seq_len          # FULL sequence (rope cache size)
         self.cp_rank = cp_rank
         self.cp_size = cp_size
+        self.cp_sharding = cp_sharding
         self.recompute_layers = recompute_layers
         self.tp_group = tp_group
         self.tp_size = tp_size
@@ -288,7 +314,8 @@ class LlamaForTraining(nn.Module):
                                tp_group=tp_group, tp_size=tp_size,
                                sp=sp and tp_size > 1, cp_group=cp_group,
                                cp_size=cp_size, cp_rank=cp_rank,
-                               cp_comm_type=cp_comm_type, fp8=fp8)
+                               cp_comm_type=cp_comm_type,
+                               cp_sharding=cp_sharding, fp8=fp8)
              for i in range(cfg.layer_num)])
         self.final_norm = K.RMSNorm(cfg.hidden_size, dtype=dtype, device=device)
         if self.sp:
@@ -307,9 +334,16 @@ class LlamaForTraining(nn.Module):
     def forward(self, tokens, labels):
         # tokens/labels: [B, S] int64 (S = the LOCAL seq slice under CP)
         B, S = tokens.shape
-        pos = (torch.arange(self.cp_rank * S, (self.cp_rank + 1) * S,
-                            device=tokens.device, dtype=torch.int32)
-               .repeat(B))
+        if self.cp_size > 1:
+            from .cp import cp_positions
+
+            pos = cp_positions(S * self.cp_size, self.cp_size,
+                               self.cp_rank,
+                               self.cp_sharding == "zigzag",
+                               tokens.device).repeat(B)
+        else:
+            pos = (torch.arange(S, device=tokens.device,
+                                dtype=torch.int32).repeat(B))
         x = self.embedding(tokens)
         if self.sp:
             from .tp import slice_seq

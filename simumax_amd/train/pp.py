@@ -39,7 +39,8 @@ class PipelineStageModel(nn.Module):
     def __init__(self, cfg: ModelConfig, seq_len: int, stage: int, pp: int,
                  dtype=torch.bfloat16, rope_base=500000.0, device=None,
                  tp_group=None, tp_size=1, tp_rank=0, cp_group=None,
-                 cp_rank=0, cp_size=1, cp_comm_type="a2a"):
+                 cp_rank=0, cp_size=1, cp_comm_type="a2a",
+                 cp_sharding="contiguous"):
         super().__init__()
         self.cfg = cfg
         self.stage = stage
@@ -47,6 +48,8 @@ class PipelineStageModel(nn.Module):
         self.tp_group = tp_group
         self.tp_size = tp_size
         self.cp_rank = cp_rank
+        self.cp_size = cp_size
+        self.cp_sharding = cp_sharding
         lo, hi = stage_layer_range(cfg.layer_num, pp, stage)
         if stage == 0:
             self.embedding = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
@@ -55,7 +58,8 @@ class PipelineStageModel(nn.Module):
             [LlamaDecoderLayer(cfg, dtype, device, layer_idx=i,
                                tp_group=tp_group, tp_size=tp_size,
                                cp_group=cp_group, cp_size=cp_size,
-                               cp_rank=cp_rank, cp_comm_type=cp_comm_type)
+                               cp_rank=cp_rank, cp_comm_type=cp_comm_type,
+                               cp_sharding=cp_sharding)
              for i in range(lo, hi)])
         if stage == pp - 1:
             assert cfg.vocab_size % tp_size == 0
@@ -79,8 +83,16 @@ class PipelineStageModel(nn.Module):
         else:
             B, S, _ = x.shape
         # global positions of the local (CP) seq shard
-        pos = (torch.arange(self.cp_rank * S, (self.cp_rank + 1) * S,
-                            device=x.device, dtype=torch.int32).repeat(B))
+        if self.cp_size > 1:
+            from .cp import cp_positions
+
+            pos = cp_positions(S * self.cp_size, self.cp_size,
+                               self.cp_rank,
+                               self.cp_sharding == "zigzag",
+                               x.device).repeat(B)
+        else:
+            pos = (torch.arange(S, device=x.device,
+                                dtype=torch.int32).repeat(B))
         for layer in self.layers:
             x = layer(x, self.rope_cs, pos)
         if self.stage == self.pp - 1:
@@ -202,7 +214,9 @@ def build_pp_trainer(model_cfg: ModelConfig, cfg, device="cpu"):
                                cp_group=cp_group, cp_rank=cp_rank,
                                cp_size=getattr(cfg, "cp_size", 1),
                                cp_comm_type=getattr(cfg, "cp_comm_type",
-                                                    "a2a"))
+                                                    "a2a"),
+                               cp_sharding=getattr(cfg, "cp_sharding",
+                                                   "contiguous"))
     # ZeRO-1 under PP: each stage's fp32 state shards over the stage's
     # own dp replicas (never across stages — they hold different params)
     opt = MixedPrecisionAdam(model.parameters(), cfg,

@@ -39,6 +39,7 @@ class TrainConfig:
     pp_size: int = 1
     cp_size: int = 1   # context parallel (seq_len = FULL sequence)
     cp_comm_type: str = "a2a"   # "a2a" (Ulysses, flash path) | "all_gather"
+    cp_sharding: str = "contiguous"   # | "zigzag" (all_gather/ring only)
     fp8: bool = False  # e4m3/e5m2 GEMMs via _scaled_mm (decoder linears)
     recompute_layers: int = 0  # full-block activation recompute for the
                                # first N layers (torch.utils.checkpoint;
@@ -460,7 +461,8 @@ def build_trainer(model_cfg: ModelConfig, cfg: TrainConfig, device="cuda",
                              tp_rank=tp_rank, sp=cfg.sequence_parallel,
                              cp_group=cp_group, cp_rank=cp_rank,
                              cp_size=cfg.cp_size,
-                             cp_comm_type=cfg.cp_comm_type, fp8=cfg.fp8,
+                             cp_comm_type=cfg.cp_comm_type,
+                             cp_sharding=cfg.cp_sharding, fp8=cfg.fp8,
                              recompute_layers=cfg.recompute_layers)
     # ZeRO-1 shards the fp32 optimizer state over the DATA-parallel group
     # (Megatron distributed optimizer); with tp > 1 that is dp_group, not

@@ -22,7 +22,8 @@ def _tiny_cfg():
                        vocab_size=512, use_swiglu=True)
 
 
-def _worker(rank, world, port, q, cp_comm_type="a2a", cp_size=2):
+def _worker(rank, world, port, q, cp_comm_type="a2a", cp_size=2,
+            cp_sharding="contiguous"):
     import torch.distributed as dist
 
     sys.path.insert(0, REPO)
@@ -41,7 +42,8 @@ def _worker(rank, world, port, q, cp_comm_type="a2a", cp_size=2):
         dp = world // cp_size
         tc = TrainConfig(seq_len=S, micro_batch_size=2, micro_batch_num=1,
                          overlap_grad_reduce=False, cp_size=cp_size,
-                         cp_comm_type=cp_comm_type)
+                         cp_comm_type=cp_comm_type,
+                         cp_sharding=cp_sharding)
         model, opt, red = build_trainer(cfg, tc, "cpu")
         c = rank % cp_size          # cp consecutive (tp=1)
         d = rank // cp_size         # dp replica
@@ -57,9 +59,12 @@ def _worker(rank, world, port, q, cp_comm_type="a2a", cp_size=2):
 
         toks, labels = make_synthetic_batch(cfg.vocab_size, dp, 2, S,
                                             "cpu", seed=77)
-        s_loc = S // cp_size
-        sl = slice(c * s_loc, (c + 1) * s_loc)
-        loss = model(toks[d][:, sl], labels[d][:, sl])
+        from simumax_amd.train.cp import cp_slice_batch
+
+        zig = cp_sharding == "zigzag"
+        toks_l = cp_slice_batch(toks[d], S, cp_size, c, zig)
+        labels_l = cp_slice_batch(labels[d], S, cp_size, c, zig)
+        loss = model(toks_l, labels_l)
         loss.backward()
         accumulate_main_grads([p for p in model.parameters()])
         red.finalize()
@@ -90,11 +95,13 @@ def _worker(rank, world, port, q, cp_comm_type="a2a", cp_size=2):
         dist.destroy_process_group()
 
 
-def _run_cp(port, cp_comm_type, world=2, cp_size=2):
+def _run_cp(port, cp_comm_type, world=2, cp_size=2,
+            cp_sharding="contiguous"):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     ps = [ctx.Process(target=_worker,
-                      args=(r, world, port, q, cp_comm_type, cp_size))
+                      args=(r, world, port, q, cp_comm_type, cp_size,
+                            cp_sharding))
           for r in range(world)]
     for p in ps:
         p.start()
@@ -130,3 +137,15 @@ def test_cp2_dp2_composition():
     """world 4 = cp2 x dp2: seq shards within the cp pair, distinct
     batches across dp, one world-spanning dp_cp grad average."""
     _run_cp(29640, "a2a", world=4, cp_size=2)
+
+
+@pytest.mark.timeout(300)
+def test_cp2_ring_zigzag():
+    """zigzag shard pairing (chunks {c, 2cp-1-c}): balanced causal load,
+    position-aware masks in the ring."""
+    _run_cp(29680, "ring", cp_sharding="zigzag")
+
+
+@pytest.mark.timeout(300)
+def test_cp2_all_gather_zigzag():
+    _run_cp(29683, "all_gather", cp_sharding="zigzag")

@@ -520,3 +520,14 @@ def test_cp_ring_mode():
     # all modes end up within a sane band of each other on time
     ts = [m.analysis_cost()["iter_time"] for m in (ring, ag, a2a)]
     assert max(ts) / min(ts) < 1.8, ts
+    # zigzag balances the causal load: per-rank attention flops drop
+    # from the contiguous worst-rank share (1 - r/cp) to (1 - r)
+    def sdp_flops(p):
+        lf = next(l for l in p.chunks[0].leaf_modules()
+                  if type(l).__name__ == "CoreAttention")
+        return lf._compute_info.fwd_flops
+
+    zig = run("ring")
+    zig.strategy.cp_sharding = "zigzag"
+    zig.run_estimate()
+    assert sdp_flops(zig) < sdp_flops(ring)
