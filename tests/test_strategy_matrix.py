@@ -25,6 +25,8 @@ GRID = [
     ("dense", 8, 1, 1, 1, 2, {}),
     ("dense", 8, 1, 1, 1, 4, {"cp_comm_type": "all_gather"}),
     ("dense", 8, 1, 1, 1, 4, {"cp_comm_type": "ring"}),
+    ("dense", 8, 1, 1, 1, 4, {"cp_comm_type": "ring",
+                              "cp_sharding": "zigzag"}),
     ("dense", 8, 2, 1, 1, 2, {"enable_sequence_parallel": True}),
     ("dense", 8, 1, 1, 1, 1, {"zero_state": 1}),
     ("dense", 8, 2, 2, 1, 2, {}),
