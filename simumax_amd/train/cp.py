@@ -1,18 +1,23 @@
-"""Ulysses-style context parallelism for the reference trainer.
+"""Context parallelism for the trainer — all three priced modes are
+executable (the reference only models them):
 
-Mirrors the simulator's CP a2a cost model (ops/dense.py CoreAttention
-cp_comm_type="a2a", reference dense_module.py:1158-1338): the sequence is
-sharded over the cp group outside attention; inside attention an
-all-to-all scatters heads and gathers sequence (q, k, v), flash attention
-runs on the full sequence with head_num/cp local heads, and a reverse
-all-to-all restores the seq-sharded layout for the out projection.
+* "a2a" (Ulysses): all-to-all scatters heads / gathers sequence around
+  flash attention (simulator cp_comm_type="a2a", reference
+  dense_module.py:1158-1338). Contiguous shards only.
+* "all_gather": K/V gathered to the full sequence, q stays seq-sharded,
+  positional causal mask (the mode the reference prices but raises
+  NotImplementedError on).
+* "ring": K/V blocks circulate over p2p with online-LSE block
+  accumulation — absent in the reference entirely.
 
-Gradient semantics: cp ranks see the same batch but different seq slices,
-so parameter grads are averaged over the dp*cp group — the trainer's
-reducer already spans the whole world when tp=ep=1, which IS dp_cp.
+Shard assignment is "contiguous" or "zigzag" (chunk pairs {c, 2cp-1-c}
+for balanced causal load; all_gather/ring only).
 
-Process-group layout matches core/utils.get_rank_group: cp consecutive
-(tp fastest, but the trainer composes cp only with pure DP for now).
+Gradient semantics: cp ranks see the same batch but different seq
+slices, so parameter grads are averaged over the dp*cp group — the
+trainer's reducer spans dp_cp. Process-group layout matches
+core/utils.get_rank_group: cp strided by tp (consecutive when tp=1);
+composes with DP, TP(+SP), EP, PP and ZeRO-1 (see tests/test_*cp*).
 """
 
 from __future__ import annotations
