@@ -252,11 +252,12 @@ def test_sim_vs_perf_agreement_moe_ep(tmp_path):
     assert res["total_time"] == pytest.approx(analytic, rel=0.10)
 
 
-def test_sim_vs_perf_agreement_cp(tmp_path):
-    """Ulysses CP a2a events (q/k/v pre + o post, mirrored in bwd) replay
-    consistently."""
+@pytest.mark.parametrize("mode", ["a2a", "all_gather", "ring"])
+def test_sim_vs_perf_agreement_cp(tmp_path, mode):
+    """CP comm events (a2a: q/k/v pre + o post; ag: gather/rs; ring:
+    per-hop p2p) replay consistently through the event simulator."""
     p = build(strategy="tp1_pp1_dp8_mbs1", model="llama2-tiny",
-              cp_size=2)
+              cp_size=2, cp_comm_type=mode)
     cost = p.analysis_cost()
     res = p.simulate(str(tmp_path))
     analytic = (cost["pipeline_time"] / cost["straggler_ratio"]
