@@ -23,7 +23,7 @@ def _tiny_cfg():
 
 
 def _worker(rank, world, port, q, cp_comm_type="a2a", cp_size=2,
-            cp_sharding="contiguous"):
+            cp_sharding="contiguous", recompute_layers=0):
     import torch.distributed as dist
 
     sys.path.insert(0, REPO)
@@ -43,7 +43,8 @@ def _worker(rank, world, port, q, cp_comm_type="a2a", cp_size=2,
         tc = TrainConfig(seq_len=S, micro_batch_size=2, micro_batch_num=1,
                          overlap_grad_reduce=False, cp_size=cp_size,
                          cp_comm_type=cp_comm_type,
-                         cp_sharding=cp_sharding)
+                         cp_sharding=cp_sharding,
+                         recompute_layers=recompute_layers)
         model, opt, red = build_trainer(cfg, tc, "cpu")
         c = rank % cp_size          # cp consecutive (tp=1)
         d = rank // cp_size         # dp replica
@@ -96,12 +97,12 @@ def _worker(rank, world, port, q, cp_comm_type="a2a", cp_size=2,
 
 
 def _run_cp(port, cp_comm_type, world=2, cp_size=2,
-            cp_sharding="contiguous"):
+            cp_sharding="contiguous", recompute_layers=0):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     ps = [ctx.Process(target=_worker,
                       args=(r, world, port, q, cp_comm_type, cp_size,
-                            cp_sharding))
+                            cp_sharding, recompute_layers))
           for r in range(world)]
     for p in ps:
         p.start()
@@ -149,3 +150,12 @@ def test_cp2_ring_zigzag():
 @pytest.mark.timeout(300)
 def test_cp2_all_gather_zigzag():
     _run_cp(29683, "all_gather", cp_sharding="zigzag")
+
+
+@pytest.mark.timeout(300)
+def test_cp2_ring_with_recompute():
+    """Activation recompute re-executes the ring's p2p exchanges inside
+    backward; layer-by-layer symmetry across ranks must keep the
+    exchange order paired (the reference model keeps full activations —
+    the recomputed grads must still match it)."""
+    _run_cp(29686, "ring", recompute_layers=2)
