@@ -14,6 +14,34 @@ def apply_insitu_overlay(system_config, summary) -> int:
     acc = system_config.accelerator
     n = 0
     for table, rows in summary.items():
+        if table == "network":
+            # measured collective wall time -> rescale the tier op's
+            # efficiency so the model's prediction for that (op, bytes,
+            # comm_num) matches the measurement exactly
+            for op_name, row in rows.items():
+                try:
+                    net = system_config.networks[row.get(
+                        "net", "high_intra_node")]
+                    cfg = net.op[op_name]
+                    args = (op_name, row["bytes"], row["comm_num"])
+                    kw = dict(net=row.get("net", "high_intra_node"))
+                    # the model is affine in 1/eff (bw term + additive
+                    # latency): probe at two eff values, solve exactly
+                    e0 = cfg.efficient_factor
+                    t1 = system_config.compute_net_op_time(*args, **kw)
+                    cfg.efficient_factor = e0 / 2
+                    t2 = system_config.compute_net_op_time(*args, **kw)
+                    A = (t2 - t1) * e0          # bw term at eff=1
+                    L = t1 - A / e0             # additive latency
+                    denom = max(row["ms"] - L, 1e-6)
+                    cfg.efficient_factor = round(
+                        min(max(A / denom, 0.02), 1.5), 4)
+                    n += 1
+                except (KeyError, AttributeError, TypeError,
+                        ZeroDivisionError):
+                    cfg.efficient_factor = e0
+                    continue
+            continue
         if table == "meta":
             if "recompute_factor" in rows:
                 acc.recompute_factor = float(rows["recompute_factor"])

@@ -279,3 +279,26 @@ def test_fp8_cache_bytes_not_in_param_count():
     assert res[True][1] == res[False][1]
     assert res[True][2] > 0 and res[False][2] == 0
     assert res[True][3] > res[False][3]
+
+
+def test_network_overlay_rescales_tier_efficiency():
+    """A measured RCCL all-reduce wall time rescales the xGMI tier's
+    per-op efficiency so the model reproduces the measurement."""
+    from simumax_amd import SystemConfig, get_simu_system_config
+    from simumax_amd.calib.insitu_overlay import apply_insitu_overlay
+
+    sysc = SystemConfig.init_from_config_file(
+        get_simu_system_config("mi355x"))
+    B, n = 160.0 * 1024**2, 8
+    t_pred = sysc.compute_net_op_time("all_reduce", B, n,
+                                      net="high_intra_node")
+    # pretend the real collective is 1.5x slower than the spec guess
+    meas = t_pred * 1.5
+    applied = apply_insitu_overlay(sysc, {"network": {"all_reduce": {
+        "bytes": B, "comm_num": n, "ms": meas,
+        "net": "high_intra_node"}}})
+    assert applied == 1
+    t_new = sysc.compute_net_op_time("all_reduce", B, n,
+                                     net="high_intra_node")
+    # the affine solve reproduces the measurement exactly (to rounding)
+    assert abs(t_new - meas) / meas < 0.01, (t_new, meas)
