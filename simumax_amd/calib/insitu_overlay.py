@@ -19,6 +19,7 @@ def apply_insitu_overlay(system_config, summary) -> int:
             # efficiency so the model's prediction for that (op, bytes,
             # comm_num) matches the measurement exactly
             for op_name, row in rows.items():
+                cfg = e0 = None
                 try:
                     net = system_config.networks[row.get(
                         "net", "high_intra_node")]
@@ -39,7 +40,8 @@ def apply_insitu_overlay(system_config, summary) -> int:
                     n += 1
                 except (KeyError, AttributeError, TypeError,
                         ZeroDivisionError):
-                    cfg.efficient_factor = e0
+                    if cfg is not None and e0 is not None:
+                        cfg.efficient_factor = e0
                     continue
             continue
         if table == "meta":
