@@ -39,11 +39,14 @@ _KNOBS = ("world_size", "tp_size", "pp_size", "ep_size", "cp_size",
           "interleaving_size")
 
 
+_STR_KNOBS = ("cp_comm_type", "cp_sharding")
+
+
 def _run(model: str, system: str, strategy: str, overrides: dict):
     st = StrategyConfig.init_from_config_file(get_simu_strategy_config(strategy))
     for k, v in overrides.items():
         if v is not None:
-            setattr(st, k, int(v))
+            setattr(st, k, str(v) if k in _STR_KNOBS else int(v))
     mc = ModelConfig.init_from_config_file(get_simu_model_config(model))
     sysc = SystemConfig.init_from_config_file(get_simu_system_config(system))
     perf = PerfLLM()
@@ -84,10 +87,12 @@ def api_analyze(model: str = Query("llama3-8b"),
                 pp_size: int = None, ep_size: int = None,
                 cp_size: int = None, micro_batch_size: int = None,
                 micro_batch_num: int = None, seq_len: int = None,
-                interleaving_size: int = None):
+                interleaving_size: int = None,
+                cp_comm_type: str = None, cp_sharding: str = None):
     loc = locals()
-    _, res = _run(model, system, strategy,
-                  {k: loc[k] for k in _KNOBS})
+    over = {k: loc[k] for k in _KNOBS}
+    over.update({k: loc[k] for k in _STR_KNOBS})
+    _, res = _run(model, system, strategy, over)
     return res
 
 

@@ -57,3 +57,13 @@ def test_artifacts_zip(client):
     z = zipfile.ZipFile(io.BytesIO(r.content))
     names = z.namelist()
     assert "analysis.json" in names and "gemm_cost.json" in names
+
+
+def test_api_analyze_cp_mode_knobs(client):
+    r = client.get("/api/analyze", params=dict(
+        model="llama2-tiny", strategy="tp1_pp1_dp8_mbs1", system="mi355x",
+        seq_len=1024, cp_size=4, cp_comm_type="ring", cp_sharding="zigzag"))
+    assert r.status_code == 200
+    body = r.json()
+    assert body["iter_time_ms"] > 0
+    assert "cp4" in body["parallelism"]
